@@ -1,0 +1,59 @@
+"""DeepFM-style tabular binary classifier (BASELINE.json config 4).
+
+fm-1st : per-category scalar weights + linear over dense;
+fm-2nd : factorization-machine pairwise interaction over the F per-feature
+         D-dim embedding vectors: 0.5 * sum_d[(sum_f v_fd)^2 - sum_f v_fd^2];
+deep   : same embeddings concat dense -> MLP tower;
+head   : logits = fm1 + fm2 + deep_head.
+"""
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import torch
+
+from shifu_amd.ops.embedding import MultiEmbedding
+from shifu_amd.ops.linear import FusedLinear
+from shifu_amd.ops.loss import predict_proba
+
+
+class DeepFM(torch.nn.Module):
+    def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
+                 hidden_nodes: List[int], activations: List[str], seed: int = 1234):
+        super().__init__()
+        self.num_dense = num_dense
+        self.embed_dim = embed_dim
+        self.vocab_sizes = list(vocab_sizes)
+        F = len(self.vocab_sizes)
+
+        self.fm_first = MultiEmbedding(self.vocab_sizes, 1, seed=seed + 11)
+        self.fm_dense = FusedLinear(num_dense, 1, activation="none", seed=seed + 12)
+        self.embeddings = MultiEmbedding(self.vocab_sizes, embed_dim, seed=seed + 21)
+
+        tower_in = num_dense + F * embed_dim
+        layers, prev = [], tower_in
+        for i, (h, a) in enumerate(zip(hidden_nodes, activations)):
+            layers.append(FusedLinear(prev, h, activation=a, seed=seed + 31 + i))
+            prev = h
+        self.tower = torch.nn.ModuleList(layers)
+        self.shifu_output_0 = FusedLinear(prev, 1, activation="none", seed=seed + 99)
+
+    def forward(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
+        B = dense.shape[0]
+        F = len(self.vocab_sizes)
+        fm1 = self.fm_first(cats).sum(dim=1) + self.fm_dense(dense).reshape(-1)
+
+        emb_flat = self.embeddings(cats)                   # [B, F*D]
+        v = emb_flat.reshape(B, F, self.embed_dim).float()
+        sum_v = v.sum(dim=1)                               # [B, D]
+        fm2 = 0.5 * (sum_v * sum_v - (v * v).sum(dim=1)).sum(dim=1)  # [B]
+
+        x = torch.cat([dense, emb_flat.to(dense.dtype)], dim=1)
+        for layer in self.tower:
+            x = layer(x)
+        deep = self.shifu_output_0(x).reshape(-1)
+        return fm1.to(deep.dtype) + fm2.to(deep.dtype) + deep
+
+    @torch.no_grad()
+    def predict(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
+        return predict_proba(self.forward(dense, cats))

@@ -1,0 +1,58 @@
+"""Wide&Deep tabular binary classifier (BASELINE.json config 3 — the
+headline benchmark model: 1M-vocab categorical embeddings + 200 dense).
+
+wide  : per-category scalar weights (a D=1 embedding arena) + a linear term
+        over the dense features;
+deep  : concat(dense, per-feature D-dim embeddings) -> MLP tower
+        (FusedLinear MFMA kernels);
+head  : logits = wide + deep_head; sigmoid fused in the loss.
+
+Not present in the reference (dense-only MLPs) — added per BASELINE.json
+configs 3/4 (SURVEY.md §2.4 row "new configs only").
+"""
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import torch
+
+from shifu_amd.ops.embedding import MultiEmbedding
+from shifu_amd.ops.linear import FusedLinear
+from shifu_amd.ops.loss import predict_proba
+
+
+class WideDeep(torch.nn.Module):
+    def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
+                 hidden_nodes: List[int], activations: List[str], seed: int = 1234):
+        super().__init__()
+        self.num_dense = num_dense
+        self.embed_dim = embed_dim
+        self.vocab_sizes = list(vocab_sizes)
+        F = len(self.vocab_sizes)
+
+        # wide part
+        self.wide_cat = MultiEmbedding(self.vocab_sizes, 1, seed=seed + 101)
+        self.wide_dense = FusedLinear(num_dense, 1, activation="none", seed=seed + 102)
+
+        # deep part
+        self.embeddings = MultiEmbedding(self.vocab_sizes, embed_dim, seed=seed + 201)
+        tower_in = num_dense + F * embed_dim
+        layers, prev = [], tower_in
+        for i, (h, a) in enumerate(zip(hidden_nodes, activations)):
+            layers.append(FusedLinear(prev, h, activation=a, seed=seed + 301 + i))
+            prev = h
+        self.tower = torch.nn.ModuleList(layers)
+        self.shifu_output_0 = FusedLinear(prev, 1, activation="none", seed=seed + 999)
+
+    def forward(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
+        wide = self.wide_cat(cats).sum(dim=1) + self.wide_dense(dense).reshape(-1)
+        emb = self.embeddings(cats)                     # [B, F*D]
+        x = torch.cat([dense, emb.to(dense.dtype)], dim=1)
+        for layer in self.tower:
+            x = layer(x)
+        deep = self.shifu_output_0(x).reshape(-1)
+        return wide.to(deep.dtype) + deep               # logits [B]
+
+    @torch.no_grad()
+    def predict(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
+        return predict_proba(self.forward(dense, cats))

@@ -1,0 +1,83 @@
+"""Categorical embedding arena: gather forward, sparse rowwise gradients.
+
+New-config op (Wide&Deep / DeepFM — BASELINE.json configs 3/4; the reference
+is dense-only, SURVEY.md §2.4 row "new configs only").
+
+Design (MI355X-first):
+* ALL F categorical tables live in ONE arena tensor [sum(V_f), D]
+  (bf16 on GPU — a 26x1M x 64 arena is 3.3 GB of the 288 GB HBM; fp32 on CPU),
+  with per-feature row offsets.  One gather kernel serves every feature.
+* forward: ids [B, F] -> out [B, F*D] (concatenated), one coalesced
+  gather kernel (D*2B contiguous bytes per row).
+* backward: returns a torch SPARSE grad (global row indices + dout rows) —
+  a dense grad would be the full arena (GBs) and its all-reduce would
+  dominate the step.  The FusedOptimizer aggregates sparse grads across
+  ranks (allgather of rows over xGMI) and applies a rowwise update kernel
+  (scatter-add with fp32 atomics on GPU — SURVEY.md §2.4 embedding row).
+"""
+from __future__ import annotations
+
+import math
+from typing import List, Sequence
+
+import torch
+
+from shifu_amd.ops.dispatch import use_hip, hip_ops
+
+
+class _EmbGatherFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, arena: torch.Tensor, flat_ids: torch.Tensor, F: int, D: int):
+        # flat_ids: [B, F] global row ids (offsets already added)
+        B = flat_ids.shape[0]
+        if use_hip(arena):
+            ext = hip_ops()
+            out = ext.embedding_gather(arena, flat_ids.contiguous())
+        else:
+            out = arena.index_select(0, flat_ids.reshape(-1)).reshape(B, F * D)
+        ctx.save_for_backward(flat_ids)
+        ctx.arena_shape = arena.shape
+        ctx.F, ctx.D = F, D
+        return out
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        (flat_ids,) = ctx.saved_tensors
+        B = flat_ids.shape[0]
+        values = dout.reshape(B * ctx.F, ctx.D)
+        indices = flat_ids.reshape(1, B * ctx.F)
+        grad = torch.sparse_coo_tensor(indices, values, ctx.arena_shape)
+        return grad, None, None, None
+
+
+class MultiEmbedding(torch.nn.Module):
+    """F categorical features -> concatenated [B, F*D] embeddings from one arena."""
+
+    def __init__(self, vocab_sizes: Sequence[int], dim: int,
+                 seed: int = 0, device: str = "cpu", dtype: torch.dtype = torch.float32):
+        super().__init__()
+        self.vocab_sizes = [int(v) for v in vocab_sizes]
+        self.dim = int(dim)
+        self.total_rows = int(sum(self.vocab_sizes))
+        offsets = torch.tensor(
+            [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
+            dtype=torch.int64)
+        self.register_buffer("offsets", offsets)
+        gen = torch.Generator().manual_seed(seed)
+        scale = 1.0 / math.sqrt(max(self.dim, 1))
+        arena = (torch.rand(self.total_rows, self.dim, generator=gen) * 2 - 1) * scale
+        self.arena = torch.nn.Parameter(arena.to(dtype))
+        self.arena._is_embedding_arena = True  # FusedOptimizer routes this to the sparse path
+
+    @property
+    def num_features(self) -> int:
+        return len(self.vocab_sizes)
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        # ids: [B, F] per-feature local ids; clamp out-of-vocab to last row
+        if ids.shape[1] != self.num_features:
+            raise ValueError(f"ids has {ids.shape[1]} features, expected {self.num_features}")
+        sizes = torch.tensor(self.vocab_sizes, device=ids.device, dtype=ids.dtype)
+        local = ids.clamp(min=0) % sizes  # hash-style fold of out-of-range ids
+        flat = local + self.offsets.to(ids.device)
+        return _EmbGatherFn.apply(self.arena, flat, self.num_features, self.dim)

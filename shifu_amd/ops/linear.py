@@ -1,0 +1,135 @@
+"""Fused dense layer: Y = act(X @ W + b).
+
+Replaces the reference's per-layer `tf.matmul(x,W)+b` + activation
+(reference: ssgd_monitor.py:57-71 nn_layer, activations ssgd_monitor.py:74-88)
+with one fused HIP kernel on GPU (MFMA tiles, bias+activation epilogue —
+SURVEY.md §2.4 K1/K2) and a pure-PyTorch fp32 reference path on CPU.
+
+Mixed-precision contract on GPU:
+* master weights are fp32 autograd leaves (optimizer updates fp32);
+* forward casts W/b to bf16 once per step and runs the bf16 MFMA kernel
+  (f32 accumulate);
+* backward produces dW/db in fp32 (wgrad GEMM accumulates f32 and stores
+  f32), dX in bf16 for the upstream layer.
+
+Activation gradients are computed from Y (not Z): sigmoid' = y(1-y),
+tanh' = 1-y^2, relu'/leakyrelu' from sign(y) — valid because all four
+activations are monotone with act(z)>0 <=> z>0 for (leaky)relu.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from shifu_amd.ops.dispatch import use_hip, hip_ops
+
+# activation ids shared with the HIP side (ops/hip/shifu_ops.hip)
+ACT_NONE, ACT_SIGMOID, ACT_TANH, ACT_RELU, ACT_LEAKYRELU = 0, 1, 2, 3, 4
+_ACT_IDS = {"none": ACT_NONE, "linear": ACT_NONE, "sigmoid": ACT_SIGMOID,
+            "tanh": ACT_TANH, "relu": ACT_RELU, "leakyrelu": ACT_LEAKYRELU}
+LEAKY_SLOPE = 0.01  # tf.nn.leaky_relu default alpha (ssgd_monitor.py:86)
+
+
+def act_id(name: str) -> int:
+    return _ACT_IDS[name.lower()]
+
+
+def _act_fwd_ref(z: torch.Tensor, act: int) -> torch.Tensor:
+    if act == ACT_SIGMOID:
+        return torch.sigmoid(z)
+    if act == ACT_TANH:
+        return torch.tanh(z)
+    if act == ACT_RELU:
+        return torch.relu(z)
+    if act == ACT_LEAKYRELU:
+        return torch.nn.functional.leaky_relu(z, LEAKY_SLOPE)
+    return z
+
+
+def _act_grad_from_y_ref(dy: torch.Tensor, y: torch.Tensor, act: int) -> torch.Tensor:
+    if act == ACT_SIGMOID:
+        return dy * y * (1.0 - y)
+    if act == ACT_TANH:
+        return dy * (1.0 - y * y)
+    if act == ACT_RELU:
+        return dy * (y > 0).to(dy.dtype)
+    if act == ACT_LEAKYRELU:
+        return dy * torch.where(y > 0, torch.ones_like(y),
+                                torch.full_like(y, LEAKY_SLOPE))
+    return dy
+
+
+class _FusedLinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor, act: int):
+        if use_hip(x):
+            ext = hip_ops()
+            wb = w.to(torch.bfloat16)
+            bb = b.to(torch.bfloat16)
+            xb = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
+            y = ext.linear_act_fwd(xb, wb, bb, act)
+            ctx.save_for_backward(xb, wb, y)
+        else:
+            z = x @ w + b
+            y = _act_fwd_ref(z, act)
+            ctx.save_for_backward(x, w, y)
+        ctx.act = act
+        ctx.hip = use_hip(x)
+        ctx.x_needs_grad = x.requires_grad
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w, y = ctx.saved_tensors
+        act = ctx.act
+        if ctx.hip:
+            ext = hip_ops()
+            dz = ext.act_grad(dy.contiguous(), y, act)          # bf16
+            dw = ext.gemm_tn_f32(x, dz)                         # fp32 [K,N] = x^T @ dz
+            db = ext.colsum_f32(dz)                             # fp32 [N]
+            dx = ext.gemm_nt_bf16(dz, w) if ctx.x_needs_grad else None  # bf16 [B,K] = dz @ w^T
+        else:
+            dz = _act_grad_from_y_ref(dy, y, act)
+            dw = x.t() @ dz
+            db = dz.sum(dim=0)
+            dx = dz @ w.t() if ctx.x_needs_grad else None
+        return dx, dw, db, None
+
+
+def fused_linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
+                 activation: str = "none") -> torch.Tensor:
+    return _FusedLinearFn.apply(x, w, b, act_id(activation))
+
+
+class FusedLinear(torch.nn.Module):
+    """Dense layer with xavier init + fused forward.
+
+    Init matches the reference's nn_layer: xavier/glorot for W
+    (tf.contrib.layers.xavier_initializer, ssgd_monitor.py:63) and small
+    truncated-normal-style bias.  The L2(0.1) regularizer on W
+    (ssgd_monitor.py:58-68) is folded into the optimizer as coupled weight
+    decay (grad += l2 * w), not materialized in the loss.
+    """
+
+    def __init__(self, in_features: int, out_features: int,
+                 activation: str = "none", seed: Optional[int] = None):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.activation = activation.lower()
+        self._act = act_id(self.activation)
+        gen = None
+        if seed is not None:
+            gen = torch.Generator().manual_seed(seed)
+        limit = math.sqrt(6.0 / (in_features + out_features))
+        w = (torch.rand(in_features, out_features, generator=gen) * 2 - 1) * limit
+        self.weight = torch.nn.Parameter(w)                    # [K, N] layout (x @ w)
+        self.bias = torch.nn.Parameter(torch.zeros(out_features))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return _FusedLinearFn.apply(x, self.weight, self.bias, self._act)
+
+    def extra_repr(self) -> str:
+        return f"in={self.in_features}, out={self.out_features}, act={self.activation}"

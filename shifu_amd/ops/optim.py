@@ -1,0 +1,166 @@
+"""Fused optimizers over the flat parameter arena + rowwise sparse embedding updates.
+
+Dense path: ONE kernel per step over the flat fp32
+master/grad/moment buffers — Adam, Adadelta (TF semantics, the reference
+default: tf.train.AdadeltaOptimizer, ssgd_monitor.py:138), and SGD.  The
+reference's per-layer l2_regularizer(0.1) (ssgd_monitor.py:58-68) is folded
+in as coupled weight decay: g += l2 * w (tf l2_regularizer(s)(w) = s*||w||^2/2
+=> d/dw = s*w).
+
+Sparse path: embedding arenas get rowwise updates on coalesced sparse grads
+(unique rows after coalesce -> no atomics needed in the update kernel):
+"sgd" or rowwise "adagrad" (fp32 accumulator per row).
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import torch
+
+from shifu_amd.ops.dispatch import use_hip, hip_ops
+from shifu_amd.ops.flat import FlatParams
+
+OPT_SGD, OPT_ADAM, OPT_ADADELTA, OPT_ADAGRAD = 0, 1, 2, 3
+_OPT_IDS = {"sgd": OPT_SGD, "adam": OPT_ADAM, "adadelta": OPT_ADADELTA,
+            "adagrad": OPT_ADAGRAD}
+
+
+class FusedOptimizer:
+    """Optimizer over (FlatParams dense arena, embedding arena params)."""
+
+    def __init__(self, flat: FlatParams, emb_params: Optional[List[torch.nn.Parameter]] = None,
+                 optimizer: str = "adadelta", lr: float = 1.0, l2_reg: float = 0.0,
+                 betas=(0.9, 0.999), eps: float = 1e-8, rho: float = 0.95,
+                 emb_optimizer: str = "adagrad", emb_lr: Optional[float] = None):
+        self.flat = flat
+        self.emb_params = list(emb_params or [])
+        self.kind = _OPT_IDS[optimizer.lower()]
+        self.lr = float(lr)
+        self.l2 = float(l2_reg)
+        self.b1, self.b2 = betas
+        self.eps = float(eps)
+        self.rho = float(rho)
+        self.step_count = 0
+        self.emb_kind = _OPT_IDS[emb_optimizer.lower()]
+        self.emb_lr = float(emb_lr if emb_lr is not None else lr)
+
+        n = flat.numel()
+        dev = flat.flat.device if n else torch.device("cpu")
+        # moment buffers (fp32): adam: m,v ; adadelta: accum, delta_accum ; adagrad: m
+        self.m = torch.zeros(n, device=dev) if self.kind in (OPT_ADAM, OPT_ADADELTA, OPT_ADAGRAD) else None
+        self.v = torch.zeros(n, device=dev) if self.kind in (OPT_ADAM, OPT_ADADELTA) else None
+        # rowwise fp32 accumulators for embedding adagrad
+        self.emb_state: Dict[int, torch.Tensor] = {}
+        if self.emb_kind == OPT_ADAGRAD:
+            for i, p in enumerate(self.emb_params):
+                self.emb_state[i] = torch.zeros(p.shape[0], device=p.device)
+
+    # ------------------------------------------------------------------ dense
+    def _dense_step_ref(self) -> None:
+        w, g = self.flat.flat, self.flat.flat_grad
+        if self.l2 != 0.0:
+            g = g.add(w, alpha=self.l2)
+        if self.kind == OPT_SGD:
+            w.add_(g, alpha=-self.lr)
+        elif self.kind == OPT_ADAM:
+            t = self.step_count
+            self.m.mul_(self.b1).add_(g, alpha=1 - self.b1)
+            self.v.mul_(self.b2).addcmul_(g, g, value=1 - self.b2)
+            bc1 = 1 - self.b1 ** t
+            bc2 = 1 - self.b2 ** t
+            # update = lr/bc1 * m / (sqrt(v)/sqrt(bc2) + eps)  (torch-Adam form)
+            step = self.lr * math.sqrt(bc2) / bc1
+            denom = self.v.sqrt().add_(self.eps * math.sqrt(bc2))
+            w.addcdiv_(self.m, denom, value=-step)
+        elif self.kind == OPT_ADADELTA:
+            # TF AdadeltaOptimizer semantics (rho, eps inside both sqrts)
+            self.m.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
+            upd = g * (self.v.add(self.eps).sqrt_() / self.m.add(self.eps).sqrt_())
+            self.v.mul_(self.rho).addcmul_(upd, upd, value=1 - self.rho)
+            w.add_(upd, alpha=-self.lr)
+        elif self.kind == OPT_ADAGRAD:
+            self.m.addcmul_(g, g, value=1.0)
+            w.addcdiv_(g, self.m.sqrt().add(self.eps), value=-self.lr)
+
+    def _dense_step_hip(self) -> None:
+        ext = hip_ops()
+        w, g = self.flat.flat, self.flat.flat_grad
+        if self.kind == OPT_SGD:
+            ext.sgd_step(w, g, self.lr, self.l2)
+        elif self.kind == OPT_ADAM:
+            ext.adam_step(w, g, self.m, self.v, self.lr, self.b1, self.b2,
+                          self.eps, self.l2, self.step_count)
+        elif self.kind == OPT_ADADELTA:
+            ext.adadelta_step(w, g, self.m, self.v, self.lr, self.rho,
+                              self.eps, self.l2)
+        elif self.kind == OPT_ADAGRAD:
+            ext.adagrad_step(w, g, self.m, self.lr, self.eps, self.l2)
+
+    # ----------------------------------------------------------------- sparse
+    def _emb_step(self, p: torch.nn.Parameter, idx: int) -> None:
+        if p.grad is None:
+            return
+        g = p.grad
+        if g.is_sparse:
+            g = g.coalesce()
+            rows, vals = g.indices()[0], g.values()
+        else:  # dense grad on a small arena (CPU tests)
+            rows = torch.nonzero(g.abs().sum(dim=1) != 0, as_tuple=False).reshape(-1)
+            vals = g[rows]
+        if rows.numel() == 0:
+            return
+        vals = vals.float()
+        if self.emb_kind == OPT_ADAGRAD:
+            acc = self.emb_state[idx]
+            if use_hip(p):
+                hip_ops().emb_adagrad_step(p.data, acc, rows.contiguous(),
+                                           vals.contiguous(), self.emb_lr, self.eps)
+            else:
+                rowsq = (vals * vals).mean(dim=1)
+                acc.index_add_(0, rows, rowsq)
+                denom = acc[rows].add(self.eps).sqrt_().unsqueeze(1)
+                p.data.index_add_(0, rows, (-self.emb_lr * vals / denom).to(p.dtype))
+        else:  # sgd
+            if use_hip(p):
+                hip_ops().emb_sgd_step(p.data, rows.contiguous(), vals.contiguous(),
+                                       self.emb_lr)
+            else:
+                p.data.index_add_(0, rows, (-self.emb_lr * vals).to(p.dtype))
+        p.grad = None
+
+    # ------------------------------------------------------------------ api
+    @torch.no_grad()
+    def step(self) -> None:
+        self.step_count += 1
+        if self.flat.numel():
+            self.flat.sync_grads()
+            if use_hip(self.flat.flat):
+                self._dense_step_hip()
+            else:
+                self._dense_step_ref()
+        for i, p in enumerate(self.emb_params):
+            self._emb_step(p, i)
+
+    def zero_grad(self) -> None:
+        self.flat.zero_grad()
+        for p in self.emb_params:
+            p.grad = None
+
+    # ------------------------------------------------------------ checkpoint
+    def state_dict(self) -> dict:
+        return {
+            "step_count": self.step_count,
+            "m": self.m, "v": self.v,
+            "emb_state": self.emb_state,
+        }
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.step_count = int(sd["step_count"])
+        if self.m is not None and sd.get("m") is not None:
+            self.m.copy_(sd["m"])
+        if self.v is not None and sd.get("v") is not None:
+            self.v.copy_(sd["v"])
+        for k, t in (sd.get("emb_state") or {}).items():
+            if int(k) in self.emb_state:
+                self.emb_state[int(k)].copy_(t)
