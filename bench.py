@@ -1,0 +1,152 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Wide&Deep tabular training throughput (samples/sec,
+whole node) — BASELINE.json headline config 3: 1M-vocab categorical
+embeddings (26 features) + 200 dense numerics, bf16 compute on MI355X.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+the driver launches via torch.distributed.run with one rank per GPU (RCCL).
+W untimed warmup steps, then exactly K timed steps bracketed by
+barrier+synchronize on both sides; elapsed is the MAX over ranks; rank 0
+prints ONE JSON line.  Each timed step is a FULL training step: embedding
+gather + fused GEMM tower forward, fused loss, backward GEMMs, bucketed
+all-reduce (dense) + sparse allgather (embeddings), fused optimizer update.
+Synthetic data (no network), random-init weights, weak scaling (per-GPU
+batch fixed).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from shifu_amd.config.model_config import ModelConfig
+from shifu_amd.config.run_config import RunConfig
+from shifu_amd.models.wide_deep import WideDeep
+from shifu_amd.ops.flat import FlatParams, split_params
+from shifu_amd.ops.loss import weighted_loss
+from shifu_amd.ops.optim import FusedOptimizer
+from shifu_amd.parallel.dist import (GradAggregator, destroy_distributed,
+                                     init_distributed, is_distributed)
+
+# headline model config (BASELINE.json config 3)
+N_DENSE = 200
+N_CAT = 26
+VOCAB = 1_000_000
+EMBED_DIM = 64
+TOWER = [1024, 512, 256]
+ACTS = ["relu", "relu", "relu"]
+PER_GPU_BATCH = 8192
+N_BATCHES = 8  # distinct resident batches cycled through the loop
+
+
+def make_batches(device, batch, rank, dtype):
+    gen = torch.Generator(device="cpu").manual_seed(1234 + rank)
+    batches = []
+    for i in range(N_BATCHES):
+        dense = torch.randn(batch, N_DENSE, generator=gen).to(device=device, dtype=dtype)
+        cats = torch.randint(0, VOCAB, (batch, N_CAT), generator=gen).to(device)
+        target = (torch.rand(batch, generator=gen) > 0.5).float().to(device)
+        weight = torch.ones(batch, device=device)
+        batches.append((dense, cats, target, weight))
+    return batches
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
+    ap.add_argument("--bucket-mb", type=int, default=128)
+    ap.add_argument("--embed-dim", type=int, default=EMBED_DIM)
+    args = ap.parse_args()
+
+    rank, world, device = init_distributed()
+    n_gpus = world if world > 1 else args.gpus
+    on_gpu = device.type == "cuda"
+    dtype = torch.bfloat16 if on_gpu else torch.float32
+
+    torch.manual_seed(777)
+    model = WideDeep(N_DENSE, [VOCAB] * N_CAT, args.embed_dim, TOWER, ACTS,
+                     seed=777).to(device)
+    if on_gpu:
+        # keep embedding arenas bf16 (HBM-resident, gathered by the HIP kernel)
+        for p in model.parameters():
+            if getattr(p, "_is_embedding_arena", False):
+                p.data = p.data.to(torch.bfloat16)
+
+    dense_params, emb_params = split_params(model)
+    flat = FlatParams(dense_params)
+    agg = GradAggregator(flat, emb_params, bucket_mb=args.bucket_mb, overlap=True)
+    opt = FusedOptimizer(flat, emb_params, optimizer="adam", lr=1e-3,
+                         l2_reg=0.0, emb_optimizer="adagrad", emb_lr=0.01)
+
+    batches = make_batches(device, args.batch, rank, dtype)
+
+    def step(i):
+        dense, cats, target, weight = batches[i % N_BATCHES]
+        logits = model(dense, cats)
+        loss = weighted_loss(logits, target, weight, "sigmoid_ce")
+        loss.backward()
+        agg.finish()
+        opt.step()
+        opt.zero_grad()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+
+    if is_distributed():
+        torch.distributed.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.time()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if on_gpu:
+        torch.cuda.synchronize()
+    if is_distributed():
+        torch.distributed.barrier()
+    elapsed = time.time() - t0
+
+    # max elapsed over ranks
+    if is_distributed():
+        t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t)
+
+    if rank == 0:
+        samples = n_gpus * args.batch * args.steps
+        out = {
+            "metric": "samples_per_sec_wide_deep",
+            "value": samples / elapsed,
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,  # reference publishes no numbers
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"wide_deep[{N_CAT}x{VOCAB}vocab*{args.embed_dim}d+{N_DENSE}dense,tower{TOWER}]",
+                "global_batch": n_gpus * args.batch,
+                "per_gpu_batch": args.batch,
+                "seq_len": None,
+                "parallelism": f"dp{n_gpus}",
+                "optimizer": "adam+rowwise_adagrad(emb)",
+                "loss": "sigmoid_ce",
+            },
+        }
+        print(json.dumps(out), flush=True)
+    destroy_distributed()
+
+
+if __name__ == "__main__":
+    main()

@@ -25,6 +25,25 @@ import torch
 from shifu_amd.ops.dispatch import use_hip, hip_ops
 
 
+def sparse_rows_values(g: torch.Tensor):
+    """(unique_rows int64, fp32 values [n,D]) from an uncoalesced sparse grad.
+
+    Manual coalesce: torch's sparse .coalesce() on bf16 CUDA values is not a
+    dependency we want on ROCm; unique+index_add in fp32 is robust and is
+    the precision we aggregate in anyway."""
+    if not g.is_sparse:
+        rows = torch.nonzero(g.abs().sum(dim=1) != 0, as_tuple=False).reshape(-1)
+        return rows, g[rows].float()
+    idx = g._indices()[0]
+    vals = g._values().float()
+    if idx.numel() == 0:
+        return idx, vals
+    uniq, inverse = torch.unique(idx, return_inverse=True)
+    out = torch.zeros(uniq.numel(), vals.shape[1], device=vals.device)
+    out.index_add_(0, inverse, vals)
+    return uniq, out
+
+
 class _EmbGatherFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, arena: torch.Tensor, flat_ids: torch.Tensor, F: int, D: int):

@@ -1,0 +1,722 @@
+// shifu_ops.hip — hand-written CDNA4 (gfx950 / MI355X) kernels for shifu_amd.
+//
+// Implements the MI355X-native replacements for the tensor ops the reference's
+// TF graphs instantiate:
+//   K1  fused dense GEMM + bias + activation forward (MFMA bf16, f32 accum)
+//   K2  backward GEMMs: dX = dZ·Wᵀ (NT), dW = Xᵀ·dZ (TN, f32 out), db colsum,
+//       fused activation-gradient
+//   K3  fused sigmoid + weighted MSE / sigmoid-CE loss (+ gradient)
+//   K4  fused Adam / Adadelta(TF semantics) / SGD / Adagrad over the flat
+//       fp32 parameter arena
+//   --  embedding arena gather + rowwise sparse updates (Wide&Deep/DeepFM)
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//   * wave64; blocks are 256 threads (4 waves)
+//   * GEMM: 128x128 tile, BK=32, mfma_f32_16x16x32_bf16, LDS staged with
+//     +16B row padding (conflict-free ds_read_b128 column groups), B held
+//     transposed in LDS so B-fragments are contiguous 16B reads
+//   * bf16 global loads vectorized (short4/short8-equivalent widths)
+//   * elementwise/optimizer kernels are grid-stride, f32x4 / bf16x8 vectorized
+// No CUDA compatibility paths; gfx950 only.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <vector>
+
+using bf16 = __hip_bfloat16;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+typedef __attribute__((ext_vector_type(4))) short s16x4;   // 8B  = 4 bf16
+typedef __attribute__((ext_vector_type(8))) short s16x8;   // 16B = 8 bf16
+
+#define DEVINL __device__ __forceinline__
+
+static inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_GPU(x) TORCH_CHECK((x).is_cuda(), #x " must be on GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+#define CHECK_BF16(x) TORCH_CHECK((x).scalar_type() == at::kBFloat16, #x " must be bf16")
+#define CHECK_F32(x) TORCH_CHECK((x).scalar_type() == at::kFloat, #x " must be fp32")
+
+// ---------------------------------------------------------------------------
+// activations (ids shared with shifu_amd/ops/linear.py)
+// ---------------------------------------------------------------------------
+enum Act { ACT_NONE = 0, ACT_SIGMOID = 1, ACT_TANH = 2, ACT_RELU = 3, ACT_LEAKY = 4 };
+#define LEAKY_SLOPE 0.01f
+
+DEVINL float act_fwd(float z, int act) {
+  switch (act) {
+    case ACT_SIGMOID: return 1.0f / (1.0f + __expf(-z));
+    case ACT_TANH:    return tanhf(z);
+    case ACT_RELU:    return z > 0.0f ? z : 0.0f;
+    case ACT_LEAKY:   return z > 0.0f ? z : LEAKY_SLOPE * z;
+    default:          return z;
+  }
+}
+
+// activation gradient expressed through y = act(z) (valid for this act set)
+DEVINL float act_grad_from_y(float y, int act) {
+  switch (act) {
+    case ACT_SIGMOID: return y * (1.0f - y);
+    case ACT_TANH:    return 1.0f - y * y;
+    case ACT_RELU:    return y > 0.0f ? 1.0f : 0.0f;
+    case ACT_LEAKY:   return y > 0.0f ? 1.0f : LEAKY_SLOPE;
+    default:          return 1.0f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// GEMM: C[M,N] = A_eff[M,K] * B_eff[K,N]  (bf16 in, f32 accumulate)
+//   TA=0: A stored [M,K] (lda=K)   TA=1: A stored [K,M] (lda=M), A_eff=A^T
+//   TB=0: B stored [K,N] (ldb=N)   TB=1: B stored [N,K] (ldb=K), B_eff=B^T
+// Epilogue: EPI_PLAIN (bf16 out), EPI_BIAS_ACT (bf16 out, +bias, act),
+//           EPI_F32 (f32 out — wgrad)
+// One 128x128 tile per 256-thread block; per wave a 64x64 sub-tile =
+// 4x4 fragments of 16x16; mfma_f32_16x16x32_bf16 over BK=32 K-steps.
+// ---------------------------------------------------------------------------
+enum Epi { EPI_PLAIN = 0, EPI_BIAS_ACT = 1, EPI_F32 = 2 };
+
+#define BM 128
+#define BN 128
+#define BK 32
+#define BKP (BK + 8)   // +16B row pad: 20-dword row stride -> conflict-free b128 groups
+
+template <int TA, int TB, int EPI, typename OUT_T>
+__global__ __launch_bounds__(256)
+void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                      OUT_T* __restrict__ C, const bf16* __restrict__ bias,
+                      int M, int N, int K, int act) {
+  __shared__ bf16 As[BM][BKP];   // A_eff[m][k]
+  __shared__ bf16 Bs[BN][BKP];   // B_eff[k][n] transposed: Bs[n][k]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;                    // 0..3
+  const int wr = (wave >> 1) * 64;              // wave row offset in tile
+  const int wc = (wave & 1) * 64;               // wave col offset in tile
+  const int m0 = blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int r16 = lane & 15;        // fragment row/col within 16
+  const int kgrp = lane >> 4;       // 0..3 -> k-offset group *8
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage A tile: As[m][k] = A_eff[m0+m][k0+k] ----
+    if (TA == 0) {
+      // A[M,K]: read 8 bf16 along k per thread (BM*(BK/8)=512 chunks, 2/thread)
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int i = tid + it * 256;
+        int m = i >> 2;                 // /(BK/8)
+        int kc = (i & 3) * 8;
+        int gm = m0 + m, gk = k0 + kc;
+        s16x4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+        if (gm < M) {
+          const bf16* src = A + (long)gm * K + gk;
+          if (gk + 8 <= K) {
+            v0 = *(const s16x4*)(src);
+            v1 = *(const s16x4*)(src + 4);
+          } else {
+            for (int j = 0; j < 8 && gk + j < K; ++j)
+              ((short*)(j < 4 ? &v0 : &v1))[j & 3] = ((const short*)src)[j];
+          }
+        }
+        *(s16x4*)&As[m][kc] = v0;
+        *(s16x4*)&As[m][kc + 4] = v1;
+      }
+    } else {
+      // A[K,M]: read 8 bf16 along m (coalesced), scatter-transpose into LDS
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int i = tid + it * 256;
+        int k = i >> 4;                 // /(BM/8)
+        int mc = (i & 15) * 8;
+        int gk = k0 + k, gm = m0 + mc;
+        short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (gk < K) {
+          const bf16* src = A + (long)gk * M + gm;
+          if (gm + 8 <= M) {
+            *(s16x4*)&tmp[0] = *(const s16x4*)(src);
+            *(s16x4*)&tmp[4] = *(const s16x4*)(src + 4);
+          } else {
+            for (int j = 0; j < 8 && gm + j < M; ++j) tmp[j] = ((const short*)src)[j];
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ((short*)&As[mc + j][k])[0] = tmp[j];
+      }
+    }
+
+    // ---- stage B tile: Bs[n][k] = B_eff[k0+k][n0+n] ----
+    if (TB == 0) {
+      // B[K,N]: read 8 bf16 along n (coalesced), scatter-transpose
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int i = tid + it * 256;
+        int k = i >> 4;
+        int nc = (i & 15) * 8;
+        int gk = k0 + k, gn = n0 + nc;
+        short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        if (gk < K) {
+          const bf16* src = B + (long)gk * N + gn;
+          if (gn + 8 <= N) {
+            *(s16x4*)&tmp[0] = *(const s16x4*)(src);
+            *(s16x4*)&tmp[4] = *(const s16x4*)(src + 4);
+          } else {
+            for (int j = 0; j < 8 && gn + j < N; ++j) tmp[j] = ((const short*)src)[j];
+          }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ((short*)&Bs[nc + j][k])[0] = tmp[j];
+      }
+    } else {
+      // B[N,K]: rows are k-contiguous -> vector LDS writes, no transpose
+#pragma unroll
+      for (int it = 0; it < 2; ++it) {
+        int i = tid + it * 256;
+        int n = i >> 2;
+        int kc = (i & 3) * 8;
+        int gn = n0 + n, gk = k0 + kc;
+        s16x4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+        if (gn < N) {
+          const bf16* src = B + (long)gn * K + gk;
+          if (gk + 8 <= K) {
+            v0 = *(const s16x4*)(src);
+            v1 = *(const s16x4*)(src + 4);
+          } else {
+            for (int j = 0; j < 8 && gk + j < K; ++j)
+              ((short*)(j < 4 ? &v0 : &v1))[j & 3] = ((const short*)src)[j];
+          }
+        }
+        *(s16x4*)&Bs[n][kc] = v0;
+        *(s16x4*)&Bs[n][kc + 4] = v1;
+      }
+    }
+
+    __syncthreads();
+
+    // ---- MFMA: 2 k-steps of 32 per k-chunk wait BK=32 -> 1 step ----
+#pragma unroll
+    for (int fi = 0; fi < 4; ++fi) {
+      bf16x8 a_frag = *(const bf16x8*)&As[wr + fi * 16 + r16][kgrp * 8];
+#pragma unroll
+      for (int fj = 0; fj < 4; ++fj) {
+        bf16x8 b_frag = *(const bf16x8*)&Bs[wc + fj * 16 + r16][kgrp * 8];
+        acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_frag, b_frag, acc[fi][fj], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: D row = (lane>>4)*4 + r, col = lane&15 (per 16x16 frag) ----
+#pragma unroll
+  for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+      int col = n0 + wc + fj * 16 + r16;
+      if (col >= N) continue;
+      float bv = 0.0f;
+      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr + fi * 16 + kgrp * 4 + r;
+        if (row >= M) continue;
+        float v = acc[fi][fj][r];
+        if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
+        if (EPI == EPI_F32) {
+          ((float*)C)[(long)row * N + col] = v;
+        } else {
+          ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
+        }
+      }
+    }
+  }
+}
+
+template <int TA, int TB, int EPI, typename OUT_T>
+static void launch_gemm(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
+                        long M, long N, long K, int act, hipStream_t s) {
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+  hipLaunchKernelGGL((gemm_tile_kernel<TA, TB, EPI, OUT_T>), grid, dim3(256), 0, s,
+                     A, B, C, bias, (int)M, (int)N, (int)K, act);
+}
+
+// ------------------------------ GEMM entry points ---------------------------
+at::Tensor linear_act_fwd(at::Tensor x, at::Tensor w, at::Tensor b, long act) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long Mb = x.size(0), K = x.size(1), N = w.size(1);
+  TORCH_CHECK(w.size(0) == K, "shape mismatch x@w");
+  auto y = at::empty({Mb, N}, x.options());
+  launch_gemm<0, 0, EPI_BIAS_ACT, bf16>(
+      (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(), (bf16*)y.data_ptr(),
+      (const bf16*)b.data_ptr(), Mb, N, K, (int)act, cur_stream());
+  return y;
+}
+
+at::Tensor gemm_nn_bf16(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long M = a.size(0), K = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == K, "shape mismatch");
+  auto c = at::empty({M, N}, a.options());
+  launch_gemm<0, 0, EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                                     (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+  return c;
+}
+
+// dx[B,K] = dz[B,N] @ w[K,N]^T
+at::Tensor gemm_nt_bf16(at::Tensor dz, at::Tensor w) {
+  CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
+  CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  long Mb = dz.size(0), N = dz.size(1), K = w.size(0);
+  TORCH_CHECK(w.size(1) == N, "shape mismatch dz@w^T");
+  auto dx = at::empty({Mb, K}, dz.options());
+  // C[M=Mb, N'=K] = dz[M,Kr=N] * (w^T)[Kr=N, N'=K]; w stored [K,N] => TB=1
+  launch_gemm<0, 1, EPI_PLAIN, bf16>((const bf16*)dz.data_ptr(), (const bf16*)w.data_ptr(),
+                                     (bf16*)dx.data_ptr(), nullptr, Mb, K, N, 0, cur_stream());
+  return dx;
+}
+
+// dw[K,N] = x[B,K]^T @ dz[B,N]   (f32 out)
+at::Tensor gemm_tn_f32(at::Tensor x, at::Tensor dz) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
+  long Bb = x.size(0), K = x.size(1), N = dz.size(1);
+  TORCH_CHECK(dz.size(0) == Bb, "shape mismatch x^T@dz");
+  auto dw = at::empty({K, N}, x.options().dtype(at::kFloat));
+  // C[M=K, N'=N] = (x^T)[M=K, Kr=B] * dz[Kr=B, N]; x stored [B,K] => TA=1
+  launch_gemm<1, 0, EPI_F32, float>((const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
+                                    (float*)dw.data_ptr(), nullptr, K, N, Bb, 0, cur_stream());
+  return dw;
+}
+
+// single-MFMA probe: d[16,16] = a[16,32] @ b[32,16] — fragment-mapping unit test
+__global__ void mfma_probe_kernel(const bf16* a, const bf16* b, float* d) {
+  int lane = threadIdx.x & 63;
+  int r16 = lane & 15, kgrp = lane >> 4;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ((__bf16*)&af)[j] = *(const __bf16*)&a[r16 * 32 + kgrp * 8 + j];
+    ((__bf16*)&bf)[j] = *(const __bf16*)&b[(kgrp * 8 + j) * 16 + r16];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(kgrp * 4 + r) * 16 + r16] = acc[r];
+}
+
+at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_BF16(a); CHECK_GPU(b); CHECK_BF16(b);
+  auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     (const bf16*)a.contiguous().data_ptr(),
+                     (const bf16*)b.contiguous().data_ptr(), (float*)d.data_ptr());
+  return d;
+}
+
+// ---------------------------------------------------------------------------
+// elementwise: fused activation gradient  dz = dy * act'(y)
+// ---------------------------------------------------------------------------
+__global__ void act_grad_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ y,
+                                bf16* __restrict__ dz, long n, int act) {
+  long i = (long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  // vectorized by 8 (16B loads)
+  long n8 = n / 8;
+  for (long v = i; v < n8; v += stride) {
+    s16x8 dyv = ((const s16x8*)dy)[v];
+    s16x8 yv = ((const s16x8*)y)[v];
+    s16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float dyf = __bfloat162float(((const bf16*)&dyv)[j]);
+      float yf = __bfloat162float(((const bf16*)&yv)[j]);
+      ((bf16*)&out)[j] = __float2bfloat16(dyf * act_grad_from_y(yf, act));
+    }
+    ((s16x8*)dz)[v] = out;
+  }
+  for (long j = n8 * 8 + i; j < n; j += stride) {
+    float dyf = __bfloat162float(dy[j]);
+    float yf = __bfloat162float(y[j]);
+    dz[j] = __float2bfloat16(dyf * act_grad_from_y(yf, act));
+  }
+}
+
+at::Tensor act_grad(at::Tensor dy, at::Tensor y, long act) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+  auto dz = at::empty_like(dy);
+  long n = dy.numel();
+  int blocks = (int)std::min((n + 2047) / 2048 + 1, (long)2048);
+  hipLaunchKernelGGL(act_grad_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                     (bf16*)dz.data_ptr(), n, (int)act);
+  return dz;
+}
+
+// ---------------------------------------------------------------------------
+// colsum: db[N] = sum_b dz[b][n]  (f32 out)
+// thread t of block bx owns column bx*256+t; row reads are coalesced.
+// ---------------------------------------------------------------------------
+__global__ void colsum_kernel(const bf16* __restrict__ dz, float* __restrict__ db,
+                              long B, long N) {
+  long n = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  float acc = 0.0f;
+  for (long b = 0; b < B; ++b) acc += __bfloat162float(dz[b * N + n]);
+  db[n] = acc;
+}
+
+at::Tensor colsum_f32(at::Tensor dz) {
+  CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
+  long B = dz.size(0), N = dz.size(1);
+  auto db = at::empty({N}, dz.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(colsum_kernel, dim3((N + 255) / 256), dim3(256), 0, cur_stream(),
+                     (const bf16*)dz.data_ptr(), (float*)db.data_ptr(), B, N);
+  return db;
+}
+
+// ---------------------------------------------------------------------------
+// fused sigmoid + weighted loss (K3)
+// fwd: p = sigmoid(z); per = w*(p-y)^2 (wmse) or w*bce (ce);
+//      reduces loss_sum and wsum (block reduce + atomicAdd)
+// bwd: dz = scale * w * dper/dz
+// ---------------------------------------------------------------------------
+enum LossKind { LOSS_WMSE = 0, LOSS_CE = 1 };
+
+__global__ void loss_fwd_kernel(const bf16* __restrict__ z, const float* __restrict__ y,
+                                const float* __restrict__ w, float* __restrict__ p,
+                                float* __restrict__ loss_sum, float* __restrict__ wsum,
+                                long n, int kind) {
+  __shared__ float red[2][8];   // per-wave partials
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  float ls = 0.0f, ws = 0.0f;
+  for (long j = i; j < n; j += stride) {
+    float zf = __bfloat162float(z[j]);
+    float pf = 1.0f / (1.0f + __expf(-zf));
+    p[j] = pf;
+    float wf = w[j], yf = y[j];
+    float per;
+    if (kind == LOSS_WMSE) {
+      float d = pf - yf;
+      per = wf * d * d;
+    } else {
+      // stable bce-with-logits: max(z,0) - z*y + log1p(exp(-|z|))
+      per = wf * (fmaxf(zf, 0.0f) - zf * yf + log1pf(__expf(-fabsf(zf))));
+    }
+    ls += per;
+    ws += wf;
+  }
+  // wave reduce
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    ls += __shfl_down(ls, off, 64);
+    ws += __shfl_down(ws, off, 64);
+  }
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  if (lane == 0) { red[0][wave] = ls; red[1][wave] = ws; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float l = 0, s = 0;
+    for (int wv = 0; wv < (int)(blockDim.x >> 6); ++wv) { l += red[0][wv]; s += red[1][wv]; }
+    atomicAdd(loss_sum, l);
+    atomicAdd(wsum, s);
+  }
+}
+
+std::vector<at::Tensor> weighted_loss_fwd(at::Tensor z, at::Tensor y, at::Tensor w, long kind) {
+  CHECK_GPU(z); CHECK_CONTIG(z); CHECK_BF16(z);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_F32(y);
+  CHECK_GPU(w); CHECK_CONTIG(w); CHECK_F32(w);
+  long n = z.numel();
+  auto p = at::empty({n}, z.options().dtype(at::kFloat));
+  auto loss_sum = at::zeros({}, z.options().dtype(at::kFloat));
+  auto wsum = at::zeros({}, z.options().dtype(at::kFloat));
+  int blocks = (int)std::min((n + 511) / 512 + 1, (long)1024);
+  hipLaunchKernelGGL(loss_fwd_kernel, dim3(blocks), dim3(512), 0, cur_stream(),
+                     (const bf16*)z.data_ptr(), (const float*)y.data_ptr(),
+                     (const float*)w.data_ptr(), (float*)p.data_ptr(),
+                     (float*)loss_sum.data_ptr(), (float*)wsum.data_ptr(), n, (int)kind);
+  return {p, loss_sum, wsum};
+}
+
+__global__ void loss_bwd_kernel(const float* __restrict__ p, const float* __restrict__ y,
+                                const float* __restrict__ w, bf16* __restrict__ dz,
+                                long n, int kind, float scale) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < n; j += stride) {
+    float pf = p[j], yf = y[j], wf = w[j];
+    float g = (kind == LOSS_WMSE) ? wf * 2.0f * (pf - yf) * pf * (1.0f - pf)
+                                  : wf * (pf - yf);
+    dz[j] = __float2bfloat16(g * scale);
+  }
+}
+
+at::Tensor weighted_loss_bwd(at::Tensor p, at::Tensor y, at::Tensor w, long kind, double scale) {
+  CHECK_GPU(p); CHECK_CONTIG(p); CHECK_F32(p);
+  long n = p.numel();
+  auto dz = at::empty({n}, p.options().dtype(at::kBFloat16));
+  int blocks = (int)std::min((n + 511) / 512 + 1, (long)1024);
+  hipLaunchKernelGGL(loss_bwd_kernel, dim3(blocks), dim3(512), 0, cur_stream(),
+                     (const float*)p.data_ptr(), (const float*)y.data_ptr(),
+                     (const float*)w.data_ptr(), (bf16*)dz.data_ptr(),
+                     n, (int)kind, (float)scale);
+  return dz;
+}
+
+// ---------------------------------------------------------------------------
+// K4 — fused optimizers over flat fp32 arena (g includes coupled L2 in-kernel)
+// ---------------------------------------------------------------------------
+__global__ void sgd_kernel(float* __restrict__ w, const float* __restrict__ g,
+                           long n, float lr, float l2) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  long n4 = n / 4;
+  for (long v = i; v < n4; v += stride) {
+    f32x4 wv = ((f32x4*)w)[v], gv = ((const f32x4*)g)[v];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) wv[j] -= lr * (gv[j] + l2 * wv[j]);
+    ((f32x4*)w)[v] = wv;
+  }
+  for (long j = n4 * 4 + i; j < n; j += stride) w[j] -= lr * (g[j] + l2 * w[j]);
+}
+
+__global__ void adam_kernel(float* __restrict__ w, const float* __restrict__ g,
+                            float* __restrict__ m, float* __restrict__ v,
+                            long n, float lr, float b1, float b2, float eps,
+                            float l2, float bc1, float sbc2) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  float step = lr * sbc2 / bc1;
+  for (long j = i; j < n; j += stride) {
+    float gj = g[j] + l2 * w[j];
+    float mj = b1 * m[j] + (1.0f - b1) * gj;
+    float vj = b2 * v[j] + (1.0f - b2) * gj * gj;
+    m[j] = mj; v[j] = vj;
+    w[j] -= step * mj / (sqrtf(vj) + eps * sbc2);
+  }
+}
+
+__global__ void adadelta_kernel(float* __restrict__ w, const float* __restrict__ g,
+                                float* __restrict__ acc, float* __restrict__ dacc,
+                                long n, float lr, float rho, float eps, float l2) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < n; j += stride) {
+    float gj = g[j] + l2 * w[j];
+    float a = rho * acc[j] + (1.0f - rho) * gj * gj;
+    acc[j] = a;
+    float upd = gj * sqrtf(dacc[j] + eps) / sqrtf(a + eps);
+    dacc[j] = rho * dacc[j] + (1.0f - rho) * upd * upd;
+    w[j] -= lr * upd;
+  }
+}
+
+__global__ void adagrad_kernel(float* __restrict__ w, const float* __restrict__ g,
+                               float* __restrict__ m, long n, float lr, float eps, float l2) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < n; j += stride) {
+    float gj = g[j] + l2 * w[j];
+    float mj = m[j] + gj * gj;
+    m[j] = mj;
+    w[j] -= lr * gj / (sqrtf(mj) + eps);
+  }
+}
+
+static int opt_blocks(long n) { return (int)std::min((n + 1023) / 1024 + 1, (long)2048); }
+
+void sgd_step(at::Tensor w, at::Tensor g, double lr, double l2) {
+  CHECK_GPU(w); CHECK_F32(w); CHECK_F32(g);
+  hipLaunchKernelGGL(sgd_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(), w.numel(),
+                     (float)lr, (float)l2);
+}
+
+void adam_step(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor v,
+               double lr, double b1, double b2, double eps, double l2, long t) {
+  CHECK_GPU(w); CHECK_F32(w);
+  float bc1 = 1.0f - powf((float)b1, (float)t);
+  float sbc2 = sqrtf(1.0f - powf((float)b2, (float)t));
+  hipLaunchKernelGGL(adam_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(),
+                     (float*)m.data_ptr(), (float*)v.data_ptr(), w.numel(),
+                     (float)lr, (float)b1, (float)b2, (float)eps, (float)l2, bc1, sbc2);
+}
+
+void adadelta_step(at::Tensor w, at::Tensor g, at::Tensor acc, at::Tensor dacc,
+                   double lr, double rho, double eps, double l2) {
+  CHECK_GPU(w); CHECK_F32(w);
+  hipLaunchKernelGGL(adadelta_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(),
+                     (float*)acc.data_ptr(), (float*)dacc.data_ptr(), w.numel(),
+                     (float)lr, (float)rho, (float)eps, (float)l2);
+}
+
+void adagrad_step(at::Tensor w, at::Tensor g, at::Tensor m,
+                  double lr, double eps, double l2) {
+  CHECK_GPU(w); CHECK_F32(w);
+  hipLaunchKernelGGL(adagrad_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(),
+                     (float*)m.data_ptr(), w.numel(), (float)lr, (float)eps, (float)l2);
+}
+
+// ---------------------------------------------------------------------------
+// embedding arena gather: out[b, f*D+d] = arena[ids[b,f], d]  (bf16)
+// one 16B (8 bf16) chunk per thread; out writes coalesced.
+// ---------------------------------------------------------------------------
+__global__ void emb_gather_kernel(const bf16* __restrict__ arena,
+                                  const long* __restrict__ ids,
+                                  bf16* __restrict__ out,
+                                  long rows_bf, long D) {
+  long chunks_per_row = D / 8;
+  long total = rows_bf * chunks_per_row;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    long rf = t / chunks_per_row;      // (b*F + f)
+    long c = t % chunks_per_row;
+    long row = ids[rf];
+    ((s16x8*)out)[t] = *(const s16x8*)(arena + row * D + c * 8);
+  }
+}
+
+__global__ void emb_gather_scalar_kernel(const bf16* __restrict__ arena,
+                                         const long* __restrict__ ids,
+                                         bf16* __restrict__ out,
+                                         long rows_bf, long D) {
+  long total = rows_bf * D;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    long rf = t / D, d = t % D;
+    out[t] = arena[ids[rf] * D + d];
+  }
+}
+
+at::Tensor embedding_gather(at::Tensor arena, at::Tensor ids) {
+  CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
+  CHECK_GPU(ids); CHECK_CONTIG(ids);
+  TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64");
+  long B = ids.size(0), F = ids.size(1), D = arena.size(1);
+  auto out = at::empty({B, F * D}, arena.options());
+  long rows = B * F;
+  if (D % 8 == 0) {
+    long total = rows * (D / 8);
+    int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
+    hipLaunchKernelGGL(emb_gather_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                       (const bf16*)arena.data_ptr(), (const long*)ids.data_ptr(),
+                       (bf16*)out.data_ptr(), rows, D);
+  } else {
+    long total = rows * D;
+    int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
+    hipLaunchKernelGGL(emb_gather_scalar_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                       (const bf16*)arena.data_ptr(), (const long*)ids.data_ptr(),
+                       (bf16*)out.data_ptr(), rows, D);
+  }
+  return out;
+}
+
+// ---------------------------------------------------------------------------
+// rowwise sparse embedding updates (rows unique after coalesce — no atomics)
+// one wave per row; lanes cover D.
+// ---------------------------------------------------------------------------
+__global__ void emb_sgd_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
+                               const float* __restrict__ vals, long nrows, long D, float lr) {
+  long r = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (r >= nrows) return;
+  int lane = threadIdx.x & 63;
+  long row = rows[r];
+  for (long d = lane; d < D; d += 64) {
+    float wv = __bfloat162float(arena[row * D + d]);
+    arena[row * D + d] = __float2bfloat16(wv - lr * vals[r * D + d]);
+  }
+}
+
+__global__ void emb_adagrad_kernel(bf16* __restrict__ arena, float* __restrict__ acc,
+                                   const long* __restrict__ rows, const float* __restrict__ vals,
+                                   long nrows, long D, float lr, float eps) {
+  long r = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (r >= nrows) return;
+  int lane = threadIdx.x & 63;
+  long row = rows[r];
+  // rowwise mean of g^2 (wave reduce over D)
+  float sq = 0.0f;
+  for (long d = lane; d < D; d += 64) {
+    float g = vals[r * D + d];
+    sq += g * g;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) sq += __shfl_down(sq, off, 64);
+  sq = __shfl(sq, 0, 64) / (float)D;
+  float a = acc[row] + sq;
+  if (lane == 0) acc[row] = a;
+  float denom = sqrtf(a) + eps;
+  for (long d = lane; d < D; d += 64) {
+    float wv = __bfloat162float(arena[row * D + d]);
+    arena[row * D + d] = __float2bfloat16(wv - lr * vals[r * D + d] / denom);
+  }
+}
+
+void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
+  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(vals);
+  long nrows = rows.numel(), D = arena.size(1);
+  if (!nrows) return;
+  int waves_per_block = 4;
+  int blocks = (int)((nrows + waves_per_block - 1) / waves_per_block);
+  hipLaunchKernelGGL(emb_sgd_kernel, dim3(blocks), dim3(64 * waves_per_block), 0, cur_stream(),
+                     (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
+                     (const float*)vals.data_ptr(), nrows, D, (float)lr);
+}
+
+void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Tensor vals,
+                      double lr, double eps) {
+  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(acc); CHECK_F32(vals);
+  long nrows = rows.numel(), D = arena.size(1);
+  if (!nrows) return;
+  int waves_per_block = 4;
+  int blocks = (int)((nrows + waves_per_block - 1) / waves_per_block);
+  hipLaunchKernelGGL(emb_adagrad_kernel, dim3(blocks), dim3(64 * waves_per_block), 0, cur_stream(),
+                     (bf16*)arena.data_ptr(), (float*)acc.data_ptr(),
+                     (const long*)rows.data_ptr(), (const float*)vals.data_ptr(),
+                     nrows, D, (float)lr, (float)eps);
+}
+
+// ---------------------------------------------------------------------------
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "shifu_amd CDNA4 (gfx950) kernels";
+  m.def("linear_act_fwd", &linear_act_fwd, "fused GEMM+bias+act forward (bf16)");
+  m.def("gemm_nn_bf16", &gemm_nn_bf16);
+  m.def("gemm_nt_bf16", &gemm_nt_bf16);
+  m.def("gemm_tn_f32", &gemm_tn_f32);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("act_grad", &act_grad);
+  m.def("colsum_f32", &colsum_f32);
+  m.def("weighted_loss_fwd", &weighted_loss_fwd);
+  m.def("weighted_loss_bwd", &weighted_loss_bwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("adam_step", &adam_step);
+  m.def("adadelta_step", &adadelta_step);
+  m.def("adagrad_step", &adagrad_step);
+  m.def("embedding_gather", &embedding_gather);
+  m.def("emb_sgd_step", &emb_sgd_step);
+  m.def("emb_adagrad_step", &emb_adagrad_step);
+}

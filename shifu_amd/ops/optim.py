@@ -101,16 +101,10 @@ class FusedOptimizer:
     def _emb_step(self, p: torch.nn.Parameter, idx: int) -> None:
         if p.grad is None:
             return
-        g = p.grad
-        if g.is_sparse:
-            g = g.coalesce()
-            rows, vals = g.indices()[0], g.values()
-        else:  # dense grad on a small arena (CPU tests)
-            rows = torch.nonzero(g.abs().sum(dim=1) != 0, as_tuple=False).reshape(-1)
-            vals = g[rows]
+        from shifu_amd.ops.embedding import sparse_rows_values
+        rows, vals = sparse_rows_values(p.grad)
         if rows.numel() == 0:
             return
-        vals = vals.float()
         if self.emb_kind == OPT_ADAGRAD:
             acc = self.emb_state[idx]
             if use_hip(p):

@@ -170,14 +170,15 @@ class GradAggregator:
                 self._aggregate_sparse(p, world)
 
     def _aggregate_sparse(self, p: torch.nn.Parameter, world: int) -> None:
-        if p.grad is None or not p.grad.is_sparse:
-            if p.grad is not None:  # dense emb grad (small test arenas)
-                dist.all_reduce(p.grad, op=dist.ReduceOp.SUM)
-                p.grad.div_(world)
+        if p.grad is None:
             return
-        g = p.grad.coalesce()
-        idx = g.indices()[0].contiguous()
-        vals = g.values().contiguous()
+        if not p.grad.is_sparse:  # dense emb grad (small test arenas)
+            dist.all_reduce(p.grad, op=dist.ReduceOp.SUM)
+            p.grad.div_(world)
+            return
+        from shifu_amd.ops.embedding import sparse_rows_values
+        idx, vals = sparse_rows_values(p.grad)
+        idx, vals = idx.contiguous(), vals.contiguous()
         dev = vals.device
         n = torch.tensor([idx.numel()], device=dev, dtype=torch.int64)
         ns = [torch.zeros_like(n) for _ in range(world)]
@@ -195,7 +196,7 @@ class GradAggregator:
         dist.all_gather(idx_out, idx_pad)
         dist.all_gather(val_out, val_pad)
         all_idx = torch.cat([t[:int(c)] for t, c in zip(idx_out, ns)])
-        all_val = torch.cat([t[:int(c)] for t, c in zip(val_out, ns)]) / world
+        all_val = (torch.cat([t[:int(c)] for t, c in zip(val_out, ns)]) / world).to(p.dtype)
         p.grad = torch.sparse_coo_tensor(all_idx.unsqueeze(0), all_val, p.shape)
 
     def remove_hooks(self) -> None:
