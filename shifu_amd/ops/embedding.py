@@ -49,10 +49,11 @@ class _EmbGatherFn(torch.autograd.Function):
     def forward(ctx, arena: torch.Tensor, flat_ids: torch.Tensor, F: int, D: int):
         # flat_ids: [B, F] global row ids (offsets already added)
         B = flat_ids.shape[0]
-        if use_hip(arena):
+        if arena.dtype == torch.bfloat16 and use_hip(arena):
             ext = hip_ops()
             out = ext.embedding_gather(arena, flat_ids.contiguous())
         else:
+            # fp32 arena (explicit rc.dtype="fp32" config) takes the eager path
             out = arena.index_select(0, flat_ids.reshape(-1)).reshape(B, F * D)
         ctx.save_for_backward(flat_ids)
         ctx.arena_shape = arena.shape

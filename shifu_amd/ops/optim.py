@@ -107,7 +107,7 @@ class FusedOptimizer:
             return
         if self.emb_kind == OPT_ADAGRAD:
             acc = self.emb_state[idx]
-            if use_hip(p):
+            if p.dtype == torch.bfloat16 and use_hip(p):
                 hip_ops().emb_adagrad_step(p.data, acc, rows.contiguous(),
                                            vals.contiguous(), self.emb_lr, self.eps)
             else:
@@ -116,7 +116,7 @@ class FusedOptimizer:
                 denom = acc[rows].add(self.eps).sqrt_().unsqueeze(1)
                 p.data.index_add_(0, rows, (-self.emb_lr * vals / denom).to(p.dtype))
         else:  # sgd
-            if use_hip(p):
+            if p.dtype == torch.bfloat16 and use_hip(p):
                 hip_ops().emb_sgd_step(p.data, rows.contiguous(), vals.contiguous(),
                                        self.emb_lr)
             else:

@@ -98,6 +98,12 @@ class Trainer:
         self.model = model.to(self.device)
         dense_dtype = torch.bfloat16 if (self.device.type == "cuda"
                                          and rc.dtype == "bf16") else torch.float32
+        if dense_dtype == torch.bfloat16:
+            # embedding arenas live bf16 in HBM (the HIP gather path);
+            # dense params stay fp32 masters (FusedLinear casts per step)
+            for p in self.model.parameters():
+                if getattr(p, "_is_embedding_arena", False):
+                    p.data = p.data.to(torch.bfloat16)
         self.dense_dtype = dense_dtype
         self.train_data = DeviceData.from_dataset(train_data, self.device, dense_dtype)
         self.valid_data = DeviceData.from_dataset(valid_data, self.device, dense_dtype)
