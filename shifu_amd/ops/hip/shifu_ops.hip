@@ -87,7 +87,10 @@ enum Epi { EPI_PLAIN = 0, EPI_BIAS_ACT = 1, EPI_F32 = 2 };
 #define BK 32
 #define BKP (BK + 8)   // +16B row pad: 20-dword row stride -> conflict-free b128 groups
 
-template <int TA, int TB, int EPI, typename OUT_T>
+// SPLITK: blockIdx.z partitions the K (reduction) range; each part atomically
+// accumulates into a zero-initialized f32 C.  Fills the 256-CU chip for
+// wgrad shapes whose M/N tile grid alone is far below 256 workgroups.
+template <int TA, int TB, int EPI, typename OUT_T, bool SPLITK = false>
 __global__ __launch_bounds__(256)
 void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                       OUT_T* __restrict__ C, const bf16* __restrict__ bias,
@@ -112,7 +115,16 @@ void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   const int r16 = lane & 15;        // fragment row/col within 16
   const int kgrp = lane >> 4;       // 0..3 -> k-offset group *8
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
+  int k_lo = 0, k_hi = K;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int chunk = ((K + nz - 1) / nz + BK - 1) / BK * BK;   // BK-aligned split
+    k_lo = blockIdx.z * chunk;
+    k_hi = min(K, k_lo + chunk);
+    if (k_lo >= K) return;
+  }
+
+  for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
     // ---- stage A tile: As[m][k] = A_eff[m0+m][k0+k] ----
     if (TA == 0) {
       // A[M,K]: read 8 bf16 along k per thread (BM*(BK/8)=512 chunks, 2/thread)
@@ -237,7 +249,8 @@ void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
         float v = acc[fi][fj][r];
         if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
         if (EPI == EPI_F32) {
-          ((float*)C)[(long)row * N + col] = v;
+          if (SPLITK) atomicAdd(&((float*)C)[(long)row * N + col], v);
+          else ((float*)C)[(long)row * N + col] = v;
         } else {
           ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
         }
@@ -250,8 +263,25 @@ template <int TA, int TB, int EPI, typename OUT_T>
 static void launch_gemm(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
                         long M, long N, long K, int act, hipStream_t s) {
   dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
-  hipLaunchKernelGGL((gemm_tile_kernel<TA, TB, EPI, OUT_T>), grid, dim3(256), 0, s,
+  hipLaunchKernelGGL((gemm_tile_kernel<TA, TB, EPI, OUT_T, false>), grid, dim3(256), 0, s,
                      A, B, C, bias, (int)M, (int)N, (int)K, act);
+}
+
+// split-K variant (EPI_F32 wgrad): pick Z so the grid fills 256 CUs (>=512
+// workgroups), each K-part >= one BK tile.  C must be zero-initialized.
+template <int TA, int TB>
+static void launch_gemm_splitk(const bf16* A, const bf16* B, float* C,
+                               long M, long N, long K, hipStream_t s) {
+  long gx = (N + BN - 1) / BN, gy = (M + BM - 1) / BM;
+  long max_z = (K + BK - 1) / BK;
+  long z = std::min<long>(std::max<long>(512 / std::max<long>(gx * gy, 1), 1), max_z);
+  if (z <= 1) {
+    launch_gemm<TA, TB, EPI_F32, float>(A, B, C, nullptr, M, N, K, 0, s);
+    return;
+  }
+  dim3 grid((unsigned)gx, (unsigned)gy, (unsigned)z);
+  hipLaunchKernelGGL((gemm_tile_kernel<TA, TB, EPI_F32, float, true>), grid, dim3(256), 0, s,
+                     A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
 }
 
 // ------------------------------ GEMM entry points ---------------------------
@@ -298,10 +328,10 @@ at::Tensor gemm_tn_f32(at::Tensor x, at::Tensor dz) {
   CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
   long Bb = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == Bb, "shape mismatch x^T@dz");
-  auto dw = at::empty({K, N}, x.options().dtype(at::kFloat));
+  auto dw = at::zeros({K, N}, x.options().dtype(at::kFloat));
   // C[M=K, N'=N] = (x^T)[M=K, Kr=B] * dz[Kr=B, N]; x stored [B,K] => TA=1
-  launch_gemm<1, 0, EPI_F32, float>((const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
-                                    (float*)dw.data_ptr(), nullptr, K, N, Bb, 0, cur_stream());
+  launch_gemm_splitk<1, 0>((const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
+                           (float*)dw.data_ptr(), K, N, Bb, cur_stream());
   return dw;
 }
 
@@ -374,22 +404,74 @@ at::Tensor act_grad(at::Tensor dy, at::Tensor y, long act) {
 // colsum: db[N] = sum_b dz[b][n]  (f32 out)
 // thread t of block bx owns column bx*256+t; row reads are coalesced.
 // ---------------------------------------------------------------------------
+// 2D grid: blockIdx.y picks a row chunk; partials atomicAdd'd — fills the
+// 256-CU chip for any (B,N) (the 1D version left 255/256 CUs idle).
 __global__ void colsum_kernel(const bf16* __restrict__ dz, float* __restrict__ db,
-                              long B, long N) {
+                              long B, long N, long rows_per_chunk) {
   long n = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
+  long b0 = (long)blockIdx.y * rows_per_chunk;
+  long b1 = min(B, b0 + rows_per_chunk);
   float acc = 0.0f;
-  for (long b = 0; b < B; ++b) acc += __bfloat162float(dz[b * N + n]);
-  db[n] = acc;
+  for (long b = b0; b < b1; ++b) acc += __bfloat162float(dz[b * N + n]);
+  if (gridDim.y == 1) db[n] = acc;
+  else atomicAdd(&db[n], acc);
+}
+
+static void colsum_grid(long B, long N, long& gx, long& chunks, long& rows_per_chunk) {
+  gx = (N + 255) / 256;
+  chunks = std::min<long>(std::max<long>(1024 / std::max<long>(gx, 1), 1),
+                          std::max<long>(B / 64, 1));
+  rows_per_chunk = (B + chunks - 1) / chunks;
 }
 
 at::Tensor colsum_f32(at::Tensor dz) {
   CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
   long B = dz.size(0), N = dz.size(1);
-  auto db = at::empty({N}, dz.options().dtype(at::kFloat));
-  hipLaunchKernelGGL(colsum_kernel, dim3((N + 255) / 256), dim3(256), 0, cur_stream(),
-                     (const bf16*)dz.data_ptr(), (float*)db.data_ptr(), B, N);
+  auto db = at::zeros({N}, dz.options().dtype(at::kFloat));
+  long gx, chunks, rpc;
+  colsum_grid(B, N, gx, chunks, rpc);
+  hipLaunchKernelGGL(colsum_kernel, dim3((unsigned)gx, (unsigned)chunks), dim3(256),
+                     0, cur_stream(),
+                     (const bf16*)dz.data_ptr(), (float*)db.data_ptr(), B, N, rpc);
   return db;
+}
+
+// fused act-grad + bias-grad: dz = dy*act'(y) AND db = colsum(dz) in ONE pass
+// over [B,N] — saves the full dz re-read a separate colsum would cost.
+__global__ void act_grad_colsum_kernel(const bf16* __restrict__ dy,
+                                       const bf16* __restrict__ y,
+                                       bf16* __restrict__ dz, float* __restrict__ db,
+                                       long B, long N, long rows_per_chunk, int act) {
+  long n = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (n >= N) return;
+  long b0 = (long)blockIdx.y * rows_per_chunk;
+  long b1 = min(B, b0 + rows_per_chunk);
+  float acc = 0.0f;
+  for (long b = b0; b < b1; ++b) {
+    long i = b * N + n;
+    float g = __bfloat162float(dy[i]) * act_grad_from_y(__bfloat162float(y[i]), act);
+    dz[i] = __float2bfloat16(g);
+    acc += g;
+  }
+  if (gridDim.y == 1) db[n] = acc;
+  else atomicAdd(&db[n], acc);
+}
+
+std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+  long B = dy.size(0), N = dy.size(1);
+  auto dz = at::empty_like(dy);
+  auto db = at::zeros({N}, dy.options().dtype(at::kFloat));
+  long gx, chunks, rpc;
+  colsum_grid(B, N, gx, chunks, rpc);
+  hipLaunchKernelGGL(act_grad_colsum_kernel, dim3((unsigned)gx, (unsigned)chunks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                     (bf16*)dz.data_ptr(), (float*)db.data_ptr(),
+                     B, N, rpc, (int)act);
+  return {dz, db};
 }
 
 // ---------------------------------------------------------------------------
@@ -710,6 +792,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);
+  m.def("act_grad_colsum", &act_grad_colsum);
   m.def("weighted_loss_fwd", &weighted_loss_fwd);
   m.def("weighted_loss_bwd", &weighted_loss_bwd);
   m.def("sgd_step", &sgd_step);

@@ -86,9 +86,9 @@ class _FusedLinearFn(torch.autograd.Function):
         act = ctx.act
         if ctx.hip:
             ext = hip_ops()
-            dz = ext.act_grad(dy.contiguous(), y, act)          # bf16
-            dw = ext.gemm_tn_f32(x, dz)                         # fp32 [K,N] = x^T @ dz
-            db = ext.colsum_f32(dz)                             # fp32 [N]
+            # one fused pass: dz = dy*act'(y) and db = colsum(dz)
+            dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
+            dw = ext.gemm_tn_f32(x, dz)                         # fp32 [K,N] = x^T @ dz (split-K)
             dx = ext.gemm_nt_bf16(dz, w) if ctx.x_needs_grad else None  # bf16 [B,K] = dz @ w^T
         else:
             dz = _act_grad_from_y_ref(dy, y, act)

@@ -84,6 +84,27 @@ def test_gemm_tn():
     assert ok, f"gemm_tn maxdiff={err}"
 
 
+def test_gemm_tn_splitk_large_batch():
+    """wgrad shape from the bench (B=8192 reduction) — exercises the split-K
+    atomic path."""
+    x = _rand_bf16(8192, 130, seed=30, scale=0.5)
+    dz = _rand_bf16(8192, 96, seed=31, scale=0.5)
+    dw = hip_ops().gemm_tn_f32(x, dz)
+    ok, err = _rel_close(dw, x.float().t() @ dz.float())
+    assert ok, f"gemm_tn splitk maxdiff={err}"
+
+
+def test_act_grad_colsum_fused():
+    dy = _rand_bf16(1000, 96, seed=40)
+    y = torch.sigmoid(_rand_bf16(1000, 96, seed=41).float()).to(torch.bfloat16)
+    dz, db = hip_ops().act_grad_colsum(dy, y, 1)
+    ref_dz = dy.float() * y.float() * (1 - y.float())
+    ok, err = _rel_close(dz, ref_dz)
+    assert ok, f"fused act_grad maxdiff={err}"
+    ok, err = _rel_close(db, dz.float().sum(0), 1e-3)
+    assert ok, f"fused colsum maxdiff={err}"
+
+
 def test_act_grad_and_colsum():
     dy = _rand_bf16(64, 33, seed=12)
     y = torch.sigmoid(_rand_bf16(64, 33, seed=13).float()).to(torch.bfloat16)
