@@ -718,68 +718,124 @@ at::Tensor embedding_gather(at::Tensor arena, at::Tensor ids) {
 }
 
 // ---------------------------------------------------------------------------
-// rowwise sparse embedding updates (rows unique after coalesce — no atomics)
-// one wave per row; lanes cover D.
+// rowwise sparse embedding updates — SORT-FREE: rows may contain DUPLICATES
+// (no torch.unique / rocprim sort on the hot path); correctness comes from
+// atomics: global_atomic_pk_add_bf16 for the arena (unsafeAtomicAdd on
+// __hip_bfloat162) and f32 atomicAdd for the adagrad accumulator.  Updates
+// are linear in the gradient, so per-duplicate application == summed-grad
+// application for a fixed denominator.
 // ---------------------------------------------------------------------------
-__global__ void emb_sgd_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
-                               const float* __restrict__ vals, long nrows, long D, float lr) {
-  long r = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  if (r >= nrows) return;
-  int lane = threadIdx.x & 63;
-  long row = rows[r];
-  for (long d = lane; d < D; d += 64) {
-    float wv = __bfloat162float(arena[row * D + d]);
-    arena[row * D + d] = __float2bfloat16(wv - lr * vals[r * D + d]);
+DEVINL void atomic_add_bf16_scalar(bf16* p, float v) {
+  // 16-bit add via CAS on the containing aligned 32-bit word (odd-D tail)
+  unsigned long addr = (unsigned long)p;
+  unsigned* word = (unsigned*)(addr & ~3UL);
+  int hi = (addr & 2) != 0;
+  unsigned old = *word, assumed;
+  do {
+    assumed = old;
+    unsigned short bits = hi ? (assumed >> 16) : (assumed & 0xffff);
+    bf16 cur = *(bf16*)&bits;
+    float nf = __bfloat162float(cur) + v;
+    bf16 nb = __float2bfloat16(nf);
+    unsigned short nbits = *(unsigned short*)&nb;
+    unsigned repl = hi ? ((assumed & 0x0000ffffu) | ((unsigned)nbits << 16))
+                       : ((assumed & 0xffff0000u) | nbits);
+    old = atomicCAS(word, assumed, repl);
+  } while (old != assumed);
+}
+
+// arena[rows[i]] += scale * vals[i] * rowscale[i] (rowscale nullable)
+__global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
+                                   const float* __restrict__ vals,
+                                   const float* __restrict__ rowscale,
+                                   long n, long D, float scale) {
+  long pairs = D / 2;
+  long total = n * (pairs ? pairs : 1);
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  if (pairs) {
+    for (long t = i; t < total; t += stride) {
+      long e = t / pairs, dp = t % pairs;
+      float sc = scale * (rowscale ? rowscale[e] : 1.0f);
+      long row = rows[e];
+      __hip_bfloat162 add;
+      add.x = __float2bfloat16(sc * vals[e * D + dp * 2]);
+      add.y = __float2bfloat16(sc * vals[e * D + dp * 2 + 1]);
+      unsafeAtomicAdd((__hip_bfloat162*)(arena + row * D + dp * 2), add);
+    }
+    if (D & 1) {  // odd tail element
+      for (long e = i; e < n; e += stride) {
+        float sc = scale * (rowscale ? rowscale[e] : 1.0f);
+        atomic_add_bf16_scalar(arena + rows[e] * D + D - 1, sc * vals[e * D + D - 1]);
+      }
+    }
+  } else {  // D == 1
+    for (long e = i; e < n; e += stride) {
+      float sc = scale * (rowscale ? rowscale[e] : 1.0f);
+      atomic_add_bf16_scalar(arena + rows[e], sc * vals[e]);
+    }
   }
 }
 
-__global__ void emb_adagrad_kernel(bf16* __restrict__ arena, float* __restrict__ acc,
-                                   const long* __restrict__ rows, const float* __restrict__ vals,
-                                   long nrows, long D, float lr, float eps) {
-  long r = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
-  if (r >= nrows) return;
+// phase 1 of adagrad: acc[rows[i]] += mean_d vals[i,d]^2  (wave per entry)
+__global__ void emb_accsq_kernel(float* __restrict__ acc, const long* __restrict__ rows,
+                                 const float* __restrict__ vals, long n, long D) {
+  long e = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (e >= n) return;
   int lane = threadIdx.x & 63;
-  long row = rows[r];
-  // rowwise mean of g^2 (wave reduce over D)
   float sq = 0.0f;
   for (long d = lane; d < D; d += 64) {
-    float g = vals[r * D + d];
+    float g = vals[e * D + d];
     sq += g * g;
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) sq += __shfl_down(sq, off, 64);
-  sq = __shfl(sq, 0, 64) / (float)D;
-  float a = acc[row] + sq;
-  if (lane == 0) acc[row] = a;
-  float denom = sqrtf(a) + eps;
-  for (long d = lane; d < D; d += 64) {
-    float wv = __bfloat162float(arena[row * D + d]);
-    arena[row * D + d] = __float2bfloat16(wv - lr * vals[r * D + d] / denom);
-  }
+  if (lane == 0) atomicAdd(&acc[rows[e]], sq / (float)D);
+}
+
+// phase 2 denominator: rowscale[i] = 1/(sqrt(acc[rows[i]]) + eps)
+__global__ void emb_denom_kernel(const float* __restrict__ acc, const long* __restrict__ rows,
+                                 float* __restrict__ rowscale, long n, float eps) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long e = i; e < n; e += stride)
+    rowscale[e] = 1.0f / (sqrtf(acc[rows[e]]) + eps);
+}
+
+static int scat_blocks(long total) {
+  return (int)std::min((total + 255) / 256 + 1, (long)4096);
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
   CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(vals);
-  long nrows = rows.numel(), D = arena.size(1);
-  if (!nrows) return;
-  int waves_per_block = 4;
-  int blocks = (int)((nrows + waves_per_block - 1) / waves_per_block);
-  hipLaunchKernelGGL(emb_sgd_kernel, dim3(blocks), dim3(64 * waves_per_block), 0, cur_stream(),
+  long n = rows.numel(), D = arena.size(1);
+  if (!n) return;
+  long total = n * std::max<long>(D / 2, 1);
+  hipLaunchKernelGGL(emb_scatter_kernel, dim3(scat_blocks(total)), dim3(256), 0, cur_stream(),
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
-                     (const float*)vals.data_ptr(), nrows, D, (float)lr);
+                     (const float*)vals.data_ptr(), nullptr, n, D, (float)-lr);
 }
 
 void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Tensor vals,
                       double lr, double eps) {
   CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(acc); CHECK_F32(vals);
-  long nrows = rows.numel(), D = arena.size(1);
-  if (!nrows) return;
-  int waves_per_block = 4;
-  int blocks = (int)((nrows + waves_per_block - 1) / waves_per_block);
-  hipLaunchKernelGGL(emb_adagrad_kernel, dim3(blocks), dim3(64 * waves_per_block), 0, cur_stream(),
-                     (bf16*)arena.data_ptr(), (float*)acc.data_ptr(),
-                     (const long*)rows.data_ptr(), (const float*)vals.data_ptr(),
-                     nrows, D, (float)lr, (float)eps);
+  long n = rows.numel(), D = arena.size(1);
+  if (!n) return;
+  auto s = cur_stream();
+  int wpb = 4;
+  hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + wpb - 1) / wpb)),
+                     dim3(64 * wpb), 0, s,
+                     (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                     (const float*)vals.data_ptr(), n, D);
+  auto rowscale = at::empty({n}, vals.options());
+  hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
+                     (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                     (float*)rowscale.data_ptr(), n, (float)eps);
+  long total = n * std::max<long>(D / 2, 1);
+  hipLaunchKernelGGL(emb_scatter_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
+                     (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
+                     (const float*)vals.data_ptr(), (const float*)rowscale.data_ptr(),
+                     n, D, (float)-lr);
 }
 
 // ---------------------------------------------------------------------------
