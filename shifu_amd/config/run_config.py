@@ -30,8 +30,13 @@ class RunConfig:
     weight_column: int = -1
     selected_numeric_columns: List[int] = field(default_factory=list)
     selected_categorical_columns: List[int] = field(default_factory=list)
+    vocab_sizes: List[int] = field(default_factory=list)  # per categorical column
     valid_set_rate: float = 0.2
     seed: int = 1234
+
+    # -- model family --
+    model_type: str = "auto"   # "mlp" | "wide_deep" | "deepfm" | "auto"
+    embed_dim: int = 16
 
     # -- model / training --
     model_config_path: Optional[str] = None
@@ -90,3 +95,10 @@ class RunConfig:
         self.weight_column = cc.weight_column
         self.selected_numeric_columns = cc.selected_numeric_columns
         self.selected_categorical_columns = cc.selected_categorical_columns
+        vs = cc.vocab_sizes()
+        self.vocab_sizes = [vs[c] for c in self.selected_categorical_columns]
+
+    def resolved_model_type(self) -> str:
+        if self.model_type != "auto":
+            return self.model_type
+        return "wide_deep" if self.selected_categorical_columns else "mlp"

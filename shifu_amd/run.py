@@ -45,12 +45,13 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
                               if rc.resolved_device() == "cuda" else 0) \
             if rc.resolved_device() == "cuda" else torch.device("cpu")
 
-        vocab = [1000] * len(rc.selected_categorical_columns)
+        vocab = rc.vocab_sizes or [1000] * len(rc.selected_categorical_columns)
         model = build_model(mc, len(rc.selected_numeric_columns), vocab,
-                            model_type="wide_deep" if vocab else "mlp",
-                            seed=rc.seed)
+                            model_type=rc.resolved_model_type(),
+                            embed_dim=rc.embed_dim, seed=rc.seed)
         trainer = Trainer(model, mc, rc, train, valid, rank=rank,
-                          world_size=world, device=device, metric_sink=metric_sink)
+                          world_size=world, device=device, metric_sink=metric_sink,
+                          heartbeat=heartbeat)
         trainer.fit()
     finally:
         destroy_distributed()

@@ -88,12 +88,19 @@ class Trainer:
                  train_data: TabularDataset, valid_data: TabularDataset,
                  rank: int = 0, world_size: int = 1,
                  device: Optional[torch.device] = None,
-                 metric_sink: Optional[Callable[[TrainingIntermediateResult], None]] = None):
+                 metric_sink: Optional[Callable[[TrainingIntermediateResult], None]] = None,
+                 heartbeat: Optional[Callable[[], None]] = None):
         self.mc, self.rc = mc, rc
         self.rank, self.world = rank, world_size
         self.device = device or torch.device(rc.resolved_device())
         self.is_chief = (rank == 0)  # chief semantics (TensorflowSession.java:443-450)
         self.metric_sink = metric_sink
+        # liveness: successor of the executor heartbeat loop
+        # (TensorflowApplicationMaster.java:63-112); called at least every
+        # heartbeat_interval_s from inside the epoch step loop
+        self.heartbeat = heartbeat
+        self._hb_interval = float(rc.heartbeat_interval_s)
+        self._hb_last = time.time()
 
         self.model = model.to(self.device)
         dense_dtype = torch.bfloat16 if (self.device.type == "cuda"
@@ -182,6 +189,9 @@ class Trainer:
             # window mode: only every update_window-th (or last) step syncs+updates
             sync = ((si + 1) % self.update_window == 0) or (si == n_steps - 1)
             losses.append(self.train_step(batch, sync=sync))
+            if self.heartbeat and time.time() - self._hb_last >= self._hb_interval:
+                self.heartbeat()
+                self._hb_last = time.time()
         if self.device.type == "cuda":
             torch.cuda.synchronize()
         train_time = time.time() - t0

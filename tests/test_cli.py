@@ -1,0 +1,67 @@
+"""End-to-end CLI test: ColumnConfig-driven run through shifu_amd.run.main
+(the `shifu train` successor path)."""
+import json
+import os
+
+import pytest
+
+from shifu_amd.data.synthetic import generate_synthetic_csv
+
+
+def test_run_main_end_to_end(tmp_path):
+    from shifu_amd.run import main
+    data_dir = str(tmp_path / "data")
+    generate_synthetic_csv(data_dir, n_rows=600, n_dense=4, vocab_sizes=[20, 30],
+                           n_files=2, seed=9)
+
+    # ColumnConfig describing the generated layout: target=0, weight=1,
+    # dense 2..5, categorical 6..7
+    cc = [{"columnNum": 0, "columnName": "target", "columnFlag": "Target"},
+          {"columnNum": 1, "columnName": "w", "columnFlag": "Weight"}]
+    cc += [{"columnNum": i, "columnName": f"d{i}", "finalSelect": True,
+            "columnType": "N"} for i in range(2, 6)]
+    cc += [{"columnNum": 6, "columnName": "c0", "finalSelect": True,
+            "columnType": "C", "vocabSize": 20},
+           {"columnNum": 7, "columnName": "c1", "finalSelect": True,
+            "columnType": "C", "vocabSize": 30}]
+    cc_path = str(tmp_path / "ColumnConfig.json")
+    with open(cc_path, "w") as f:
+        json.dump(cc, f)
+
+    mc = {"train": {"numTrainEpochs": 2, "validSetRate": 0.2,
+                    "params": {"NumHiddenLayers": 2, "NumHiddenNodes": [16, 8],
+                               "ActivationFunc": ["relu", "relu"],
+                               "LearningRate": 0.02, "Optimizer": "adam",
+                               "Loss": "sigmoid_ce", "MiniBatchSize": 64,
+                               "L2Reg": 0.0}}}
+    mc_path = str(tmp_path / "ModelConfig.json")
+    with open(mc_path, "w") as f:
+        json.dump(mc, f)
+
+    run_cfg = {
+        "num_gpus": 2, "backend": "gloo", "master_port": 29731,
+        "training_data_path": [data_dir],
+        "tmp_model_path": str(tmp_path / "ckpt"),
+        "final_model_path": str(tmp_path / "final"),
+        "log_dir": str(tmp_path / "logs"),
+        "model_type": "wide_deep", "embed_dim": 4,
+        "device": "cpu",
+    }
+    rc_path = str(tmp_path / "run.json")
+    with open(rc_path, "w") as f:
+        json.dump(run_cfg, f)
+
+    rcode = main(["--run-config", rc_path, "--model-config", mc_path,
+                  "--column-config", cc_path])
+    assert rcode == 0
+    # exported artifacts
+    final = tmp_path / "final"
+    assert (final / "GenericModelConfig.json").exists()
+    gmc = json.loads((final / "GenericModelConfig.json").read_text())
+    assert gmc["outputnames"] == ["shifu_output_0"]
+    graph = json.loads((final / "graph.json").read_text())
+    assert graph["family"] == "wide_deep"
+    assert graph["vocab_sizes"] == [20, 30]
+    # progress board written
+    board = (tmp_path / "logs" / "progress.board").read_text()
+    assert "epoch 0:" in board and "epoch 1:" in board
