@@ -84,8 +84,9 @@ enum Epi { EPI_PLAIN = 0, EPI_BIAS_ACT = 1, EPI_F32 = 2 };
 
 #define BM 128
 #define BN 128
-#define BK 32
-#define BKP (BK + 8)   // +16B row pad: 20-dword row stride -> conflict-free b128 groups
+#define BK 64
+#define BKP (BK + 8)   // +16B row pad: 36-dword row stride; 16-lane b128 groups
+                       // hit 16 distinct banks (gcd(36,64)=4, r<16 all distinct)
 
 // SPLITK: blockIdx.z partitions the K (reduction) range; each part atomically
 // accumulates into a zero-initialized f32 C.  Fills the 256-CU chip for
@@ -127,107 +128,95 @@ void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   for (int k0 = k_lo; k0 < k_hi; k0 += BK) {
     // ---- stage A tile: As[m][k] = A_eff[m0+m][k0+k] ----
     if (TA == 0) {
-      // A[M,K]: read 8 bf16 along k per thread (BM*(BK/8)=512 chunks, 2/thread)
+      // A[M,K]: 16B loads along k; 4 chunks/thread (BM*BK/8 = 1024 chunks)
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < 4; ++it) {
         int i = tid + it * 256;
-        int m = i >> 2;                 // /(BK/8)
-        int kc = (i & 3) * 8;
+        int m = i >> 3;                 // /(BK/8)
+        int kc = (i & 7) * 8;
         int gm = m0 + m, gk = k0 + kc;
-        s16x4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+        s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gm < M) {
           const bf16* src = A + (long)gm * K + gk;
-          if (gk + 8 <= K) {
-            v0 = *(const s16x4*)(src);
-            v1 = *(const s16x4*)(src + 4);
-          } else {
-            for (int j = 0; j < 8 && gk + j < K; ++j)
-              ((short*)(j < 4 ? &v0 : &v1))[j & 3] = ((const short*)src)[j];
-          }
+          if (gk + 8 <= K) v = *(const s16x8*)(src);
+          else for (int j = 0; j < 8; ++j)
+            ((short*)&v)[j] = (gk + j < K) ? ((const short*)src)[j] : (short)0;
         }
-        *(s16x4*)&As[m][kc] = v0;
-        *(s16x4*)&As[m][kc + 4] = v1;
+        *(s16x8*)&As[m][kc] = v;
       }
     } else {
-      // A[K,M]: read 8 bf16 along m (coalesced), scatter-transpose into LDS
+      // A[K,M]: 16B loads along m (coalesced), scatter-transpose into LDS
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < 4; ++it) {
         int i = tid + it * 256;
         int k = i >> 4;                 // /(BM/8)
         int mc = (i & 15) * 8;
         int gk = k0 + k, gm = m0 + mc;
-        short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gk < K) {
           const bf16* src = A + (long)gk * M + gm;
-          if (gm + 8 <= M) {
-            *(s16x4*)&tmp[0] = *(const s16x4*)(src);
-            *(s16x4*)&tmp[4] = *(const s16x4*)(src + 4);
-          } else {
-            for (int j = 0; j < 8 && gm + j < M; ++j) tmp[j] = ((const short*)src)[j];
-          }
+          if (gm + 8 <= M) v = *(const s16x8*)(src);
+          else for (int j = 0; j < 8; ++j)
+            ((short*)&v)[j] = (gm + j < M) ? ((const short*)src)[j] : (short)0;
         }
 #pragma unroll
-        for (int j = 0; j < 8; ++j) ((short*)&As[mc + j][k])[0] = tmp[j];
+        for (int j = 0; j < 8; ++j) ((short*)&As[mc + j][k])[0] = ((short*)&v)[j];
       }
     }
 
     // ---- stage B tile: Bs[n][k] = B_eff[k0+k][n0+n] ----
     if (TB == 0) {
-      // B[K,N]: read 8 bf16 along n (coalesced), scatter-transpose
+      // B[K,N]: 16B loads along n (coalesced), scatter-transpose
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < 4; ++it) {
         int i = tid + it * 256;
         int k = i >> 4;
         int nc = (i & 15) * 8;
         int gk = k0 + k, gn = n0 + nc;
-        short tmp[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+        s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gk < K) {
           const bf16* src = B + (long)gk * N + gn;
-          if (gn + 8 <= N) {
-            *(s16x4*)&tmp[0] = *(const s16x4*)(src);
-            *(s16x4*)&tmp[4] = *(const s16x4*)(src + 4);
-          } else {
-            for (int j = 0; j < 8 && gn + j < N; ++j) tmp[j] = ((const short*)src)[j];
-          }
+          if (gn + 8 <= N) v = *(const s16x8*)(src);
+          else for (int j = 0; j < 8; ++j)
+            ((short*)&v)[j] = (gn + j < N) ? ((const short*)src)[j] : (short)0;
         }
 #pragma unroll
-        for (int j = 0; j < 8; ++j) ((short*)&Bs[nc + j][k])[0] = tmp[j];
+        for (int j = 0; j < 8; ++j) ((short*)&Bs[nc + j][k])[0] = ((short*)&v)[j];
       }
     } else {
-      // B[N,K]: rows are k-contiguous -> vector LDS writes, no transpose
+      // B[N,K]: rows are k-contiguous -> 16B vector LDS writes, no transpose
 #pragma unroll
-      for (int it = 0; it < 2; ++it) {
+      for (int it = 0; it < 4; ++it) {
         int i = tid + it * 256;
-        int n = i >> 2;
-        int kc = (i & 3) * 8;
+        int n = i >> 3;
+        int kc = (i & 7) * 8;
         int gn = n0 + n, gk = k0 + kc;
-        s16x4 v0 = {0, 0, 0, 0}, v1 = {0, 0, 0, 0};
+        s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
         if (gn < N) {
           const bf16* src = B + (long)gn * K + gk;
-          if (gk + 8 <= K) {
-            v0 = *(const s16x4*)(src);
-            v1 = *(const s16x4*)(src + 4);
-          } else {
-            for (int j = 0; j < 8 && gk + j < K; ++j)
-              ((short*)(j < 4 ? &v0 : &v1))[j & 3] = ((const short*)src)[j];
-          }
+          if (gk + 8 <= K) v = *(const s16x8*)(src);
+          else for (int j = 0; j < 8; ++j)
+            ((short*)&v)[j] = (gk + j < K) ? ((const short*)src)[j] : (short)0;
         }
-        *(s16x4*)&Bs[n][kc] = v0;
-        *(s16x4*)&Bs[n][kc + 4] = v1;
+        *(s16x8*)&Bs[n][kc] = v;
       }
     }
 
     __syncthreads();
 
-    // ---- MFMA: 2 k-steps of 32 per k-chunk wait BK=32 -> 1 step ----
+    // ---- MFMA: 2 k-steps of K=32 per staged BK=64 tile ----
 #pragma unroll
-    for (int fi = 0; fi < 4; ++fi) {
-      bf16x8 a_frag = *(const bf16x8*)&As[wr + fi * 16 + r16][kgrp * 8];
+    for (int ks = 0; ks < 2; ++ks) {
+      const int ko = ks * 32 + kgrp * 8;
 #pragma unroll
-      for (int fj = 0; fj < 4; ++fj) {
-        bf16x8 b_frag = *(const bf16x8*)&Bs[wc + fj * 16 + r16][kgrp * 8];
-        acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            a_frag, b_frag, acc[fi][fj], 0, 0, 0);
+      for (int fi = 0; fi < 4; ++fi) {
+        bf16x8 a_frag = *(const bf16x8*)&As[wr + fi * 16 + r16][ko];
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          bf16x8 b_frag = *(const bf16x8*)&Bs[wc + fj * 16 + r16][ko];
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[fi][fj], 0, 0, 0);
+        }
       }
     }
     __syncthreads();
