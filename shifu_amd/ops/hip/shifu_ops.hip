@@ -324,6 +324,253 @@ at::Tensor gemm_tn_f32(at::Tensor x, at::Tensor dz) {
   return dw;
 }
 
+// ---------------------------------------------------------------------------
+// v3 "NT" GEMM — THE fast path: both operands stored reduction-major.
+//   C[M,N] = A[M,K] * B[N,K]^T   (bf16 in, f32 accumulate)
+// Every hot GEMM routes here (fwd with weights stored [out,in]; dgrad with a
+// per-step weight transpose; wgrad on transposed activations), so no
+// scatter-transpose staging exists on the hot path.  Staging is
+// global_load_lds (16 B/lane direct-to-LDS DMA, no VGPR round trip) into a
+// LINEAR LDS image with the XOR swizzle applied on the per-lane SOURCE
+// address and on the fragment reads (guide §5.4 rule 21); partial tiles fall
+// back to a guarded ds_write path producing the identical swizzled image.
+// Swizzle: granule(16B) index g at row r lives at g ^ (r&7) -> b128 fragment
+// reads are <=2-way (rows r, r+8 share a bank), vs up-to-8-way unswizzled.
+// ---------------------------------------------------------------------------
+#define NT_BM 128
+#define NT_BN 128
+#define NT_BK 64
+#define NT_LDS_BYTES (2 * NT_BM * NT_BK * 2)
+
+typedef const __attribute__((address_space(1))) unsigned int* gas_ptr;
+typedef __attribute__((address_space(3))) unsigned int* las_ptr;
+
+// stage a 128x64 bf16 tile of P[RM rows, K cols] at (r0, k0) into lds
+// (swizzled image).  Full tiles: 4 glds per wave; edges: guarded ds_writes.
+DEVINL void nt_stage(const bf16* __restrict__ P, bf16* lds, int r0, int k0,
+                     int RM, int K, int tid) {
+  const int lane = tid & 63, wave = tid >> 6;
+  if (r0 + NT_BM <= RM && k0 + NT_BK <= K) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int r = wave * 32 + c * 8 + (lane >> 3);
+      int gsrc = (lane & 7) ^ (r & 7);
+      const bf16* src = P + (long)(r0 + r) * K + k0 + gsrc * 8;
+      bf16* dst = lds + (wave * 32 + c * 8) * NT_BK;   // wave-uniform; +lane*16B implicit
+      __builtin_amdgcn_global_load_lds((gas_ptr)src, (las_ptr)dst, 16, 0, 0);
+    }
+  } else {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int i = tid + it * 256;
+      int r = i >> 3, g = i & 7;
+      int gr = r0 + r, gk = k0 + g * 8;
+      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      if (gr < RM) {
+        const bf16* src = P + (long)gr * K + gk;
+        if (gk + 8 <= K) v = *(const s16x8*)(src);
+        else for (int j = 0; j < 8; ++j)
+          ((short*)&v)[j] = (gk + j < K) ? ((const short*)src)[j] : (short)0;
+      }
+      *(s16x8*)(lds + r * NT_BK + ((g ^ (r & 7)) * 8)) = v;
+    }
+  }
+}
+
+template <int EPI, typename OUT_T, bool SPLITK = false>
+__global__ __launch_bounds__(256)
+void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                    OUT_T* __restrict__ C, const bf16* __restrict__ bias,
+                    int M, int N, int K, int act) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* As = (bf16*)smem;                          // [128][64] swizzled
+  bf16* Bs = (bf16*)(smem + NT_BM * NT_BK * 2);    // [128][64] swizzled
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int m0 = blockIdx.y * NT_BM;
+  const int n0 = blockIdx.x * NT_BN;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int r16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int rx = r16 & 7;            // row&7 for every fragment row this lane reads
+
+  int k_lo = 0, k_hi = K;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int chunk = ((K + nz - 1) / nz + NT_BK - 1) / NT_BK * NT_BK;
+    k_lo = blockIdx.z * chunk;
+    k_hi = min(K, k_lo + chunk);
+    if (k_lo >= K) return;
+  }
+
+  for (int k0 = k_lo; k0 < k_hi; k0 += NT_BK) {
+    nt_stage(A, As, m0, k0, M, K, tid);
+    nt_stage(B, Bs, n0, k0, N, K, tid);
+    __syncthreads();   // hipcc emits the vmcnt(0) drain for in-flight glds here
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int g = ks * 4 + kgrp;           // 16B granule index within the row
+      const int go = (g ^ rx) * 8;           // swizzled element offset
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        bf16x8 a_frag = *(const bf16x8*)(As + (wr + fi * 16 + r16) * NT_BK + go);
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          bf16x8 b_frag = *(const bf16x8*)(Bs + (wc + fj * 16 + r16) * NT_BK + go);
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag, b_frag, acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+      int col = n0 + wc + fj * 16 + r16;
+      if (col >= N) continue;
+      float bv = 0.0f;
+      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr + fi * 16 + kgrp * 4 + r;
+        if (row >= M) continue;
+        float v = acc[fi][fj][r];
+        if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
+        if (EPI == EPI_F32) {
+          if (SPLITK) atomicAdd(&((float*)C)[(long)row * N + col], v);
+          else ((float*)C)[(long)row * N + col] = v;
+        } else {
+          ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
+        }
+      }
+    }
+  }
+}
+
+template <int EPI, typename OUT_T>
+static void launch_nt(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
+                      long M, long N, long K, int act, hipStream_t s) {
+  dim3 grid((N + NT_BN - 1) / NT_BN, (M + NT_BM - 1) / NT_BM);
+  hipLaunchKernelGGL((gemm_nt_kernel<EPI, OUT_T, false>), grid, dim3(256),
+                     NT_LDS_BYTES, s, A, B, C, bias, (int)M, (int)N, (int)K, act);
+}
+
+static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
+                                 long M, long N, long K, hipStream_t s) {
+  long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
+  long max_z = (K + NT_BK - 1) / NT_BK;
+  long z = std::min<long>(std::max<long>(512 / std::max<long>(gx * gy, 1), 1), max_z);
+  if (z <= 1) { launch_nt<EPI_F32, float>(A, B, C, nullptr, M, N, K, 0, s); return; }
+  dim3 grid((unsigned)gx, (unsigned)gy, (unsigned)z);
+  hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>), grid, dim3(256),
+                     NT_LDS_BYTES, s, A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
+}
+
+// y[B,N] = act(x[B,K] @ w[N,K]^T + b)   — the hot forward (weights [out,in])
+at::Tensor linear_nt_fwd(at::Tensor x, at::Tensor w, at::Tensor b, long act) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  CHECK_GPU(w); CHECK_CONTIG(w); CHECK_BF16(w);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long Mb = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "shape mismatch x@w^T");
+  auto y = at::empty({Mb, N}, x.options());
+  launch_nt<EPI_BIAS_ACT, bf16>((const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+                                (bf16*)y.data_ptr(), (const bf16*)b.data_ptr(),
+                                Mb, N, K, (int)act, cur_stream());
+  return y;
+}
+
+// c[M,N] = a[M,K] @ b[N,K]^T (plain bf16 out)
+at::Tensor gemm_ntv3_bf16(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
+  auto c = at::empty({M, N}, a.options());
+  launch_nt<EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                             (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+  return c;
+}
+
+// c[M,N] f32 = a[M,K] @ b[N,K]^T with split-K (wgrad)
+at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
+  auto c = at::zeros({M, N}, a.options().dtype(at::kFloat));
+  launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                       (float*)c.data_ptr(), M, N, K, cur_stream());
+  return c;
+}
+
+// ---------------------------------------------------------------------------
+// tiled bf16 transpose: out[C,R] = in[R,C]^T.  64x64 LDS tiles (+8 pad),
+// 16B coalesced global loads AND stores; scalar traffic stays inside LDS.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256)
+void transpose_bf16_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
+                           int R, int C) {
+  __shared__ bf16 tile[64][72];
+  int rt = blockIdx.y * 64, ct = blockIdx.x * 64;
+  int tid = threadIdx.x;
+  // load 64x64: 512 chunks of 8, 2 per thread
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int i = tid + it * 256;
+    int r = i >> 3, g = (i & 7) * 8;
+    int gr = rt + r, gc = ct + g;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (gr < R) {
+      const bf16* src = in + (long)gr * C + gc;
+      if (gc + 8 <= C) v = *(const s16x8*)src;
+      else for (int j = 0; j < 8; ++j)
+        ((short*)&v)[j] = (gc + j < C) ? ((const short*)src)[j] : (short)0;
+    }
+    *(s16x8*)&tile[r][g] = v;
+  }
+  __syncthreads();
+  // store 64x64 transposed: thread writes out[ct+c][rt+r..r+7]
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int i = tid + it * 256;
+    int c = i >> 3, g = (i & 7) * 8;
+    int oc = ct + c, orr = rt + g;
+    if (oc >= C) continue;
+    s16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ((short*)&v)[j] = ((short*)&tile[g + j][c])[0];
+    bf16* dst = out + (long)oc * R + orr;
+    if (orr + 8 <= R) *(s16x8*)dst = v;
+    else for (int j = 0; j < 8 && orr + j < R; ++j) ((short*)dst)[j] = ((short*)&v)[j];
+  }
+}
+
+at::Tensor transpose_bf16(at::Tensor t) {
+  CHECK_GPU(t); CHECK_CONTIG(t); CHECK_BF16(t);
+  long R = t.size(0), C = t.size(1);
+  auto out = at::empty({C, R}, t.options());
+  dim3 grid((unsigned)((C + 63) / 64), (unsigned)((R + 63) / 64));
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const bf16*)t.data_ptr(), (bf16*)out.data_ptr(), (int)R, (int)C);
+  return out;
+}
+
+
 // single-MFMA probe: d[16,16] = a[16,32] @ b[32,16] — fragment-mapping unit test
 __global__ void mfma_probe_kernel(const bf16* a, const bf16* b, float* d) {
   int lane = threadIdx.x & 63;
@@ -834,6 +1081,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nn_bf16", &gemm_nn_bf16);
   m.def("gemm_nt_bf16", &gemm_nt_bf16);
   m.def("gemm_tn_f32", &gemm_tn_f32);
+  m.def("linear_nt_fwd", &linear_nt_fwd);
+  m.def("gemm_ntv3_bf16", &gemm_ntv3_bf16);
+  m.def("gemm_ntv3_f32", &gemm_ntv3_f32);
+  m.def("transpose_bf16", &transpose_bf16);
   m.def("mfma_probe", &mfma_probe);
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);

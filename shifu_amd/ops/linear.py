@@ -1,16 +1,19 @@
-"""Fused dense layer: Y = act(X @ W + b).
+"""Fused dense layer: Y = act(X @ W^T + b), weights stored [out, in].
 
 Replaces the reference's per-layer `tf.matmul(x,W)+b` + activation
 (reference: ssgd_monitor.py:57-71 nn_layer, activations ssgd_monitor.py:74-88)
-with one fused HIP kernel on GPU (MFMA tiles, bias+activation epilogue —
-SURVEY.md §2.4 K1/K2) and a pure-PyTorch fp32 reference path on CPU.
+with hand-written CDNA4 kernels.
 
-Mixed-precision contract on GPU:
-* master weights are fp32 autograd leaves (optimizer updates fp32);
-* forward casts W/b to bf16 once per step and runs the bf16 MFMA kernel
-  (f32 accumulate);
-* backward produces dW/db in fp32 (wgrad GEMM accumulates f32 and stores
-  f32), dX in bf16 for the upstream layer.
+Layout strategy (MI355X-first): every hot GEMM runs the v3 "NT" kernel whose
+two operands are both stored reduction-major (global_load_lds staging, no
+scatter-transposes):
+  fwd   : y  = nt(x [B,K], w [N,K]) + bias/act epilogue
+  dgrad : dx = nt(dz [B,N], w^T [K,N])        (w^T = one small per-step
+                                               tiled-transpose kernel)
+  wgrad : dw = nt(dz^T [N,B], x^T [K,B])      (activation transposes, f32
+                                               out, split-K)
+Mixed precision: fp32 master weights (autograd leaves), bf16 compute,
+fp32 dW/db, bf16 dX.
 
 Activation gradients are computed from Y (not Z): sigmoid' = y(1-y),
 tanh' = 1-y^2, relu'/leakyrelu' from sign(y) — valid because all four
@@ -64,15 +67,16 @@ def _act_grad_from_y_ref(dy: torch.Tensor, y: torch.Tensor, act: int) -> torch.T
 class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor, act: int):
+        # w: [out, in]
         if use_hip(x):
             ext = hip_ops()
             wb = w.to(torch.bfloat16)
             bb = b.to(torch.bfloat16)
             xb = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
-            y = ext.linear_act_fwd(xb, wb, bb, act)
+            y = ext.linear_nt_fwd(xb.contiguous(), wb, bb, act)
             ctx.save_for_backward(xb, wb, y)
         else:
-            z = x @ w + b
+            z = x @ w.t() + b
             y = _act_fwd_ref(z, act)
             ctx.save_for_backward(x, w, y)
         ctx.act = act
@@ -88,29 +92,34 @@ class _FusedLinearFn(torch.autograd.Function):
             ext = hip_ops()
             # one fused pass: dz = dy*act'(y) and db = colsum(dz)
             dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
-            dw = ext.gemm_tn_f32(x, dz)                         # fp32 [K,N] = x^T @ dz (split-K)
-            dx = ext.gemm_nt_bf16(dz, w) if ctx.x_needs_grad else None  # bf16 [B,K] = dz @ w^T
+            xT = ext.transpose_bf16(x)                # [K,B]
+            dzT = ext.transpose_bf16(dz)              # [N,B]
+            dw = ext.gemm_ntv3_f32(dzT, xT)           # fp32 [N,K], split-K
+            dx = None
+            if ctx.x_needs_grad:
+                wT = ext.transpose_bf16(w)            # [K,N]
+                dx = ext.gemm_ntv3_bf16(dz, wT)       # bf16 [B,K]
         else:
             dz = _act_grad_from_y_ref(dy, y, act)
-            dw = x.t() @ dz
+            dw = dz.t() @ x                           # [N,K]
             db = dz.sum(dim=0)
-            dx = dz @ w.t() if ctx.x_needs_grad else None
+            dx = dz @ w if ctx.x_needs_grad else None
         return dx, dw, db, None
 
 
 def fused_linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
                  activation: str = "none") -> torch.Tensor:
+    """y = act(x @ w^T + b) with w stored [out, in]."""
     return _FusedLinearFn.apply(x, w, b, act_id(activation))
 
 
 class FusedLinear(torch.nn.Module):
-    """Dense layer with xavier init + fused forward.
+    """Dense layer with xavier init + fused forward; weight stored [out, in].
 
     Init matches the reference's nn_layer: xavier/glorot for W
-    (tf.contrib.layers.xavier_initializer, ssgd_monitor.py:63) and small
-    truncated-normal-style bias.  The L2(0.1) regularizer on W
-    (ssgd_monitor.py:58-68) is folded into the optimizer as coupled weight
-    decay (grad += l2 * w), not materialized in the loss.
+    (tf.contrib.layers.xavier_initializer, ssgd_monitor.py:63) and zero bias.
+    The L2(0.1) regularizer on W (ssgd_monitor.py:58-68) is folded into the
+    optimizer as coupled weight decay (grad += l2 * w).
     """
 
     def __init__(self, in_features: int, out_features: int,
@@ -124,8 +133,8 @@ class FusedLinear(torch.nn.Module):
         if seed is not None:
             gen = torch.Generator().manual_seed(seed)
         limit = math.sqrt(6.0 / (in_features + out_features))
-        w = (torch.rand(in_features, out_features, generator=gen) * 2 - 1) * limit
-        self.weight = torch.nn.Parameter(w)                    # [K, N] layout (x @ w)
+        w = (torch.rand(out_features, in_features, generator=gen) * 2 - 1) * limit
+        self.weight = torch.nn.Parameter(w)                    # [N, K]
         self.bias = torch.nn.Parameter(torch.zeros(out_features))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -18,7 +18,7 @@ from shifu_amd.ops.optim import FusedOptimizer
 def test_fused_linear_matches_autograd(act):
     torch.manual_seed(0)
     x = torch.randn(16, 7, requires_grad=True)
-    w = torch.randn(7, 5, requires_grad=True)
+    w = torch.randn(5, 7, requires_grad=True)   # [out, in]
     b = torch.randn(5, requires_grad=True)
 
     y = fused_linear(x, w, b, act)
@@ -28,7 +28,7 @@ def test_fused_linear_matches_autograd(act):
     x2 = x.detach().clone().requires_grad_(True)
     w2 = w.detach().clone().requires_grad_(True)
     b2 = b.detach().clone().requires_grad_(True)
-    z = x2 @ w2 + b2
+    z = x2 @ w2.t() + b2
     acts = {"none": lambda t: t, "sigmoid": torch.sigmoid, "tanh": torch.tanh,
             "relu": torch.relu,
             "leakyrelu": lambda t: torch.nn.functional.leaky_relu(t, 0.01)}

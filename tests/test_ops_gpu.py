@@ -67,6 +67,46 @@ def test_linear_act_fwd(act):
     assert ok, f"linear_act_fwd act={act} maxdiff={err}"
 
 
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (300, 96, 200), (8192, 1024, 1864),
+                                   (100, 1, 37), (513, 250, 129)])
+def test_gemm_ntv3(M, N, K):
+    """v3 fast-path GEMM (glds + swizzle): C = A @ B^T, transpose-detecting."""
+    a = _rand_bf16(M, K, seed=M + 3)
+    b = _rand_bf16(N, K, seed=N + 4)
+    c = hip_ops().gemm_ntv3_bf16(a, b)
+    ok, err = _rel_close(c, a.float() @ b.float().t())
+    assert ok, f"gemm_ntv3 {M}x{N}x{K} maxdiff={err}"
+
+
+def test_gemm_ntv3_f32_splitk():
+    a = _rand_bf16(1024, 8192, seed=50, scale=0.5)   # dzT [N,B]
+    b = _rand_bf16(1864, 8192, seed=51, scale=0.5)   # xT [K,B]
+    c = hip_ops().gemm_ntv3_f32(a, b)
+    assert c.dtype == torch.float32
+    ok, err = _rel_close(c, a.float() @ b.float().t())
+    assert ok, f"ntv3 splitk maxdiff={err}"
+
+
+@pytest.mark.parametrize("act", [0, 1, 3])
+def test_linear_nt_fwd(act):
+    x = _rand_bf16(300, 150, seed=60)
+    w = _rand_bf16(70, 150, seed=61, scale=0.3)   # [out, in]
+    b = _rand_bf16(70, seed=62)
+    y = hip_ops().linear_nt_fwd(x, w, b, act)
+    z = x.float() @ w.float().t() + b.float()
+    acts = {0: lambda t: t, 1: torch.sigmoid, 3: torch.relu}
+    ok, err = _rel_close(y, acts[act](z))
+    assert ok, f"linear_nt_fwd act={act} maxdiff={err}"
+
+
+@pytest.mark.parametrize("R,C", [(64, 64), (8192, 1864), (100, 37), (513, 1)])
+def test_transpose_bf16(R, C):
+    t = _rand_bf16(R, C, seed=R + C)
+    tt = hip_ops().transpose_bf16(t)
+    assert tt.shape == (C, R)
+    assert torch.equal(tt.float().cpu(), t.float().t().cpu())
+
+
 def test_gemm_nt():
     dz = _rand_bf16(320, 96, seed=8)
     w = _rand_bf16(130, 96, seed=9)
@@ -210,7 +250,7 @@ def test_fused_linear_autograd_gpu_vs_cpu():
     from shifu_amd.ops.linear import fused_linear
     torch.manual_seed(0)
     x32 = torch.randn(256, 96)
-    w32 = torch.randn(96, 48) * 0.2
+    w32 = torch.randn(48, 96) * 0.2   # [out, in]
     b32 = torch.randn(48) * 0.1
 
     # CPU reference

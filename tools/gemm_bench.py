@@ -12,13 +12,16 @@ import torch
 from shifu_amd.ops.dispatch import hip_ops
 
 SHAPES = [
-    # (name, M, N, K, kind)  kind: nn|nt|tn  (C[M,N], reduction K)
+    # (name, M, N, K, kind)  kind: nn|nt|tn|v3|v3f  (C[M,N], reduction K)
     ("l1_fwd", 8192, 1024, 1864, "nn"),
-    ("l2_fwd", 8192, 512, 1024, "nn"),
+    ("l1_fwd3", 8192, 1024, 1864, "v3"),
+    ("l2_fwd3", 8192, 512, 1024, "v3"),
     ("l1_dgrad", 8192, 1864, 1024, "nt"),
+    ("l1_dgrd3", 8192, 1864, 1024, "v3"),
     ("l1_wgrad", 1864, 1024, 8192, "tn"),
-    ("l2_wgrad", 1024, 512, 8192, "tn"),
+    ("l1_wgrd3", 1024, 1864, 8192, "v3f"),
     ("sq4096", 4096, 4096, 4096, "nn"),
+    ("sq4096v3", 4096, 4096, 4096, "v3"),
 ]
 
 
@@ -54,15 +57,26 @@ def main():
             b = torch.randn(N, K, device="cuda").to(torch.bfloat16)   # w [K',N']
             ours = lambda: ext.gemm_nt_bf16(a, b)
             ref = lambda: a @ b.t()
-        else:  # tn
+        elif kind == "tn":
             a = torch.randn(K, M, device="cuda").to(torch.bfloat16)   # x [B,K']
             b = torch.randn(K, N, device="cuda").to(torch.bfloat16)   # dz [B,N']
             ours = lambda: ext.gemm_tn_f32(a, b)
             ref = lambda: (a.t().float() @ b.float())
+        elif kind == "v3":
+            a = torch.randn(M, K, device="cuda").to(torch.bfloat16)
+            b = torch.randn(N, K, device="cuda").to(torch.bfloat16)
+            ours = lambda: ext.gemm_ntv3_bf16(a, b)
+            ref = lambda: a @ b.t()
+        else:  # v3f: wgrad incl. the two activation transposes
+            a0 = torch.randn(K, M, device="cuda").to(torch.bfloat16)  # dz [B,N']
+            b0 = torch.randn(K, N, device="cuda").to(torch.bfloat16)  # x [B,K']
+            ours = lambda: ext.gemm_ntv3_f32(ext.transpose_bf16(a0), ext.transpose_bf16(b0))
+            ref = lambda: a0.t().float() @ b0.float()
+            a, b = ext.transpose_bf16(a0), ext.transpose_bf16(b0)
 
         c1 = ours().float()
         c2 = (a.float() @ b.float()) if kind == "nn" else \
-             (a.float() @ b.float().t()) if kind == "nt" else \
+             (a.float() @ b.float().t()) if kind in ("nt", "v3", "v3f") else \
              (a.float().t() @ b.float())
         rel = float((c1 - c2).abs().max() / c2.abs().max().clamp_min(1e-3))
 
