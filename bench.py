@@ -63,6 +63,8 @@ def main():
     ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
     ap.add_argument("--bucket-mb", type=int, default=128)
     ap.add_argument("--embed-dim", type=int, default=EMBED_DIM)
+    ap.add_argument("--graphs", choices=["auto", "on", "off"], default="auto",
+                    help="hipGraph-capture the whole training step")
     args = ap.parse_args()
 
     rank, world, device = init_distributed()
@@ -87,15 +89,43 @@ def main():
 
     batches = make_batches(device, args.batch, rank, dtype)
 
-    def step(i):
-        dense, cats, target, weight = batches[i % N_BATCHES]
-        logits = model(dense, cats)
-        loss = weighted_loss(logits, target, weight, "sigmoid_ce")
-        loss.backward()
-        agg.finish()
-        opt.step()
-        opt.zero_grad()
-        return loss
+    use_graphs = (args.graphs == "on" or
+                  (args.graphs == "auto" and on_gpu and world == 1))
+
+    if use_graphs:
+        # static input buffers; per step we copy the rotating batch in and
+        # replay ONE hipGraph covering fwd+loss+bwd+optimizer
+        from shifu_amd.train.graph import GraphedStep
+        sdense, scats, starget, sweight = [t.clone() for t in batches[0]]
+
+        def graph_body():
+            logits = model(sdense, scats)
+            loss = weighted_loss(logits, starget, sweight, "sigmoid_ce")
+            loss.backward()
+            agg.finish()
+            opt.step()
+            opt.zero_grad()
+            return loss
+
+        stepper = GraphedStep(graph_body, warmup=3)
+
+        def step(i):
+            dense, cats, target, weight = batches[i % N_BATCHES]
+            sdense.copy_(dense)
+            scats.copy_(cats)
+            starget.copy_(target)
+            sweight.copy_(weight)
+            return stepper.run()
+    else:
+        def step(i):
+            dense, cats, target, weight = batches[i % N_BATCHES]
+            logits = model(dense, cats)
+            loss = weighted_loss(logits, target, weight, "sigmoid_ce")
+            loss.backward()
+            agg.finish()
+            opt.step()
+            opt.zero_grad()
+            return loss
 
     for i in range(args.warmup):
         step(i)
@@ -142,6 +172,7 @@ def main():
                 "parallelism": f"dp{n_gpus}",
                 "optimizer": "adam+rowwise_adagrad(emb)",
                 "loss": "sigmoid_ce",
+                "hipgraphs": use_graphs,
             },
         }
         print(json.dumps(out), flush=True)

@@ -880,6 +880,46 @@ void adam_step(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor v,
                      (float)lr, (float)b1, (float)b2, (float)eps, (float)l2, bc1, sbc2);
 }
 
+// graph-safe Adam: the step counter lives on DEVICE so hipGraph replays see
+// a fresh bias correction every replay (a host-computed bc would be frozen
+// into the captured kernel args).
+__global__ void step_incr_kernel(float* step_buf) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) step_buf[0] += 1.0f;
+}
+
+__global__ void adam_dev_kernel(float* __restrict__ w, const float* __restrict__ g,
+                                float* __restrict__ m, float* __restrict__ v,
+                                const float* __restrict__ step_buf, long n,
+                                float lr, float b1, float b2, float eps, float l2) {
+  float t = step_buf[0];
+  float bc1 = 1.0f - powf(b1, t);
+  float sbc2 = sqrtf(1.0f - powf(b2, t));
+  float step = lr * sbc2 / bc1;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long j = i; j < n; j += stride) {
+    float gj = g[j] + l2 * w[j];
+    float mj = b1 * m[j] + (1.0f - b1) * gj;
+    float vj = b2 * v[j] + (1.0f - b2) * gj * gj;
+    m[j] = mj; v[j] = vj;
+    w[j] -= step * mj / (sqrtf(vj) + eps * sbc2);
+  }
+}
+
+void adam_step_dev(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor v,
+                   at::Tensor step_buf, double lr, double b1, double b2,
+                   double eps, double l2) {
+  CHECK_GPU(w); CHECK_F32(w); CHECK_F32(step_buf);
+  auto s = cur_stream();
+  hipLaunchKernelGGL(step_incr_kernel, dim3(1), dim3(64), 0, s,
+                     (float*)step_buf.data_ptr());
+  hipLaunchKernelGGL(adam_dev_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, s,
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(),
+                     (float*)m.data_ptr(), (float*)v.data_ptr(),
+                     (const float*)step_buf.data_ptr(), w.numel(),
+                     (float)lr, (float)b1, (float)b2, (float)eps, (float)l2);
+}
+
 void adadelta_step(at::Tensor w, at::Tensor g, at::Tensor acc, at::Tensor dacc,
                    double lr, double rho, double eps, double l2) {
   CHECK_GPU(w); CHECK_F32(w);
@@ -1093,6 +1133,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("weighted_loss_bwd", &weighted_loss_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("adam_step", &adam_step);
+  m.def("adam_step_dev", &adam_step_dev);
   m.def("adadelta_step", &adadelta_step);
   m.def("adagrad_step", &adagrad_step);
   m.def("embedding_gather", &embedding_gather);

@@ -89,8 +89,12 @@ class FusedOptimizer:
         if self.kind == OPT_SGD:
             ext.sgd_step(w, g, self.lr, self.l2)
         elif self.kind == OPT_ADAM:
-            ext.adam_step(w, g, self.m, self.v, self.lr, self.b1, self.b2,
-                          self.eps, self.l2, self.step_count)
+            # device-side step counter -> hipGraph-replayable bias correction
+            if not hasattr(self, "_step_buf"):
+                self._step_buf = torch.zeros(1, device=w.device)
+                self._step_buf.fill_(float(self.step_count - 1))
+            ext.adam_step_dev(w, g, self.m, self.v, self._step_buf,
+                              self.lr, self.b1, self.b2, self.eps, self.l2)
         elif self.kind == OPT_ADADELTA:
             ext.adadelta_step(w, g, self.m, self.v, self.lr, self.rho,
                               self.eps, self.l2)
@@ -151,6 +155,8 @@ class FusedOptimizer:
 
     def load_state_dict(self, sd: dict) -> None:
         self.step_count = int(sd["step_count"])
+        if hasattr(self, "_step_buf"):
+            self._step_buf.fill_(float(self.step_count))
         if self.m is not None and sd.get("m") is not None:
             self.m.copy_(sd["m"])
         if self.v is not None and sd.get("v") is not None:
