@@ -78,6 +78,8 @@ class MultiEmbedding(torch.nn.Module):
             [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
             dtype=torch.int64)
         self.register_buffer("offsets", offsets)
+        # device-resident so forward makes no host->device copies (hipGraph-safe)
+        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64))
         gen = torch.Generator().manual_seed(seed)
         scale = 1.0 / math.sqrt(max(self.dim, 1))
         arena = (torch.rand(self.total_rows, self.dim, generator=gen) * 2 - 1) * scale
@@ -92,7 +94,6 @@ class MultiEmbedding(torch.nn.Module):
         # ids: [B, F] per-feature local ids; clamp out-of-vocab to last row
         if ids.shape[1] != self.num_features:
             raise ValueError(f"ids has {ids.shape[1]} features, expected {self.num_features}")
-        sizes = torch.tensor(self.vocab_sizes, device=ids.device, dtype=ids.dtype)
-        local = ids.clamp(min=0) % sizes  # hash-style fold of out-of-range ids
-        flat = local + self.offsets.to(ids.device)
+        local = ids.clamp(min=0) % self.sizes  # hash-style fold of out-of-range ids
+        flat = local + self.offsets
         return _EmbGatherFn.apply(self.arena, flat, self.num_features, self.dim)
