@@ -777,7 +777,8 @@ std::vector<at::Tensor> weighted_loss_fwd(at::Tensor z, at::Tensor y, at::Tensor
 
 __global__ void loss_bwd_kernel(const float* __restrict__ p, const float* __restrict__ y,
                                 const float* __restrict__ w, bf16* __restrict__ dz,
-                                long n, int kind, float scale) {
+                                long n, int kind, const float* __restrict__ scale_dev) {
+  float scale = scale_dev[0];   // device scalar: no host sync, hipGraph-safe
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long j = i; j < n; j += stride) {
@@ -788,15 +789,17 @@ __global__ void loss_bwd_kernel(const float* __restrict__ p, const float* __rest
   }
 }
 
-at::Tensor weighted_loss_bwd(at::Tensor p, at::Tensor y, at::Tensor w, long kind, double scale) {
+at::Tensor weighted_loss_bwd(at::Tensor p, at::Tensor y, at::Tensor w, long kind,
+                             at::Tensor scale) {
   CHECK_GPU(p); CHECK_CONTIG(p); CHECK_F32(p);
+  CHECK_F32(scale);
   long n = p.numel();
   auto dz = at::empty({n}, p.options().dtype(at::kBFloat16));
   int blocks = (int)std::min((n + 511) / 512 + 1, (long)1024);
   hipLaunchKernelGGL(loss_bwd_kernel, dim3(blocks), dim3(512), 0, cur_stream(),
                      (const float*)p.data_ptr(), (const float*)y.data_ptr(),
                      (const float*)w.data_ptr(), (bf16*)dz.data_ptr(),
-                     n, (int)kind, (float)scale);
+                     n, (int)kind, (const float*)scale.data_ptr());
   return dz;
 }
 

@@ -61,7 +61,8 @@ class _WeightedLossFn(torch.autograd.Function):
         if ctx.hip:
             ext = hip_ops()
             dz = ext.weighted_loss_bwd(p, y.contiguous(), w.contiguous(),
-                                       ctx.kind, float(scale))
+                                       ctx.kind,
+                                       scale.reshape(1).float().contiguous())
         else:
             yf, wf = y.float(), w.float()
             if ctx.kind == LOSS_WMSE:

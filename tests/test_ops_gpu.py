@@ -176,7 +176,8 @@ def test_weighted_loss(kind):
     assert abs(float(ls) - float(per.sum())) < 1e-2 * max(float(per.sum()), 1.0)
     assert abs(float(ws) - float(w.sum())) < 1e-3 * float(w.sum())
 
-    dz = hip_ops().weighted_loss_bwd(p, y, w, kind, 0.125)
+    dz = hip_ops().weighted_loss_bwd(p, y, w, kind,
+                                     torch.tensor([0.125], device="cuda"))
     if kind == 0:
         ref = w * 2 * (pref - y) * pref * (1 - pref) * 0.125
     else:
