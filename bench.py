@@ -108,6 +108,12 @@ def main():
             return loss
 
         stepper = GraphedStep(graph_body, warmup=3)
+        try:
+            stepper.capture()
+        except Exception as e:  # robust on fresh boxes: fall back to eager
+            print(f"# hipGraph capture failed ({e}); falling back to eager",
+                  flush=True)
+            use_graphs = False
 
         def step(i):
             dense, cats, target, weight = batches[i % N_BATCHES]
@@ -116,7 +122,7 @@ def main():
             starget.copy_(target)
             sweight.copy_(weight)
             return stepper.run()
-    else:
+    if not use_graphs:
         def step(i):
             dense, cats, target, weight = batches[i % N_BATCHES]
             logits = model(dense, cats)
