@@ -65,6 +65,9 @@ def main():
     ap.add_argument("--embed-dim", type=int, default=EMBED_DIM)
     ap.add_argument("--graphs", choices=["auto", "on", "off"], default="auto",
                     help="hipGraph-capture the whole training step")
+    ap.add_argument("--emb-mode", choices=["auto", "dp", "ep"], default="auto",
+                    help="embedding parallelism: replicated+sparse-allgather (dp) "
+                         "or sharded+all-to-all (ep); auto=ep when world>1")
     args = ap.parse_args()
 
     rank, world, device = init_distributed()
@@ -73,8 +76,10 @@ def main():
     dtype = torch.bfloat16 if on_gpu else torch.float32
 
     torch.manual_seed(777)
+    use_ep = (args.emb_mode == "ep" or (args.emb_mode == "auto" and world > 1))
     model = WideDeep(N_DENSE, [VOCAB] * N_CAT, args.embed_dim, TOWER, ACTS,
-                     seed=777).to(device)
+                     seed=777, sharded_embeddings=use_ep, world=world,
+                     rank=rank).to(device)
     if on_gpu:
         # keep embedding arenas bf16 (HBM-resident, gathered by the HIP kernel)
         for p in model.parameters():
@@ -175,7 +180,8 @@ def main():
                 "global_batch": n_gpus * args.batch,
                 "per_gpu_batch": args.batch,
                 "seq_len": None,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"dp{n_gpus}+ep{n_gpus}(emb)" if use_ep
+                                else f"dp{n_gpus}"),
                 "optimizer": "adam+rowwise_adagrad(emb)",
                 "loss": "sigmoid_ce",
                 "hipgraphs": use_graphs,

@@ -19,16 +19,24 @@ from shifu_amd.ops.loss import predict_proba
 
 class DeepFM(torch.nn.Module):
     def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
-                 hidden_nodes: List[int], activations: List[str], seed: int = 1234):
+                 hidden_nodes: List[int], activations: List[str], seed: int = 1234,
+                 sharded_embeddings: bool = False, world: int = 1, rank: int = 0):
         super().__init__()
         self.num_dense = num_dense
         self.embed_dim = embed_dim
         self.vocab_sizes = list(vocab_sizes)
         F = len(self.vocab_sizes)
 
-        self.fm_first = MultiEmbedding(self.vocab_sizes, 1, seed=seed + 11)
+        if sharded_embeddings and world > 1:
+            from shifu_amd.parallel.ep import ShardedEmbedding
+            emb = lambda d, s: ShardedEmbedding(self.vocab_sizes, d, seed=s,
+                                                world=world, rank=rank)
+        else:
+            emb = lambda d, s: MultiEmbedding(self.vocab_sizes, d, seed=s)
+
+        self.fm_first = emb(1, seed + 11)
         self.fm_dense = FusedLinear(num_dense, 1, activation="none", seed=seed + 12)
-        self.embeddings = MultiEmbedding(self.vocab_sizes, embed_dim, seed=seed + 21)
+        self.embeddings = emb(embed_dim, seed + 21)
 
         tower_in = num_dense + F * embed_dim
         layers, prev = [], tower_in

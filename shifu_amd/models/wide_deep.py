@@ -23,19 +23,27 @@ from shifu_amd.ops.loss import predict_proba
 
 class WideDeep(torch.nn.Module):
     def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
-                 hidden_nodes: List[int], activations: List[str], seed: int = 1234):
+                 hidden_nodes: List[int], activations: List[str], seed: int = 1234,
+                 sharded_embeddings: bool = False, world: int = 1, rank: int = 0):
         super().__init__()
         self.num_dense = num_dense
         self.embed_dim = embed_dim
         self.vocab_sizes = list(vocab_sizes)
         F = len(self.vocab_sizes)
 
+        if sharded_embeddings and world > 1:
+            from shifu_amd.parallel.ep import ShardedEmbedding
+            emb = lambda d, s: ShardedEmbedding(self.vocab_sizes, d, seed=s,
+                                                world=world, rank=rank)
+        else:
+            emb = lambda d, s: MultiEmbedding(self.vocab_sizes, d, seed=s)
+
         # wide part
-        self.wide_cat = MultiEmbedding(self.vocab_sizes, 1, seed=seed + 101)
+        self.wide_cat = emb(1, seed + 101)
         self.wide_dense = FusedLinear(num_dense, 1, activation="none", seed=seed + 102)
 
         # deep part
-        self.embeddings = MultiEmbedding(self.vocab_sizes, embed_dim, seed=seed + 201)
+        self.embeddings = emb(embed_dim, seed + 201)
         tower_in = num_dense + F * embed_dim
         layers, prev = [], tower_in
         for i, (h, a) in enumerate(zip(hidden_nodes, activations)):

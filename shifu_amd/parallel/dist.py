@@ -167,6 +167,9 @@ class GradAggregator:
 
         if is_distributed() and self.sync_enabled:
             for p in self.emb_params:
+                if getattr(p, "_is_ep_sharded", False):
+                    continue  # sharded arenas: backward all-to-all already
+                              # delivered the complete per-row gradient
                 self._aggregate_sparse(p, world)
 
     def _aggregate_sparse(self, p: torch.nn.Parameter, world: int) -> None:
