@@ -10,7 +10,7 @@ import os
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
 from setuptools import setup
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+from torch.utils.cpp_extension import BuildExtension, CppExtension, CUDAExtension
 
 setup(
     name="shifu_amd",
@@ -24,7 +24,13 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
             },
-        )
+        ),
+        CppExtension(
+            name="shifu_amd.io._shifu_io",
+            sources=["shifu_amd/io/cpp/csv_reader.cpp"],
+            libraries=["z"],
+            extra_compile_args=["-O3", "-std=c++17"],
+        ),
     ],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )

@@ -29,9 +29,10 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
     init_distributed(rc.resolved_backend(), rc.master_addr, rc.master_port)
     try:
         files = list_training_files(rc.training_data_path)
-        full = load_csv_files(files, rc.selected_numeric_columns,
-                              rc.selected_categorical_columns,
-                              rc.target_column, rc.weight_column, rc.delimiter)
+        from shifu_amd.io import load_csv_native
+        full = load_csv_native(files, rc.selected_numeric_columns,
+                               rc.selected_categorical_columns,
+                               rc.target_column, rc.weight_column, rc.delimiter)
         # deterministic split BEFORE sharding so every rank agrees on the
         # valid set (reference splits per-worker after file sharding; a
         # shared-seed global split keeps valid metrics comparable)
