@@ -1056,6 +1056,22 @@ __global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restr
   }
 }
 
+// small-D phase 1: one THREAD per entry (a wave per entry would idle 63/64
+// lanes at D=1 — the wide-column arena case)
+__global__ void emb_accsq_small_kernel(float* __restrict__ acc, const long* __restrict__ rows,
+                                       const float* __restrict__ vals, long n, long D) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long e = i; e < n; e += stride) {
+    float sq = 0.0f;
+    for (long d = 0; d < D; ++d) {
+      float g = vals[e * D + d];
+      sq += g * g;
+    }
+    atomicAdd(&acc[rows[e]], sq / (float)D);
+  }
+}
+
 // phase 1 of adagrad: acc[rows[i]] += mean_d vals[i,d]^2  (wave per entry)
 __global__ void emb_accsq_kernel(float* __restrict__ acc, const long* __restrict__ rows,
                                  const float* __restrict__ vals, long n, long D) {
@@ -1101,11 +1117,17 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
   long n = rows.numel(), D = arena.size(1);
   if (!n) return;
   auto s = cur_stream();
-  int wpb = 4;
-  hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + wpb - 1) / wpb)),
-                     dim3(64 * wpb), 0, s,
-                     (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
-                     (const float*)vals.data_ptr(), n, D);
+  if (D <= 8) {
+    hipLaunchKernelGGL(emb_accsq_small_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
+                       (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                       (const float*)vals.data_ptr(), n, D);
+  } else {
+    int wpb = 4;
+    hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + wpb - 1) / wpb)),
+                       dim3(64 * wpb), 0, s,
+                       (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                       (const float*)vals.data_ptr(), n, D);
+  }
   auto rowscale = at::empty({n}, vals.options());
   hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
                      (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
@@ -1115,6 +1137,113 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
                      (const float*)vals.data_ptr(), (const float*)rowscale.data_ptr(),
                      n, D, (float)-lr);
+}
+
+// ---------------------------------------------------------------------------
+// GEMV path for 1-unit heads (shifu_output_0 / wide linear): a 128x128 MFMA
+// tile on N=1 wastes 127/128 of the math — these run at memory speed.
+// ---------------------------------------------------------------------------
+// y[b] = act(sum_k x[b,k]*w[k] + bias) : one wave per row, shuffle-reduce
+__global__ __launch_bounds__(256)
+void gemv_fwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ w,
+                     const bf16* __restrict__ bias, bf16* __restrict__ y,
+                     long B, long K, int act) {
+  long row = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= B) return;
+  int lane = threadIdx.x & 63;
+  float acc = 0.0f;
+  for (long k = lane; k < K; k += 64)
+    acc += __bfloat162float(x[row * K + k]) * __bfloat162float(w[k]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  if (lane == 0) y[row] = __float2bfloat16(act_fwd(acc + __bfloat162float(bias[0]), act));
+}
+
+// dw[k] = sum_b x[b,k]*dz[b] ; db = sum_b dz[b]  (column-parallel + atomics)
+__global__ void gemv_wgrad_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
+                                  float* __restrict__ dw, float* __restrict__ db,
+                                  long B, long K, long rows_per_chunk) {
+  long k = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long b0 = (long)blockIdx.y * rows_per_chunk;
+  long b1 = min(B, b0 + rows_per_chunk);
+  if (k < K) {
+    float acc = 0.0f;
+    for (long b = b0; b < b1; ++b)
+      acc += __bfloat162float(x[b * K + k]) * __bfloat162float(dz[b]);
+    if (gridDim.y == 1) dw[k] = acc;
+    else atomicAdd(&dw[k], acc);
+  }
+  if (blockIdx.x == 0 && threadIdx.x < 64) {   // one wave reduces db chunk
+    float acc = 0.0f;
+    for (long b = b0 + threadIdx.x; b < b1; b += 64)
+      acc += __bfloat162float(dz[b]);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+    if (threadIdx.x == 0) atomicAdd(&db[0], acc);
+  }
+}
+
+// dx[b,k] = dz[b] * w[k]  (outer product, vectorized by 8)
+__global__ void gemv_dgrad_kernel(const bf16* __restrict__ dz, const bf16* __restrict__ w,
+                                  bf16* __restrict__ dx, long B, long K) {
+  long chunks = K / 8;
+  long total = B * chunks;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    long b = t / chunks, c = (t % chunks) * 8;
+    float d = __bfloat162float(dz[b]);
+    s16x8 wv = *(const s16x8*)(w + c);
+    s16x8 out;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      ((bf16*)&out)[j] = __float2bfloat16(d * __bfloat162float(((const bf16*)&wv)[j]));
+    *(s16x8*)(dx + b * K + c) = out;
+  }
+  // scalar tail columns
+  for (long b = i; b < B && (K & 7); b += stride) {
+    float d = __bfloat162float(dz[b]);
+    for (long k = chunks * 8; k < K; ++k)
+      dx[b * K + k] = __float2bfloat16(d * __bfloat162float(w[k]));
+  }
+}
+
+at::Tensor gemv_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  long B = x.size(0), K = x.size(1);
+  auto y = at::empty({B, 1}, x.options());
+  int wpb = 4;
+  hipLaunchKernelGGL(gemv_fwd_kernel, dim3((unsigned)((B + wpb - 1) / wpb)),
+                     dim3(64 * wpb), 0, cur_stream(),
+                     (const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+                     (const bf16*)bias.data_ptr(), (bf16*)y.data_ptr(), B, K, (int)act);
+  return y;
+}
+
+std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
+                                 bool need_dx) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  long B = x.size(0), K = x.size(1);
+  auto dw = at::zeros({1, K}, x.options().dtype(at::kFloat));
+  auto db = at::zeros({1}, x.options().dtype(at::kFloat));
+  long gx, chunks, rpc;
+  colsum_grid(B, K, gx, chunks, rpc);
+  hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)gx, (unsigned)chunks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
+                     (float*)dw.data_ptr(), (float*)db.data_ptr(), B, K, rpc);
+  at::Tensor dx;
+  if (need_dx) {
+    dx = at::empty({B, K}, x.options());
+    long total = B * std::max<long>(K / 8, 1);
+    hipLaunchKernelGGL(gemv_dgrad_kernel, dim3(scat_blocks(total)), dim3(256),
+                       0, cur_stream(),
+                       (const bf16*)dz.data_ptr(), (const bf16*)w.data_ptr(),
+                       (bf16*)dx.data_ptr(), B, K);
+  } else {
+    dx = dw;  // placeholder; caller ignores dx when need_dx=false
+  }
+  return {dw, db, dx};
 }
 
 // ---------------------------------------------------------------------------
@@ -1128,6 +1257,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_ntv3_bf16", &gemm_ntv3_bf16);
   m.def("gemm_ntv3_f32", &gemm_ntv3_f32);
   m.def("transpose_bf16", &transpose_bf16);
+  m.def("gemv_fwd", &gemv_fwd);
+  m.def("gemv_bwd", &gemv_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);

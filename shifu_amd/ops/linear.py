@@ -73,7 +73,11 @@ class _FusedLinearFn(torch.autograd.Function):
             wb = w.to(torch.bfloat16)
             bb = b.to(torch.bfloat16)
             xb = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
-            y = ext.linear_nt_fwd(xb.contiguous(), wb, bb, act)
+            xb = xb.contiguous()
+            if w.shape[0] == 1:   # 1-unit head: GEMV path (memory-speed)
+                y = ext.gemv_fwd(xb, wb, bb, act)
+            else:
+                y = ext.linear_nt_fwd(xb, wb, bb, act)
             ctx.save_for_backward(xb, wb, y)
         else:
             z = x @ w.t() + b
@@ -90,6 +94,12 @@ class _FusedLinearFn(torch.autograd.Function):
         act = ctx.act
         if ctx.hip:
             ext = hip_ops()
+            if w.shape[0] == 1:   # 1-unit head: GEMV backward
+                dz = ext.act_grad(dy.contiguous(), y, act).reshape(-1)
+                dw, db, dx = ext.gemv_bwd(x, w.reshape(-1), dz, ctx.x_needs_grad)
+                if not ctx.x_needs_grad:
+                    dx = None
+                return dx, dw, db, None
             # one fused pass: dz = dy*act'(y) and db = colsum(dz)
             dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
             xT = ext.transpose_bf16(x)                # [K,B]

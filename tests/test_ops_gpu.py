@@ -107,6 +107,49 @@ def test_transpose_bf16(R, C):
     assert torch.equal(tt.float().cpu(), t.float().t().cpu())
 
 
+@pytest.mark.parametrize("act", [0, 1])
+def test_gemv_head_path(act):
+    """1-unit head GEMV fwd/bwd vs reference (the shifu_output_0 shape)."""
+    x = _rand_bf16(1000, 37, seed=70)
+    w = _rand_bf16(1, 37, seed=71, scale=0.3)
+    b = _rand_bf16(1, seed=72)
+    y = hip_ops().gemv_fwd(x, w, b, act)
+    z = x.float() @ w.float().t() + b.float()
+    ref_y = torch.sigmoid(z) if act == 1 else z
+    ok, err = _rel_close(y, ref_y)
+    assert ok, f"gemv_fwd maxdiff={err}"
+
+    dz = _rand_bf16(1000, seed=73)
+    dw, db, dx = hip_ops().gemv_bwd(x, w.reshape(-1), dz, True)
+    ok, err = _rel_close(dw, (dz.float().reshape(1, -1) @ x.float()))
+    assert ok, f"gemv dw maxdiff={err}"
+    assert abs(float(db) - float(dz.float().sum())) < 0.05 * max(abs(float(dz.float().sum())), 1)
+    ok, err = _rel_close(dx, dz.float().reshape(-1, 1) @ w.float())
+    assert ok, f"gemv dx maxdiff={err}"
+
+
+def test_fused_linear_head_autograd_gpu():
+    """FusedLinear with out_features=1 routes through GEMV; compare vs CPU."""
+    from shifu_amd.ops.linear import fused_linear
+    torch.manual_seed(2)
+    x32 = torch.randn(512, 64)
+    w32 = torch.randn(1, 64) * 0.2
+    b32 = torch.randn(1) * 0.1
+    xc = x32.clone().requires_grad_(True)
+    wc = w32.clone().requires_grad_(True)
+    bc = b32.clone().requires_grad_(True)
+    fused_linear(xc, wc, bc, "none").pow(2).sum().backward()
+    xg = x32.to(torch.bfloat16).cuda().requires_grad_(True)
+    wg = w32.clone().cuda().requires_grad_(True)
+    bg = b32.clone().cuda().requires_grad_(True)
+    fused_linear(xg, wg, bg, "none").float().pow(2).sum().backward()
+    for got, want, name in [(wg.grad, wc.grad, "dw"), (bg.grad, bc.grad, "db"),
+                            (xg.grad, xc.grad, "dx")]:
+        rel = float((got.float().cpu() - want).abs().max() /
+                    want.abs().max().clamp_min(1e-3))
+        assert rel < 0.08, f"{name} rel {rel}"
+
+
 def test_gemm_nt():
     dz = _rand_bf16(320, 96, seed=8)
     w = _rand_bf16(130, 96, seed=9)
