@@ -26,17 +26,19 @@ from shifu_amd.ops.dispatch import use_hip, hip_ops
 
 
 def sparse_rows_values(g: torch.Tensor):
-    """(rows int64, fp32 values [n,D]) from a sparse grad.
+    """(rows int64, values [n,D] in grad dtype) from a sparse grad.
 
     Rows MAY CONTAIN DUPLICATES: every consumer tolerates them — the HIP
     update kernels scatter with atomics (global_atomic_pk_add_bf16 / CAS),
     the CPU path uses index_add_, and updates are linear in the gradient.
     Skipping the dedup avoids a rocprim sort (torch.unique) per step, which
-    profiling showed at ~20% of the Wide&Deep step."""
+    profiling showed at ~20% of the Wide&Deep step.  Values keep their
+    native dtype (bf16 on GPU): the update kernels read bf16 directly and
+    the DP allgather wires half the bytes."""
     if not g.is_sparse:
         rows = torch.nonzero(g.abs().sum(dim=1) != 0, as_tuple=False).reshape(-1)
-        return rows, g[rows].float()
-    return g._indices()[0], g._values().float()
+        return rows, g[rows]
+    return g._indices()[0], g._values()
 
 
 class _EmbGatherFn(torch.autograd.Function):

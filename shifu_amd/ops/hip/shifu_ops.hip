@@ -317,7 +317,8 @@ at::Tensor gemm_tn_f32(at::Tensor x, at::Tensor dz) {
   CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
   long Bb = x.size(0), K = x.size(1), N = dz.size(1);
   TORCH_CHECK(dz.size(0) == Bb, "shape mismatch x^T@dz");
-  auto dw = at::zeros({K, N}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({K, N}, x.options().dtype(at::kFloat));
+  hipMemsetAsync(dw.data_ptr(), 0, (size_t)K * N * 4, cur_stream());
   // C[M=K, N'=N] = (x^T)[M=K, Kr=B] * dz[Kr=B, N]; x stored [B,K] => TA=1
   launch_gemm_splitk<1, 0>((const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
                            (float*)dw.data_ptr(), K, N, Bb, cur_stream());
@@ -512,7 +513,8 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
   long M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
-  auto c = at::zeros({M, N}, a.options().dtype(at::kFloat));
+  auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
+  hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
   launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                        (float*)c.data_ptr(), M, N, K, cur_stream());
   return c;
@@ -675,23 +677,47 @@ at::Tensor colsum_f32(at::Tensor dz) {
 
 // fused act-grad + bias-grad: dz = dy*act'(y) AND db = colsum(dz) in ONE pass
 // over [B,N] — saves the full dz re-read a separate colsum would cost.
+// each thread owns 8 consecutive columns (16B vector loads/stores per row)
 __global__ void act_grad_colsum_kernel(const bf16* __restrict__ dy,
                                        const bf16* __restrict__ y,
                                        bf16* __restrict__ dz, float* __restrict__ db,
                                        long B, long N, long rows_per_chunk, int act) {
-  long n = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (n >= N) return;
+  long n0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (n0 >= N) return;
   long b0 = (long)blockIdx.y * rows_per_chunk;
   long b1 = min(B, b0 + rows_per_chunk);
-  float acc = 0.0f;
-  for (long b = b0; b < b1; ++b) {
-    long i = b * N + n;
-    float g = __bfloat162float(dy[i]) * act_grad_from_y(__bfloat162float(y[i]), act);
-    dz[i] = __float2bfloat16(g);
-    acc += g;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (n0 + 8 <= N) {
+    for (long b = b0; b < b1; ++b) {
+      long i = b * N + n0;
+      s16x8 dyv = *(const s16x8*)(dy + i);
+      s16x8 yv = *(const s16x8*)(y + i);
+      s16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = __bfloat162float(((const bf16*)&dyv)[j]) *
+                  act_grad_from_y(__bfloat162float(((const bf16*)&yv)[j]), act);
+        ((bf16*)&out)[j] = __float2bfloat16(g);
+        acc[j] += g;
+      }
+      *(s16x8*)(dz + i) = out;
+    }
+  } else {
+    for (long b = b0; b < b1; ++b)
+      for (long n = n0; n < N; ++n) {
+        long i = b * N + n;
+        float g = __bfloat162float(dy[i]) * act_grad_from_y(__bfloat162float(y[i]), act);
+        dz[i] = __float2bfloat16(g);
+        acc[n - n0] += g;
+      }
   }
-  if (gridDim.y == 1) db[n] = acc;
-  else atomicAdd(&db[n], acc);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    long n = n0 + j;
+    if (n >= N) break;
+    if (gridDim.y == 1) db[n] = acc[j];
+    else atomicAdd(&db[n], acc[j]);
+  }
 }
 
 std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
@@ -701,7 +727,7 @@ std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
   auto dz = at::empty_like(dy);
   auto db = at::zeros({N}, dy.options().dtype(at::kFloat));
   long gx, chunks, rpc;
-  colsum_grid(B, N, gx, chunks, rpc);
+  colsum_grid(B, (N + 7) / 8, gx, chunks, rpc);
   hipLaunchKernelGGL(act_grad_colsum_kernel, dim3((unsigned)gx, (unsigned)chunks),
                      dim3(256), 0, cur_stream(),
                      (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
@@ -1025,7 +1051,7 @@ DEVINL void atomic_add_bf16_scalar(bf16* p, float v) {
 
 // arena[rows[i]] += scale * vals[i] * rowscale[i] (rowscale nullable)
 __global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
-                                   const float* __restrict__ vals,
+                                   const bf16* __restrict__ vals,
                                    const float* __restrict__ rowscale,
                                    long n, long D, float scale) {
   long pairs = D / 2;
@@ -1038,20 +1064,21 @@ __global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restr
       float sc = scale * (rowscale ? rowscale[e] : 1.0f);
       long row = rows[e];
       __hip_bfloat162 add;
-      add.x = __float2bfloat16(sc * vals[e * D + dp * 2]);
-      add.y = __float2bfloat16(sc * vals[e * D + dp * 2 + 1]);
+      add.x = __float2bfloat16(sc * __bfloat162float(vals[e * D + dp * 2]));
+      add.y = __float2bfloat16(sc * __bfloat162float(vals[e * D + dp * 2 + 1]));
       unsafeAtomicAdd((__hip_bfloat162*)(arena + row * D + dp * 2), add);
     }
     if (D & 1) {  // odd tail element
       for (long e = i; e < n; e += stride) {
         float sc = scale * (rowscale ? rowscale[e] : 1.0f);
-        atomic_add_bf16_scalar(arena + rows[e] * D + D - 1, sc * vals[e * D + D - 1]);
+        atomic_add_bf16_scalar(arena + rows[e] * D + D - 1,
+                               sc * __bfloat162float(vals[e * D + D - 1]));
       }
     }
   } else {  // D == 1
     for (long e = i; e < n; e += stride) {
       float sc = scale * (rowscale ? rowscale[e] : 1.0f);
-      atomic_add_bf16_scalar(arena + rows[e], sc * vals[e]);
+      atomic_add_bf16_scalar(arena + rows[e], sc * __bfloat162float(vals[e]));
     }
   }
 }
@@ -1059,13 +1086,13 @@ __global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restr
 // small-D phase 1: one THREAD per entry (a wave per entry would idle 63/64
 // lanes at D=1 — the wide-column arena case)
 __global__ void emb_accsq_small_kernel(float* __restrict__ acc, const long* __restrict__ rows,
-                                       const float* __restrict__ vals, long n, long D) {
+                                       const bf16* __restrict__ vals, long n, long D) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long e = i; e < n; e += stride) {
     float sq = 0.0f;
     for (long d = 0; d < D; ++d) {
-      float g = vals[e * D + d];
+      float g = __bfloat162float(vals[e * D + d]);
       sq += g * g;
     }
     atomicAdd(&acc[rows[e]], sq / (float)D);
@@ -1074,13 +1101,13 @@ __global__ void emb_accsq_small_kernel(float* __restrict__ acc, const long* __re
 
 // phase 1 of adagrad: acc[rows[i]] += mean_d vals[i,d]^2  (wave per entry)
 __global__ void emb_accsq_kernel(float* __restrict__ acc, const long* __restrict__ rows,
-                                 const float* __restrict__ vals, long n, long D) {
+                                 const bf16* __restrict__ vals, long n, long D) {
   long e = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (e >= n) return;
   int lane = threadIdx.x & 63;
   float sq = 0.0f;
   for (long d = lane; d < D; d += 64) {
-    float g = vals[e * D + d];
+    float g = __bfloat162float(vals[e * D + d]);
     sq += g * g;
   }
 #pragma unroll
@@ -1102,31 +1129,31 @@ static int scat_blocks(long total) {
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
-  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(vals);
+  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_BF16(vals);
   long n = rows.numel(), D = arena.size(1);
   if (!n) return;
   long total = n * std::max<long>(D / 2, 1);
   hipLaunchKernelGGL(emb_scatter_kernel, dim3(scat_blocks(total)), dim3(256), 0, cur_stream(),
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
-                     (const float*)vals.data_ptr(), nullptr, n, D, (float)-lr);
+                     (const bf16*)vals.data_ptr(), nullptr, n, D, (float)-lr);
 }
 
 void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Tensor vals,
                       double lr, double eps) {
-  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(acc); CHECK_F32(vals);
+  CHECK_GPU(arena); CHECK_BF16(arena); CHECK_F32(acc); CHECK_BF16(vals);
   long n = rows.numel(), D = arena.size(1);
   if (!n) return;
   auto s = cur_stream();
   if (D <= 8) {
     hipLaunchKernelGGL(emb_accsq_small_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
-                       (const float*)vals.data_ptr(), n, D);
+                       (const bf16*)vals.data_ptr(), n, D);
   } else {
     int wpb = 4;
     hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + wpb - 1) / wpb)),
                        dim3(64 * wpb), 0, s,
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
-                       (const float*)vals.data_ptr(), n, D);
+                       (const bf16*)vals.data_ptr(), n, D);
   }
   auto rowscale = at::empty({n}, vals.options());
   hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
@@ -1135,7 +1162,7 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
   long total = n * std::max<long>(D / 2, 1);
   hipLaunchKernelGGL(emb_scatter_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
-                     (const float*)vals.data_ptr(), (const float*)rowscale.data_ptr(),
+                     (const bf16*)vals.data_ptr(), (const float*)rowscale.data_ptr(),
                      n, D, (float)-lr);
 }
 
@@ -1224,8 +1251,10 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
                                  bool need_dx) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
   long B = x.size(0), K = x.size(1);
-  auto dw = at::zeros({1, K}, x.options().dtype(at::kFloat));
-  auto db = at::zeros({1}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({1, K}, x.options().dtype(at::kFloat));
+  auto db = at::empty({1}, x.options().dtype(at::kFloat));
+  hipMemsetAsync(dw.data_ptr(), 0, (size_t)K * 4, cur_stream());
+  hipMemsetAsync(db.data_ptr(), 0, 4, cur_stream());
   long gx, chunks, rpc;
   colsum_grid(B, K, gx, chunks, rpc);
   hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)gx, (unsigned)chunks),

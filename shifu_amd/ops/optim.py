@@ -115,16 +115,17 @@ class FusedOptimizer:
                 hip_ops().emb_adagrad_step(p.data, acc, rows.contiguous(),
                                            vals.contiguous(), self.emb_lr, self.eps)
             else:
-                rowsq = (vals * vals).mean(dim=1)
+                vf = vals.float()
+                rowsq = (vf * vf).mean(dim=1)
                 acc.index_add_(0, rows, rowsq)
                 denom = acc[rows].add(self.eps).sqrt_().unsqueeze(1)
-                p.data.index_add_(0, rows, (-self.emb_lr * vals / denom).to(p.dtype))
+                p.data.index_add_(0, rows, (-self.emb_lr * vf / denom).to(p.dtype))
         else:  # sgd
             if p.dtype == torch.bfloat16 and use_hip(p):
                 hip_ops().emb_sgd_step(p.data, rows.contiguous(), vals.contiguous(),
                                        self.emb_lr)
             else:
-                p.data.index_add_(0, rows, (-self.emb_lr * vals).to(p.dtype))
+                p.data.index_add_(0, rows, (-self.emb_lr * vals.float()).to(p.dtype))
         p.grad = None
 
     # ------------------------------------------------------------------ api

@@ -273,18 +273,19 @@ def test_emb_updates_gpu():
     arena = torch.randn(200, 32).to(torch.bfloat16).cuda()
     ref = arena.float().cpu().clone()
     rows = torch.tensor([3, 77, 150]).cuda()
-    vals = torch.randn(3, 32).cuda()
+    vals = torch.randn(3, 32).to(torch.bfloat16).cuda()
     hip_ops().emb_sgd_step(arena, rows, vals, 0.5)
-    ref[rows.cpu()] -= 0.5 * vals.cpu()
+    ref[rows.cpu()] -= 0.5 * vals.float().cpu()
     assert torch.allclose(arena.float().cpu(), ref, atol=1e-2)
 
     acc = torch.zeros(200).cuda()
     arena2 = torch.randn(200, 32).to(torch.bfloat16).cuda()
     ref2 = arena2.float().cpu().clone()
     hip_ops().emb_adagrad_step(arena2, acc, rows, vals, 0.1, 1e-8)
-    rowsq = (vals * vals).mean(dim=1)
+    vf = vals.float()
+    rowsq = (vf * vf).mean(dim=1)
     denom = (rowsq.sqrt() + 1e-8).unsqueeze(1)
-    ref2[rows.cpu()] -= (0.1 * vals / denom).cpu()
+    ref2[rows.cpu()] -= (0.1 * vf / denom).cpu()
     assert torch.allclose(arena2.float().cpu(), ref2, atol=1e-2)
     assert torch.allclose(acc[rows].cpu(), rowsq.cpu(), atol=1e-5)
 
