@@ -462,6 +462,243 @@ void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 }
 
+// ---------------------------------------------------------------------------
+// v4 "NT" GEMM — 256x256 tile, 8 waves, deep-pipelined glds ring.
+// Same operand contract as v3 (A [M,K], B [N,K], both reduction-major) but:
+//   * 512 threads = 8 waves as 2(M) x 4(N); per-wave output 128x64
+//     (acc[8][4] f32x4 = 128 VGPRs); B fragments for a whole K-tile are
+//     read once per tile and held in registers (32 VGPRs).
+//   * 4 phases per K-tile; phase q computes accumulator rows 2q..2q+1
+//     (16 MFMAs) after reading 4 A-fragments; one 16 KB half-tile of a
+//     FUTURE K-tile is staged per phase by glds into a 2-deep ring
+//     (A-halves one tile ahead, B-halves two tiles ahead — each region is
+//     staged in the first phase after its previous tile's readers passed a
+//     barrier, so placement-independent by construction).
+//   * counted s_waitcnt vmcnt(4) at each K-tile boundary (never 0 in the
+//     loop) + raw s_barrier per phase — loads stay in flight across
+//     barriers (guide T3+T4); s_setprio(1) wraps each MFMA cluster (T5).
+//   * K-edge tiles fall back to a guarded ds_write stage producing the
+//     same swizzled image (granule g of row r lives at g ^ (r&7)).
+// ---------------------------------------------------------------------------
+#define V4_BM 256
+#define V4_BN 256
+#define V4_BK 64
+// LDS regions (bytes): A ring buffer = tile parity; 32KB per operand-tile
+#define V4_A(par) ((par) * 65536)
+#define V4_B(par) ((par) * 65536 + 32768)
+
+DEVINL void v4_stage_half_glds(const bf16* __restrict__ P, char* lds_region,
+                               int R0, int half, int k0, int K, int tid) {
+  // one 16KB half (rows half*128..+127 of the 256-row region), 2 glds/thread
+  const int lane = tid & 63, wave = tid >> 6;
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int byte = c * 8192 + wave * 1024 + lane * 16;
+    int row = byte >> 7;               // within half
+    int g = (byte >> 4) & 7;
+    int gsrc = g ^ (row & 7);          // swizzle on the SOURCE address
+    const bf16* src = P + (long)(R0 + half * 128 + row) * K + k0 + gsrc * 8;
+    char* dst = lds_region + half * 16384 + c * 8192 + wave * 1024;
+    __builtin_amdgcn_global_load_lds((gas_ptr)src, (las_ptr)dst, 16, 0, 0);
+  }
+}
+
+DEVINL void v4_stage_half_guarded(const bf16* __restrict__ P, char* lds_region,
+                                  int R0, int half, int k0, int RM, int K, int tid) {
+  // same image via guarded loads + ds_writes (edge tiles); 2 chunks/thread
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int byte = c * 8192 + (tid >> 6) * 1024 + (tid & 63) * 16;
+    int row = byte >> 7;
+    int g = (byte >> 4) & 7;
+    int gsrc = g ^ (row & 7);
+    int grow = R0 + half * 128 + row;
+    long gk = (long)k0 + gsrc * 8;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (grow < RM) {
+      const bf16* src = P + (long)grow * K + gk;
+      if (gk + 8 <= K) v = *(const s16x8*)src;
+      else for (int j = 0; j < 8; ++j)
+        ((short*)&v)[j] = (gk + j < K) ? ((const short*)src)[j] : (short)0;
+    }
+    *(s16x8*)(lds_region + half * 16384 + byte) = v;
+  }
+}
+
+// stage one slot: slot 0/1 = A half0/1, slot 2/3 = B half0/1, of k-tile kt
+DEVINL void v4_stage_slot(const bf16* A, const bf16* B, char* smem, int slot,
+                          int kt, int m0, int n0, int M, int N, int K, int tid) {
+  int k0 = kt * V4_BK;
+  int par = kt & 1;
+  bool full_k = (k0 + V4_BK <= K);
+  if (slot < 2) {
+    bool full = full_k && (m0 + V4_BM <= M);
+    if (full) v4_stage_half_glds(A, smem + V4_A(par), m0, slot, k0, K, tid);
+    else v4_stage_half_guarded(A, smem + V4_A(par), m0, slot, k0, M, K, tid);
+  } else {
+    bool full = full_k && (n0 + V4_BN <= N);
+    if (full) v4_stage_half_glds(B, smem + V4_B(par), n0, slot - 2, k0, K, tid);
+    else v4_stage_half_guarded(B, smem + V4_B(par), n0, slot - 2, k0, N, K, tid);
+  }
+}
+
+template <int EPI, typename OUT_T, bool SPLITK = false>
+__global__ __launch_bounds__(512, 1)
+void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                       OUT_T* __restrict__ C, const bf16* __restrict__ bias,
+                       int M, int N, int K, int act) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;            // 0..1 -> M offset wm*128
+  const int wn = wave & 3;             // 0..3 -> N offset wn*64
+  const int m0 = blockIdx.y * V4_BM;
+  const int n0 = blockIdx.x * V4_BN;
+  const int r16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int rx = r16 & 7;
+
+  int kt_lo = 0, kt_hi = (K + V4_BK - 1) / V4_BK;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int per = (kt_hi + nz - 1) / nz;
+    kt_lo = blockIdx.z * per;
+    kt_hi = min(kt_hi, kt_lo + per);
+    if (kt_lo >= kt_hi) return;
+  }
+  const int n_kt = kt_hi - kt_lo;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // prologue: A(t0), B(t0), B(t1) fully; then enter the loop which stages
+  // A(t+1) in phases 0-1 and B(t+2) in phases 1-2 of each tile t.
+  v4_stage_slot(A, B, smem, 0, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 1, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 2, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 3, kt_lo, m0, n0, M, N, K, tid);
+  if (n_kt > 1) {
+    v4_stage_slot(A, B, smem, 2, kt_lo + 1, m0, n0, M, N, K, tid);
+    v4_stage_slot(A, B, smem, 3, kt_lo + 1, m0, n0, M, N, K, tid);
+  }
+  asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < n_kt; ++t) {
+    const int kt = kt_lo + t;
+    const int par = kt & 1;
+    const char* Ar = smem + V4_A(par);
+    const char* Br = smem + V4_B(par);
+
+    // B fragments for the whole tile, held in registers
+    bf16x8 bfr[4][2];
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        int row = wn * 64 + nf * 16 + r16;
+        int g = ks * 4 + kgrp;
+        bfr[nf][ks] = *(const bf16x8*)(Br + row * 128 + ((g ^ rx) * 16));
+      }
+
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      // A fragments for accumulator rows 2q, 2q+1
+      bf16x8 afr[2][2];
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          int row = wm * 128 + (q * 2 + m2) * 16 + r16;
+          int g = ks * 4 + kgrp;
+          afr[m2][ks] = *(const bf16x8*)(Ar + row * 128 + ((g ^ rx) * 16));
+        }
+      // stage schedule: q0 -> A half0 (t+1); q1 -> A half1 (t+1) + B half0
+      // (t+2); q2 -> B half1 (t+2)
+      if (q == 0 && t + 1 < n_kt)
+        v4_stage_slot(A, B, smem, 0, kt + 1, m0, n0, M, N, K, tid);
+      if (q == 1) {
+        if (t + 1 < n_kt) v4_stage_slot(A, B, smem, 1, kt + 1, m0, n0, M, N, K, tid);
+        if (t + 2 < n_kt) v4_stage_slot(A, B, smem, 2, kt + 2, m0, n0, M, N, K, tid);
+      }
+      if (q == 2 && t + 2 < n_kt)
+        v4_stage_slot(A, B, smem, 3, kt + 2, m0, n0, M, N, K, tid);
+
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf)
+            acc[q * 2 + m2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[m2][ks], bfr[nf][ks], acc[q * 2 + m2][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      // phase barrier: bound wave skew so the next phase's staging never
+      // overwrites a region a lagging wave still reads.  lgkmcnt(0) flushes
+      // this wave's ds_writes (edge-tile guarded staging) before the barrier.
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    // K-tile boundary: allow the 2 newest slots (4 glds) to stay in flight
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int col = n0 + wn * 64 + nf * 16 + r16;
+      if (col >= N) continue;
+      float bv = 0.0f;
+      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wm * 128 + mf * 16 + kgrp * 4 + r;
+        if (row >= M) continue;
+        float v = acc[mf][nf][r];
+        if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
+        if (EPI == EPI_F32) {
+          if (SPLITK) atomicAdd(&((float*)C)[(long)row * N + col], v);
+          else ((float*)C)[(long)row * N + col] = v;
+        } else {
+          ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
+        }
+      }
+    }
+  }
+}
+
+#define V4_LDS_BYTES 131072
+
+template <int EPI, typename OUT_T>
+static void launch_nt_v4(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
+                         long M, long N, long K, int act, hipStream_t s) {
+  dim3 grid((N + V4_BN - 1) / V4_BN, (M + V4_BM - 1) / V4_BM);
+  auto kern = gemm_nt_v4_kernel<EPI, OUT_T, false>;
+  hipFuncSetAttribute((const void*)kern,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
+  hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
+                     A, B, C, bias, (int)M, (int)N, (int)K, act);
+}
+
+static void launch_nt_v4_splitk_f32(const bf16* A, const bf16* B, float* C,
+                                    long M, long N, long K, hipStream_t s, long z) {
+  dim3 grid((unsigned)((N + V4_BN - 1) / V4_BN),
+            (unsigned)((M + V4_BM - 1) / V4_BM), (unsigned)z);
+  auto kern = gemm_nt_v4_kernel<EPI_F32, float, true>;
+  hipFuncSetAttribute((const void*)kern,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
+  hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
+                     A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
+}
+
 template <int EPI, typename OUT_T>
 static void launch_nt(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
                       long M, long N, long K, int act, hipStream_t s) {
@@ -481,6 +718,12 @@ static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
                      NT_LDS_BYTES, s, A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
 }
 
+// v4 (256^2 pipelined) pays when both tile dims fill 256 rows and the K loop
+// is deep enough to amortize the ring prologue.
+static inline bool use_v4(long M, long N, long K) {
+  return M >= 512 && N >= 256 && K >= 256;
+}
+
 // y[B,N] = act(x[B,K] @ w[N,K]^T + b)   — the hot forward (weights [out,in])
 at::Tensor linear_nt_fwd(at::Tensor x, at::Tensor w, at::Tensor b, long act) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
@@ -489,9 +732,14 @@ at::Tensor linear_nt_fwd(at::Tensor x, at::Tensor w, at::Tensor b, long act) {
   long Mb = x.size(0), K = x.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "shape mismatch x@w^T");
   auto y = at::empty({Mb, N}, x.options());
-  launch_nt<EPI_BIAS_ACT, bf16>((const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
-                                (bf16*)y.data_ptr(), (const bf16*)b.data_ptr(),
-                                Mb, N, K, (int)act, cur_stream());
+  if (use_v4(Mb, N, K))
+    launch_nt_v4<EPI_BIAS_ACT, bf16>((const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+                                     (bf16*)y.data_ptr(), (const bf16*)b.data_ptr(),
+                                     Mb, N, K, (int)act, cur_stream());
+  else
+    launch_nt<EPI_BIAS_ACT, bf16>((const bf16*)x.data_ptr(), (const bf16*)w.data_ptr(),
+                                  (bf16*)y.data_ptr(), (const bf16*)b.data_ptr(),
+                                  Mb, N, K, (int)act, cur_stream());
   return y;
 }
 
@@ -502,8 +750,12 @@ at::Tensor gemm_ntv3_bf16(at::Tensor a, at::Tensor b) {
   long M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
   auto c = at::empty({M, N}, a.options());
-  launch_nt<EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-                             (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+  if (use_v4(M, N, K))
+    launch_nt_v4<EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                                  (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+  else
+    launch_nt<EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                               (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
   return c;
 }
 
@@ -515,8 +767,21 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
   auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
   hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
-  launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-                       (float*)c.data_ptr(), M, N, K, cur_stream());
+  long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
+  long kt4 = (K + V4_BK - 1) / V4_BK;
+  if (M >= 256 && N >= 256 && kt4 >= 4) {
+    long z = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
+                            std::max<long>(kt4 / 4, 1));
+    if (z <= 1)
+      launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                                   (float*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+    else
+      launch_nt_v4_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                              (float*)c.data_ptr(), M, N, K, cur_stream(), z);
+  } else {
+    launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                         (float*)c.data_ptr(), M, N, K, cur_stream());
+  }
   return c;
 }
 

@@ -78,6 +78,30 @@ def test_gemm_ntv3(M, N, K):
     assert ok, f"gemm_ntv3 {M}x{N}x{K} maxdiff={err}"
 
 
+@pytest.mark.parametrize("M,N,K", [(512, 256, 256), (512, 512, 1864),
+                                   (4096, 4096, 4096), (8192, 256, 512),
+                                   (600, 300, 777)])
+def test_gemm_nt_v4_shapes(M, N, K):
+    """256x256 deep-pipelined kernel (v4 route), incl. K-edge + M/N-edge tiles."""
+    a = _rand_bf16(M, K, seed=M + 11)
+    b = _rand_bf16(N, K, seed=N + 12)
+    c = hip_ops().gemm_ntv3_bf16(a, b)
+    ok, err = _rel_close(c, a.float() @ b.float().t())
+    assert ok, f"v4 {M}x{N}x{K} maxdiff={err}"
+
+
+def test_gemm_nt_v4_race_screen():
+    """The v4 pipeline uses counted vmcnt across raw barriers — a wrong count
+    is a data race.  Non-splitk runs must be bitwise-identical across
+    repeats."""
+    a = _rand_bf16(1024, 512, seed=90)
+    b = _rand_bf16(512, 512, seed=91)
+    first = hip_ops().gemm_ntv3_bf16(a, b)
+    for _ in range(5):
+        again = hip_ops().gemm_ntv3_bf16(a, b)
+        assert torch.equal(first, again), "v4 nondeterminism: race in pipeline"
+
+
 def test_gemm_ntv3_f32_splitk():
     a = _rand_bf16(1024, 8192, seed=50, scale=0.5)   # dzT [N,B]
     b = _rand_bf16(1864, 8192, seed=51, scale=0.5)   # xT [K,B]
