@@ -585,7 +585,12 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     v4_stage_slot(A, B, smem, 2, kt_lo + 1, m0, n0, M, N, K, tid);
     v4_stage_slot(A, B, smem, 3, kt_lo + 1, m0, n0, M, N, K, tid);
   }
-  asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+  {
+    bool b1_glds = (n_kt > 1) && ((kt_lo + 1) * V4_BK + V4_BK <= K) &&
+                   (n0 + V4_BN <= N);
+    if (b1_glds) asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+    else asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  }
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < n_kt; ++t) {
@@ -644,8 +649,16 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
-    // K-tile boundary: allow the 2 newest slots (4 glds) to stay in flight
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    // K-tile boundary.  The only loads allowed to stay in flight are the
+    // B(t+2) slots (4 glds) — and only when they were actually staged by
+    // glds; with a guarded/absent B(t+2), everything must drain or the
+    // A(t+1) glds could outlive this wait and be read unlanded next tile.
+    {
+      bool b2_glds = (t + 2 < n_kt) && ((kt + 2) * V4_BK + V4_BK <= K) &&
+                     (n0 + V4_BN <= N);
+      if (b2_glds) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
     __builtin_amdgcn_s_barrier();
   }
 
@@ -721,7 +734,10 @@ static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
 // v4 (256^2 pipelined) pays when both tile dims fill 256 rows and the K loop
 // is deep enough to amortize the ring prologue.
 static inline bool use_v4(long M, long N, long K) {
-  return M >= 512 && N >= 256 && K >= 256;
+  // v4 runs 1 block/CU (128 KB LDS); below ~200 blocks it leaves CUs idle
+  // and the 128^2-tile v3 (2-3 blocks/CU) wins.
+  long blocks = ((M + V4_BM - 1) / V4_BM) * ((N + V4_BN - 1) / V4_BN);
+  return M >= 512 && N >= 256 && K >= 256 && blocks >= 200;
 }
 
 // y[B,N] = act(x[B,K] @ w[N,K]^T + b)   — the hot forward (weights [out,in])
@@ -769,7 +785,8 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
   long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
   long kt4 = (K + V4_BK - 1) / V4_BK;
-  if (M >= 256 && N >= 256 && kt4 >= 4) {
+  if (M >= 256 && N >= 256 && kt4 >= 4 &&
+      gx4 * gy4 * std::max<long>(kt4 / 4, 1) >= 200) {
     long z = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
                             std::max<long>(kt4 / 4, 1));
     if (z <= 1)
