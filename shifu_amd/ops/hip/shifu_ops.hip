@@ -785,10 +785,12 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
   long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
   long kt4 = (K + V4_BK - 1) / V4_BK;
-  if (M >= 256 && N >= 256 && kt4 >= 4 &&
-      gx4 * gy4 * std::max<long>(kt4 / 4, 1) >= 200) {
-    long z = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
-                            std::max<long>(kt4 / 4, 1));
+  // v4 split-K needs >=8 K-tiles per part to amortize its ring prologue,
+  // and >=200 blocks to fill the chip at 1 block/CU.
+  long z4 = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
+                           std::max<long>(kt4 / 8, 1));
+  if (M >= 256 && N >= 256 && kt4 >= 8 && gx4 * gy4 * z4 >= 200) {
+    long z = z4;
     if (z <= 1)
       launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                                    (float*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
@@ -1437,7 +1439,7 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                        (const bf16*)vals.data_ptr(), n, D);
   }
-  auto rowscale = at::empty({n}, vals.options());
+  auto rowscale = at::empty({n}, vals.options().dtype(at::kFloat));
   hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
                      (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                      (float*)rowscale.data_ptr(), n, (float)eps);
