@@ -1004,8 +1004,12 @@ std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
   long B = dy.size(0), N = dy.size(1);
   auto dz = at::empty_like(dy);
   auto db = at::zeros({N}, dy.options().dtype(at::kFloat));
-  long gx, chunks, rpc;
-  colsum_grid(B, (N + 7) / 8, gx, chunks, rpc);
+  // 8 columns per thread -> few column-blocks; scale row-chunks so total
+  // threads stay ~10x CU count (HBM saturation needs >=64k threads)
+  long gx = ((N + 7) / 8 + 255) / 256;
+  long chunks = std::min<long>(std::max<long>(768 / gx, 1),
+                               std::max<long>(B / 16, 1));
+  long rpc = (B + chunks - 1) / chunks;
   hipLaunchKernelGGL(act_grad_colsum_kernel, dim3((unsigned)gx, (unsigned)chunks),
                      dim3(256), 0, cur_stream(),
                      (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
