@@ -63,6 +63,7 @@ class RunConfig:
     # -- misc --
     log_dir: str = "./logs"
     device: str = "auto"                 # "cuda" | "cpu" | "auto"
+    enable_trace: bool = False           # per-phase step timing (utils/trace.py)
 
     def resolved_backend(self) -> str:
         if self.backend != "auto":

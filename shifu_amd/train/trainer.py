@@ -132,18 +132,27 @@ class Trainer:
         self.global_step = 0
         self.start_epoch = 0
         self._rng = np.random.default_rng(rc.seed + rank)
+        from shifu_amd.utils.trace import StepTracer
+        self.tracer = StepTracer(enabled=rc.enable_trace,
+                                 use_gpu_events=(self.device.type == "cuda"))
 
     # ------------------------------------------------------------------ steps
     def train_step(self, batch: DeviceData, sync: bool = True) -> float:
         """One forward/backward/(all-reduce)/update step; returns loss."""
+        tr = self.tracer
         self.aggregator.set_sync(sync)
-        logits = self.model(batch.dense, batch.cats)
-        loss = weighted_loss(logits, batch.target, batch.weight, self.loss_kind)
-        loss.backward()
+        with tr.phase("fwd"):
+            logits = self.model(batch.dense, batch.cats)
+        with tr.phase("loss"):
+            loss = weighted_loss(logits, batch.target, batch.weight, self.loss_kind)
+        with tr.phase("bwd"):
+            loss.backward()
         if sync:
-            self.aggregator.finish()
-            self.optimizer.step()
-            self.optimizer.zero_grad()
+            with tr.phase("comm"):
+                self.aggregator.finish()
+            with tr.phase("opt"):
+                self.optimizer.step()
+                self.optimizer.zero_grad()
         self.global_step += 1
         return float(loss.detach())
 
@@ -230,6 +239,11 @@ class Trainer:
                                      self.global_step, self.model, self.optimizer)
             if is_distributed():
                 torch.distributed.barrier()
+        if self.tracer.enabled:
+            self.tracer.export_chrome_trace(
+                os.path.join(self.rc.log_dir, f"trace-rank{self.rank}.json"))
+            if self.is_chief:
+                print(self.tracer.summary_line(), flush=True)
         if self.is_chief:
             export_model(self.model, self.rc.final_model_path,
                          model_name=self.mc.model_name, algorithm=self.mc.algorithm,
