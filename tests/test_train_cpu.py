@@ -172,3 +172,18 @@ def test_auc_score_sanity():
     assert auc_score(np.array([0.1, 0.2, 0.8, 0.9]), labels) == 1.0
     assert auc_score(np.array([0.9, 0.8, 0.2, 0.1]), labels) == 0.0
     assert abs(auc_score(np.array([0.5, 0.5, 0.5, 0.5]), labels) - 0.5) < 1e-9
+
+
+def test_per_epoch_aggregation_semantics(tmp_path):
+    """The reference's headline mode applies ONE global update per epoch
+    (SyncReplicasOptimizer with replicas_to_aggregate = total_rows/batch,
+    ssgd_monitor.py:139-140).  update_window >= steps/epoch reproduces it:
+    exactly one optimizer step per epoch."""
+    train, valid = _data(n=640)
+    mc = _mc(epochs=3, batch=64)
+    mc.params.update_window = 10 ** 6
+    rc = RunConfig(tmp_model_path=str(tmp_path / "ckpt"),
+                   final_model_path=str(tmp_path / "final"))
+    tr = Trainer(ShifuMLP(10, [16, 8], ["relu", "relu"]), mc, rc, train, valid)
+    tr.fit()
+    assert tr.optimizer.step_count == 3  # one aggregated update per epoch
