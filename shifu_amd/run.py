@@ -16,7 +16,7 @@ import torch
 from shifu_amd.config.model_config import ColumnConfig, ModelConfig
 from shifu_amd.config.run_config import RunConfig
 from shifu_amd.data.csv_loader import list_training_files, load_csv_files
-from shifu_amd.data.sharding import shard_rows
+from shifu_amd.data.sharding import shard_files, shard_rows
 from shifu_amd.models.mlp import build_model
 from shifu_amd.parallel.dist import destroy_distributed, init_distributed
 from shifu_amd.parallel.launcher import Launcher
@@ -30,16 +30,28 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
     try:
         files = list_training_files(rc.training_data_path)
         from shifu_amd.io import load_csv_native
-        full = load_csv_native(files, rc.selected_numeric_columns,
-                               rc.selected_categorical_columns,
-                               rc.target_column, rc.weight_column, rc.delimiter)
-        # deterministic split BEFORE sharding so every rank agrees on the
-        # valid set (reference splits per-worker after file sharding; a
-        # shared-seed global split keeps valid metrics comparable)
-        train, valid = full.split(rc.valid_set_rate, seed=rc.seed)
         import numpy as np
-        s, e = shard_rows(len(train), rank, world)
-        train = train.subset(np.arange(s, e))
+        if world > 1 and len(files) >= world:
+            # reference behavior (TrainingDataSet.java:55-89): round-robin
+            # FILES across workers, each rank parses only its shard — the
+            # 100M-row config never parses the full dataset per rank.
+            # Valid split is then per-rank (seeded per rank, like the
+            # reference's per-worker random split).
+            my_files = shard_files(files, rank, world)
+            shard = load_csv_native(my_files, rc.selected_numeric_columns,
+                                    rc.selected_categorical_columns,
+                                    rc.target_column, rc.weight_column, rc.delimiter)
+            train, valid = shard.split(rc.valid_set_rate, seed=rc.seed + rank)
+        else:
+            # few big files: every rank parses, then row-range shards; a
+            # shared-seed global split keeps valid metrics identical across
+            # ranks
+            full = load_csv_native(files, rc.selected_numeric_columns,
+                                   rc.selected_categorical_columns,
+                                   rc.target_column, rc.weight_column, rc.delimiter)
+            train, valid = full.split(rc.valid_set_rate, seed=rc.seed)
+            s, e = shard_rows(len(train), rank, world)
+            train = train.subset(np.arange(s, e))
 
         device = torch.device(rc.resolved_device(),
                               rank % max(torch.cuda.device_count(), 1)
