@@ -52,9 +52,8 @@ class DeepFM(torch.nn.Module):
         fm1 = self.fm_first(cats).sum(dim=1) + self.fm_dense(dense).reshape(-1)
 
         emb_flat = self.embeddings(cats)                   # [B, F*D]
-        v = emb_flat.reshape(B, F, self.embed_dim).float()
-        sum_v = v.sum(dim=1)                               # [B, D]
-        fm2 = 0.5 * (sum_v * sum_v - (v * v).sum(dim=1)).sum(dim=1)  # [B]
+        from shifu_amd.ops.fm import fm_second_order
+        fm2 = fm_second_order(emb_flat, F, self.embed_dim)  # [B] (fused on GPU)
 
         x = torch.cat([dense, emb_flat.to(dense.dtype)], dim=1)
         for layer in self.tower:

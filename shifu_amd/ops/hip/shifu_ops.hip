@@ -1533,6 +1533,73 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
 }
 
 // ---------------------------------------------------------------------------
+// DeepFM second-order interaction (fused):
+//   fm2[b] = 0.5 * sum_d [ (sum_f v[b,f,d])^2 - sum_f v[b,f,d]^2 ]
+// fwd: one wave per row b, lane d accumulates over F; saves s[b,d]=sum_f v
+// bwd: dv[b,f,d] = (s[b,d] - v[b,f,d]) * dout[b]
+// ---------------------------------------------------------------------------
+__global__ void fm2_fwd_kernel(const bf16* __restrict__ emb, float* __restrict__ fm2,
+                               float* __restrict__ s, long B, long F, long D) {
+  long b = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (b >= B) return;
+  int lane = threadIdx.x & 63;
+  float acc = 0.0f;
+  for (long d = lane; d < D; d += 64) {
+    float sum = 0.0f, sq = 0.0f;
+    for (long f = 0; f < F; ++f) {
+      float v = __bfloat162float(emb[b * F * D + f * D + d]);
+      sum += v;
+      sq += v * v;
+    }
+    s[b * D + d] = sum;
+    acc += sum * sum - sq;
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  if (lane == 0) fm2[b] = 0.5f * acc;
+}
+
+__global__ void fm2_bwd_kernel(const bf16* __restrict__ emb, const float* __restrict__ s,
+                               const float* __restrict__ dout, bf16* __restrict__ demb,
+                               long B, long F, long D) {
+  long total = B * F * D;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    long b = t / (F * D);
+    long d = t % D;
+    float v = __bfloat162float(emb[t]);
+    demb[t] = __float2bfloat16((s[b * D + d] - v) * dout[b]);
+  }
+}
+
+std::vector<at::Tensor> fm2_fwd(at::Tensor emb, long F, long D) {
+  CHECK_GPU(emb); CHECK_CONTIG(emb); CHECK_BF16(emb);
+  long B = emb.size(0);
+  auto fm2 = at::empty({B}, emb.options().dtype(at::kFloat));
+  auto sum = at::empty({B, D}, emb.options().dtype(at::kFloat));
+  int wpb = 4;
+  hipLaunchKernelGGL(fm2_fwd_kernel, dim3((unsigned)((B + wpb - 1) / wpb)),
+                     dim3(64 * wpb), 0, cur_stream(),
+                     (const bf16*)emb.data_ptr(), (float*)fm2.data_ptr(),
+                     (float*)sum.data_ptr(), B, F, D);
+  return {fm2, sum};
+}
+
+at::Tensor fm2_bwd(at::Tensor emb, at::Tensor s, at::Tensor dout, long F, long D) {
+  CHECK_GPU(emb); CHECK_CONTIG(emb); CHECK_BF16(emb);
+  CHECK_F32(s); CHECK_F32(dout);
+  long B = emb.size(0);
+  auto demb = at::empty_like(emb);
+  long total = B * F * D;
+  hipLaunchKernelGGL(fm2_bwd_kernel, dim3(scat_blocks(total)), dim3(256), 0, cur_stream(),
+                     (const bf16*)emb.data_ptr(), (const float*)s.data_ptr(),
+                     (const float*)dout.contiguous().data_ptr(),
+                     (bf16*)demb.data_ptr(), B, F, D);
+  return demb;
+}
+
+// ---------------------------------------------------------------------------
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "shifu_amd CDNA4 (gfx950) kernels";
   m.def("linear_act_fwd", &linear_act_fwd, "fused GEMM+bias+act forward (bf16)");
@@ -1545,6 +1612,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose_bf16", &transpose_bf16);
   m.def("gemv_fwd", &gemv_fwd);
   m.def("gemv_bwd", &gemv_bwd);
+  m.def("fm2_fwd", &fm2_fwd);
+  m.def("fm2_bwd", &fm2_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);

@@ -175,3 +175,20 @@ def test_emb_optimizer_sgd_rowwise():
     assert torch.allclose(emb.arena.data[3], before[3] - 0.1 * 2.0 * torch.ones(2))
     assert torch.allclose(emb.arena.data[7], before[7] - 0.1 * 1.0 * torch.ones(2))
     assert torch.equal(emb.arena.data[0], before[0])
+
+
+def test_fm_second_order_matches_autograd():
+    from shifu_amd.ops.fm import fm_second_order
+    torch.manual_seed(0)
+    B, F, D = 8, 3, 4
+    e1 = torch.randn(B, F * D, requires_grad=True)
+    out = fm_second_order(e1, F, D)
+    out.pow(2).sum().backward()
+
+    e2 = e1.detach().clone().requires_grad_(True)
+    v = e2.reshape(B, F, D)
+    s = v.sum(dim=1)
+    ref = 0.5 * (s * s - (v * v).sum(dim=1)).sum(dim=1)
+    ref.pow(2).sum().backward()
+    assert torch.allclose(out, ref, atol=1e-5)
+    assert torch.allclose(e1.grad, e2.grad, atol=1e-5)
