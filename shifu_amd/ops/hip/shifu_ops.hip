@@ -720,11 +720,20 @@ static void launch_nt(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
                      NT_LDS_BYTES, s, A, B, C, bias, (int)M, (int)N, (int)K, act);
 }
 
+static long splitk_target_blocks() {
+  static long t = [] {
+    const char* e = getenv("SHIFU_SPLITK_TARGET");
+    return e ? atol(e) : 512L;
+  }();
+  return t;
+}
+
 static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
                                  long M, long N, long K, hipStream_t s) {
   long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
   long max_z = (K + NT_BK - 1) / NT_BK;
-  long z = std::min<long>(std::max<long>(512 / std::max<long>(gx * gy, 1), 1), max_z);
+  long z = std::min<long>(
+      std::max<long>(splitk_target_blocks() / std::max<long>(gx * gy, 1), 1), max_z);
   if (z <= 1) { launch_nt<EPI_F32, float>(A, B, C, nullptr, M, N, K, 0, s); return; }
   dim3 grid((unsigned)gx, (unsigned)gy, (unsigned)z);
   hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>), grid, dim3(256),
