@@ -87,7 +87,9 @@ def main():
                 p.data = p.data.to(torch.bfloat16)
 
     dense_params, emb_params = split_params(model)
-    flat = FlatParams(dense_params)
+    from shifu_amd.ops.flat import bind_mirrors
+    flat = FlatParams(dense_params, mirror_bf16=on_gpu)
+    bind_mirrors(model, flat)
     agg = GradAggregator(flat, emb_params, bucket_mb=args.bucket_mb, overlap=True)
     opt = FusedOptimizer(flat, emb_params, optimizer="adam", lr=1e-3,
                          l2_reg=0.0, emb_optimizer="adagrad", emb_lr=0.01)

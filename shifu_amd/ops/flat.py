@@ -35,8 +35,9 @@ class FlatParams:
     """Flattens a list of same-dtype parameters into one buffer and rebinds
     each parameter's .data (and .grad) to a view of it."""
 
-    def __init__(self, params: List[torch.nn.Parameter]):
+    def __init__(self, params: List[torch.nn.Parameter], mirror_bf16: bool = False):
         self.params = list(params)
+        self.mirror = None
         if not self.params:
             self.flat = torch.zeros(0)
             self.flat_grad = torch.zeros(0)
@@ -59,6 +60,22 @@ class FlatParams:
             p.grad = self.flat_grad[off:off + n].view_as(p.data)
             self._offsets.append((off, n))
             off += n
+        if mirror_bf16:
+            # ONE bf16 compute copy of the whole arena, refreshed by a single
+            # cast kernel per step (vs a small cast kernel per layer) — the
+            # fused-linear forward reads per-param views of it.
+            self.mirror = self.flat.to(torch.bfloat16)
+
+    def refresh_mirror(self) -> None:
+        if self.mirror is not None:
+            self.mirror.copy_(self.flat)
+
+    def mirror_view(self, p: torch.nn.Parameter):
+        if self.mirror is None:
+            return None
+        i = self.params.index(p)
+        off, n = self._offsets[i]
+        return self.mirror[off:off + n].view_as(p.data)
 
     def numel(self) -> int:
         return self.flat.numel()
@@ -84,3 +101,16 @@ class FlatParams:
     def grad_views(self) -> Iterable[torch.Tensor]:
         for off, n in self._offsets:
             yield self.flat_grad[off:off + n]
+
+
+def bind_mirrors(module: torch.nn.Module, flat: "FlatParams") -> None:
+    """Attach per-layer bf16 mirror views to every FusedLinear whose params
+    live in this arena (no-op without a mirror)."""
+    if flat.mirror is None:
+        return
+    from shifu_amd.ops.linear import FusedLinear
+    pset = {id(p): i for i, p in enumerate(flat.params)}
+    for m in module.modules():
+        if isinstance(m, FusedLinear) and id(m.weight) in pset and id(m.bias) in pset:
+            m._w_mirror = flat.mirror_view(m.weight)
+            m._b_mirror = flat.mirror_view(m.bias)

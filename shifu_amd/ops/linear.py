@@ -66,12 +66,15 @@ def _act_grad_from_y_ref(dy: torch.Tensor, y: torch.Tensor, act: int) -> torch.T
 
 class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor, act: int):
-        # w: [out, in]
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor, act: int,
+                w_mirror=None, b_mirror=None):
+        # w: [out, in].  w_mirror/b_mirror: bf16 views of the flat arena's
+        # compute copy (refreshed once per step by the optimizer) — when
+        # absent, cast per call.
         if use_hip(x):
             ext = hip_ops()
-            wb = w.to(torch.bfloat16)
-            bb = b.to(torch.bfloat16)
+            wb = (w_mirror if w_mirror is not None else w.to(torch.bfloat16)).contiguous()
+            bb = (b_mirror if b_mirror is not None else b.to(torch.bfloat16)).contiguous()
             xb = x if x.dtype == torch.bfloat16 else x.to(torch.bfloat16)
             xb = xb.contiguous()
             if w.shape[0] == 1:   # 1-unit head: GEMV path (memory-speed)
@@ -99,7 +102,7 @@ class _FusedLinearFn(torch.autograd.Function):
                 dw, db, dx = ext.gemv_bwd(x, w.reshape(-1), dz, ctx.x_needs_grad)
                 if not ctx.x_needs_grad:
                     dx = None
-                return dx, dw, db, None
+                return dx, dw, db, None, None, None
             # one fused pass: dz = dy*act'(y) and db = colsum(dz)
             dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
             xT = ext.transpose_bf16(x)                # [K,B]
@@ -114,7 +117,7 @@ class _FusedLinearFn(torch.autograd.Function):
             dw = dz.t() @ x                           # [N,K]
             db = dz.sum(dim=0)
             dx = dz @ w if ctx.x_needs_grad else None
-        return dx, dw, db, None
+        return dx, dw, db, None, None, None
 
 
 def fused_linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
@@ -148,7 +151,9 @@ class FusedLinear(torch.nn.Module):
         self.bias = torch.nn.Parameter(torch.zeros(out_features))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return _FusedLinearFn.apply(x, self.weight, self.bias, self._act)
+        return _FusedLinearFn.apply(x, self.weight, self.bias, self._act,
+                                    getattr(self, "_w_mirror", None),
+                                    getattr(self, "_b_mirror", None))
 
     def extra_repr(self) -> str:
         return f"in={self.in_features}, out={self.out_features}, act={self.activation}"

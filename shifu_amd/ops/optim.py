@@ -138,6 +138,7 @@ class FusedOptimizer:
                 self._dense_step_hip()
             else:
                 self._dense_step_ref()
+            self.flat.refresh_mirror()   # one arena-wide bf16 cast per step
         for i, p in enumerate(self.emb_params):
             self._emb_step(p, i)
 

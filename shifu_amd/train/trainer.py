@@ -116,7 +116,10 @@ class Trainer:
         self.valid_data = DeviceData.from_dataset(valid_data, self.device, dense_dtype)
 
         dense_params, emb_params = split_params(self.model)
-        self.flat = FlatParams(dense_params)
+        from shifu_amd.ops.flat import bind_mirrors
+        self.flat = FlatParams(dense_params,
+                               mirror_bf16=(dense_dtype == torch.bfloat16))
+        bind_mirrors(self.model, self.flat)
         self.emb_params = emb_params
         self.aggregator = GradAggregator(self.flat, emb_params,
                                          bucket_mb=rc.bucket_mb,
@@ -223,6 +226,7 @@ class Trainer:
         path = ckpt.latest_checkpoint(self.rc.tmp_model_path)
         if path:
             info = ckpt.load_checkpoint(path, self.model, self.optimizer, self.device)
+            self.flat.refresh_mirror()
             self.start_epoch = int(info["epoch"]) + 1
             self.global_step = int(info["global_step"])
 
