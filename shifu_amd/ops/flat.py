@@ -73,7 +73,7 @@ class FlatParams:
     def mirror_view(self, p: torch.nn.Parameter):
         if self.mirror is None:
             return None
-        i = self.params.index(p)
+        i = next(i for i, q in enumerate(self.params) if q is p)
         off, n = self._offsets[i]
         return self.mirror[off:off + n].view_as(p.data)
 
