@@ -743,6 +743,11 @@ static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
 // v4 (256^2 pipelined) pays when both tile dims fill 256 rows and the K loop
 // is deep enough to amortize the ring prologue.
 static inline bool use_v4(long M, long N, long K) {
+  static int disable = [] {
+    const char* e = getenv("SHIFU_DISABLE_V4");
+    return e ? atoi(e) : 0;
+  }();
+  if (disable) return false;
   // v4 runs 1 block/CU (128 KB LDS); below ~200 blocks it leaves CUs idle
   // and the 128^2-tile v3 (2-3 blocks/CU) wins.
   long blocks = ((M + V4_BM - 1) / V4_BM) * ((N + V4_BN - 1) / V4_BN);
@@ -1365,19 +1370,27 @@ __global__ void emb_accsq_small_kernel(float* __restrict__ acc, const long* __re
   }
 }
 
-// phase 1 of adagrad: acc[rows[i]] += mean_d vals[i,d]^2  (wave per entry)
+// phase 1 of adagrad: acc[rows[i]] += mean_d vals[i,d]^2.
+// A wave covers TWO entries (lanes 0-31 / 32-63), each lane loading a bf16
+// PAIR (4 B) -> 256 B coalesced per wave access (a lane-per-element layout
+// read only 2 B/lane: 0.56 TB/s measured; this is the fix).  Requires even D.
 __global__ void emb_accsq_kernel(float* __restrict__ acc, const long* __restrict__ rows,
                                  const bf16* __restrict__ vals, long n, long D) {
-  long e = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  long e = (long)blockIdx.x * (blockDim.x >> 5) + (threadIdx.x >> 5);
   if (e >= n) return;
-  int lane = threadIdx.x & 63;
+  int lane = threadIdx.x & 31;
+  long pairs = D >> 1;
   float sq = 0.0f;
-  for (long d = lane; d < D; d += 64) {
-    float g = __bfloat162float(vals[e * D + d]);
-    sq += g * g;
+  const unsigned* v2 = (const unsigned*)(vals + e * D);
+  for (long p = lane; p < pairs; p += 32) {
+    unsigned u = v2[p];
+    unsigned short lo = (unsigned short)(u & 0xffff), hi = (unsigned short)(u >> 16);
+    float a = __bfloat162float(*(const bf16*)&lo);
+    float b = __bfloat162float(*(const bf16*)&hi);
+    sq += a * a + b * b;
   }
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) sq += __shfl_down(sq, off, 64);
+  for (int off = 16; off > 0; off >>= 1) sq += __shfl_down(sq, off, 32);
   if (lane == 0) atomicAdd(&acc[rows[e]], sq / (float)D);
 }
 
@@ -1410,14 +1423,14 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
   long n = rows.numel(), D = arena.size(1);
   if (!n) return;
   auto s = cur_stream();
-  if (D <= 8) {
+  if (D <= 8 || (D & 1)) {
     hipLaunchKernelGGL(emb_accsq_small_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                        (const bf16*)vals.data_ptr(), n, D);
   } else {
-    int wpb = 4;
-    hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + wpb - 1) / wpb)),
-                       dim3(64 * wpb), 0, s,
+    int epb = 8;  // 256 threads = 8 waves-of-32 lanes... 8 half-waves/block
+    hipLaunchKernelGGL(emb_accsq_kernel, dim3((unsigned)((n + epb - 1) / epb)),
+                       dim3(256), 0, s,
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                        (const bf16*)vals.data_ptr(), n, D);
   }

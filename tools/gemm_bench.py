@@ -20,6 +20,8 @@ SHAPES = [
     ("l1_dgrd3", 8192, 1864, 1024, "v3"),
     ("l1_wgrad", 1864, 1024, 8192, "tn"),
     ("l1_wgrd3", 1024, 1864, 8192, "v3f"),
+    ("l1f16k", 16384, 1024, 1864, "v3"),
+    ("l1d16k", 16384, 1864, 1024, "v3"),
     ("sq4096", 4096, 4096, 4096, "nn"),
     ("sq4096v3", 4096, 4096, 4096, "v3"),
 ]
