@@ -223,14 +223,22 @@ void gemm_tile_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 
   // ---- epilogue: D row = (lane>>4)*4 + r, col = lane&15 (per 16x16 frag) ----
+  // bias hoisted: this lane's column set is 4 values, not 16
+  float bvs[4] = {0.f, 0.f, 0.f, 0.f};
+  if (EPI == EPI_BIAS_ACT) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+      int col = n0 + wc + fj * 16 + r16;
+      if (col < N) bvs[fj] = __bfloat162float(bias[col]);
+    }
+  }
 #pragma unroll
   for (int fi = 0; fi < 4; ++fi) {
 #pragma unroll
     for (int fj = 0; fj < 4; ++fj) {
       int col = n0 + wc + fj * 16 + r16;
       if (col >= N) continue;
-      float bv = 0.0f;
-      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+      float bv = bvs[fj];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr + fi * 16 + kgrp * 4 + r;
@@ -437,14 +445,22 @@ void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     __syncthreads();
   }
 
+  // bias hoisted: this lane's column set is 4 values, not 16
+  float bvs[4] = {0.f, 0.f, 0.f, 0.f};
+  if (EPI == EPI_BIAS_ACT) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+      int col = n0 + wc + fj * 16 + r16;
+      if (col < N) bvs[fj] = __bfloat162float(bias[col]);
+    }
+  }
 #pragma unroll
   for (int fi = 0; fi < 4; ++fi) {
 #pragma unroll
     for (int fj = 0; fj < 4; ++fj) {
       int col = n0 + wc + fj * 16 + r16;
       if (col >= N) continue;
-      float bv = 0.0f;
-      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+      float bv = bvs[fj];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr + fi * 16 + kgrp * 4 + r;
@@ -662,15 +678,22 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
     __builtin_amdgcn_s_barrier();
   }
 
-  // epilogue
+  // epilogue (bias hoisted: 4 loads per lane, not 32)
+  float bvs[4] = {0.f, 0.f, 0.f, 0.f};
+  if (EPI == EPI_BIAS_ACT) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int col = n0 + wn * 64 + nf * 16 + r16;
+      if (col < N) bvs[nf] = __bfloat162float(bias[col]);
+    }
+  }
 #pragma unroll
   for (int mf = 0; mf < 8; ++mf) {
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       int col = n0 + wn * 64 + nf * 16 + r16;
       if (col >= N) continue;
-      float bv = 0.0f;
-      if (EPI == EPI_BIAS_ACT) bv = __bfloat162float(bias[col]);
+      float bv = bvs[nf];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wm * 128 + mf * 16 + kgrp * 4 + r;
