@@ -20,6 +20,7 @@ from __future__ import annotations
 
 import os
 import queue
+import sys
 import time
 import traceback
 from collections import defaultdict
@@ -39,6 +40,15 @@ MSG_METRIC, MSG_HEARTBEAT, MSG_DONE, MSG_ERROR = "metric", "hb", "done", "error"
 def _rank_main(rank: int, world: int, rc_dict: dict, mc_dict: dict,
                entry: Callable, q) -> None:
     """Child process body: set env, run the user entry(rank, world, rc, mc, sink)."""
+    # per-rank log files (successor of YARN container logs)
+    log_dir = rc_dict.get("log_dir") or "./logs"
+    try:
+        os.makedirs(log_dir, exist_ok=True)
+        logf = open(os.path.join(log_dir, f"rank-{rank}.log"), "a", buffering=1)
+        sys.stdout = logf
+        sys.stderr = logf
+    except OSError:
+        pass
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
     os.environ["LOCAL_RANK"] = str(rank)

@@ -36,7 +36,8 @@ def latest_checkpoint(ckpt_dir: str) -> Optional[str]:
 
 
 def save_checkpoint(ckpt_dir: str, epoch: int, global_step: int,
-                    model: torch.nn.Module, optimizer, keep_last: int = 3) -> str:
+                    model: torch.nn.Module, optimizer, keep_last: int = 3,
+                    extra: Optional[dict] = None) -> str:
     os.makedirs(ckpt_dir, exist_ok=True)
     path = checkpoint_path(ckpt_dir, epoch)
     tmp = path + ".tmp"
@@ -46,6 +47,7 @@ def save_checkpoint(ckpt_dir: str, epoch: int, global_step: int,
         "model": {k: v.cpu() for k, v in model.state_dict().items()},
         "optimizer": _optim_state_cpu(optimizer),
         "torch_rng": torch.get_rng_state(),
+        "extra": extra or {},
     }, tmp)
     os.replace(tmp, path)  # atomic publish — a crashed writer never corrupts
     # prune old checkpoints
@@ -82,4 +84,5 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
                           ({kk: tt.to(device) for kk, tt in v.items()} if isinstance(v, dict) else v))
                       for k, v in opt_sd.items()}
         optimizer.load_state_dict(opt_sd)
-    return {"epoch": blob["epoch"], "global_step": blob["global_step"]}
+    return {"epoch": blob["epoch"], "global_step": blob["global_step"],
+            "extra": blob.get("extra", {})}

@@ -229,6 +229,9 @@ class Trainer:
             self.flat.refresh_mirror()
             self.start_epoch = int(info["epoch"]) + 1
             self.global_step = int(info["global_step"])
+            np_rng = (info.get("extra") or {}).get("np_rng")
+            if np_rng is not None and self.rank == 0:
+                self._rng.bit_generator.state = np_rng
 
     def fit(self) -> List[TrainingIntermediateResult]:
         self.maybe_resume()
@@ -240,7 +243,8 @@ class Trainer:
                 self.metric_sink(r)
             if self.is_chief and (epoch + 1) % self.rc.checkpoint_every_epochs == 0:
                 ckpt.save_checkpoint(self.rc.tmp_model_path, epoch,
-                                     self.global_step, self.model, self.optimizer)
+                                     self.global_step, self.model, self.optimizer,
+                                     extra={"np_rng": self._rng.bit_generator.state})
             if is_distributed():
                 torch.distributed.barrier()
         if self.tracer.enabled:
