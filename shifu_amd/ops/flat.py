@@ -77,6 +77,11 @@ class FlatParams:
         off, n = self._offsets[i]
         return self.mirror[off:off + n].view_as(p.data)
 
+    def grad_view(self, p: torch.nn.Parameter):
+        i = next(i for i, q in enumerate(self.params) if q is p)
+        off, n = self._offsets[i]
+        return self.flat_grad[off:off + n].view_as(p.data)
+
     def numel(self) -> int:
         return self.flat.numel()
 
@@ -114,3 +119,6 @@ def bind_mirrors(module: torch.nn.Module, flat: "FlatParams") -> None:
         if isinstance(m, FusedLinear) and id(m.weight) in pset and id(m.bias) in pset:
             m._w_mirror = flat.mirror_view(m.weight)
             m._b_mirror = flat.mirror_view(m.bias)
+            if m.out_features > 1:  # heads keep the autograd GEMV path
+                m._w_gradview = flat.grad_view(m.weight)
+                m._b_gradview = flat.grad_view(m.bias)

@@ -22,11 +22,41 @@ activations are monotone with act(z)>0 <=> z>0 for (leaky)relu.
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
 
 from shifu_amd.ops.dispatch import use_hip, hip_ops
+
+# ---------------------------------------------------------------------------
+# async wgrad: weight/bias gradients are computed on a side HIP stream and
+# accumulated DIRECTLY into the flat-grad arena views, overlapping the
+# dgrad chain (and, multi-GPU, the all-reduce) that continues on the main
+# stream.  autograd then receives None for w/b (no AccumulateGrad work).
+# Consumers (optimizer / aggregator) call drain_wgrad_events() before
+# touching the flat grads.
+# ---------------------------------------------------------------------------
+_ASYNC_WGRAD = os.environ.get("SHIFU_ASYNC_WGRAD", "1") == "1"
+_WGRAD_STREAM = None
+_WGRAD_EVENTS: list = []
+
+
+def _wgrad_stream():
+    global _WGRAD_STREAM
+    if _WGRAD_STREAM is None:
+        _WGRAD_STREAM = torch.cuda.Stream()
+    return _WGRAD_STREAM
+
+
+def drain_wgrad_events(stream=None) -> None:
+    """Make `stream` (default: current) wait for all pending async wgrads."""
+    if not _WGRAD_EVENTS:
+        return
+    s = stream or torch.cuda.current_stream()
+    for ev in _WGRAD_EVENTS:
+        s.wait_event(ev)
+    _WGRAD_EVENTS.clear()
 
 # activation ids shared with the HIP side (ops/hip/shifu_ops.hip)
 ACT_NONE, ACT_SIGMOID, ACT_TANH, ACT_RELU, ACT_LEAKYRELU = 0, 1, 2, 3, 4
@@ -67,7 +97,7 @@ def _act_grad_from_y_ref(dy: torch.Tensor, y: torch.Tensor, act: int) -> torch.T
 class _FusedLinearFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, b: torch.Tensor, act: int,
-                w_mirror=None, b_mirror=None):
+                w_mirror=None, b_mirror=None, w_gradview=None, b_gradview=None):
         # w: [out, in].  w_mirror/b_mirror: bf16 views of the flat arena's
         # compute copy (refreshed once per step by the optimizer) — when
         # absent, cast per call.
@@ -89,6 +119,7 @@ class _FusedLinearFn(torch.autograd.Function):
         ctx.act = act
         ctx.hip = use_hip(x)
         ctx.x_needs_grad = x.requires_grad
+        ctx.gradviews = (w_gradview, b_gradview)
         return y
 
     @staticmethod
@@ -97,14 +128,40 @@ class _FusedLinearFn(torch.autograd.Function):
         act = ctx.act
         if ctx.hip:
             ext = hip_ops()
+            gv_w, gv_b = ctx.gradviews
             if w.shape[0] == 1:   # 1-unit head: GEMV backward
                 dz = ext.act_grad(dy.contiguous(), y, act).reshape(-1)
                 dw, db, dx = ext.gemv_bwd(x, w.reshape(-1), dz, ctx.x_needs_grad)
                 if not ctx.x_needs_grad:
                     dx = None
-                return dx, dw, db, None, None, None
+                return dx, dw, db, None, None, None, None, None
             # one fused pass: dz = dy*act'(y) and db = colsum(dz)
             dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
+            if _ASYNC_WGRAD and gv_w is not None and gv_b is not None:
+                # wgrad on a side stream, accumulated straight into the flat
+                # arena; autograd gets None (no AccumulateGrad for w/b)
+                main = torch.cuda.current_stream()
+                ev = torch.cuda.Event()
+                ev.record(main)
+                ws = _wgrad_stream()
+                ws.wait_event(ev)
+                with torch.cuda.stream(ws):
+                    xT = ext.transpose_bf16(x)
+                    dzT = ext.transpose_bf16(dz)
+                    dwv = ext.gemm_ntv3_f32(dzT, xT)
+                    gv_w.add_(dwv)
+                    gv_b.add_(db)
+                    done = torch.cuda.Event()
+                    done.record(ws)
+                # caching-allocator safety: main-stream tensors used on ws
+                for t in (x, dz, db):
+                    t.record_stream(ws)
+                _WGRAD_EVENTS.append(done)
+                dx = None
+                if ctx.x_needs_grad:
+                    wT = ext.transpose_bf16(w)
+                    dx = ext.gemm_ntv3_bf16(dz, wT)
+                return dx, None, None, None, None, None, None, None
             xT = ext.transpose_bf16(x)                # [K,B]
             dzT = ext.transpose_bf16(dz)              # [N,B]
             dw = ext.gemm_ntv3_f32(dzT, xT)           # fp32 [N,K], split-K
@@ -117,7 +174,7 @@ class _FusedLinearFn(torch.autograd.Function):
             dw = dz.t() @ x                           # [N,K]
             db = dz.sum(dim=0)
             dx = dz @ w if ctx.x_needs_grad else None
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None, None
 
 
 def fused_linear(x: torch.Tensor, w: torch.Tensor, b: torch.Tensor,
@@ -153,7 +210,9 @@ class FusedLinear(torch.nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return _FusedLinearFn.apply(x, self.weight, self.bias, self._act,
                                     getattr(self, "_w_mirror", None),
-                                    getattr(self, "_b_mirror", None))
+                                    getattr(self, "_b_mirror", None),
+                                    getattr(self, "_w_gradview", None),
+                                    getattr(self, "_b_gradview", None))
 
     def extra_repr(self) -> str:
         return f"in={self.in_features}, out={self.out_features}, act={self.activation}"

@@ -133,6 +133,9 @@ class FusedOptimizer:
     def step(self) -> None:
         self.step_count += 1
         if self.flat.numel():
+            if self.flat.flat.is_cuda:
+                from shifu_amd.ops.linear import drain_wgrad_events
+                drain_wgrad_events()
             self.flat.sync_grads()
             if use_hip(self.flat.flat):
                 self._dense_step_hip()

@@ -146,6 +146,9 @@ class GradAggregator:
         """Call after backward(): waits for in-flight buckets / runs the
         non-overlapped path, averages, and aggregates sparse embedding grads."""
         world = dist.get_world_size() if is_distributed() else 1
+        if self.flat.numel() and self.flat.flat.is_cuda:
+            from shifu_amd.ops.linear import drain_wgrad_events
+            drain_wgrad_events()
         if is_distributed() and self.sync_enabled and self.flat.numel():
             if self.overlap:
                 for w in self._works:
