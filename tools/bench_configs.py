@@ -26,7 +26,9 @@ def _train_throughput(model, batches, steps, warmup, optimizer="adam", lr=1e-3):
     from shifu_amd.ops.loss import weighted_loss
     from shifu_amd.ops.optim import FusedOptimizer
     dense_params, emb_params = split_params(model)
-    flat = FlatParams(dense_params)
+    from shifu_amd.ops.flat import bind_mirrors
+    flat = FlatParams(dense_params, mirror_bf16=torch.cuda.is_available())
+    bind_mirrors(model, flat)
     opt = FusedOptimizer(flat, emb_params, optimizer=optimizer, lr=lr)
 
     def step(i):
