@@ -12,8 +12,11 @@ scatter-transposes):
                                                tiled-transpose kernel)
   wgrad : dw = nt(dz^T [N,B], x^T [K,B])      (activation transposes, f32
                                                out, split-K)
-Mixed precision: fp32 master weights (autograd leaves), bf16 compute,
-fp32 dW/db, bf16 dX.
+Mixed precision: fp32 master weights (autograd leaves), bf16 compute via
+per-layer views of the flat arena's bf16 mirror (ONE arena-wide cast per
+step — ops/flat.py), fp32 dW/db, bf16 dX.  Weight/bias grads can run on a
+side HIP stream (async wgrad) accumulated directly into the flat-grad
+arena, overlapping the dgrad chain.
 
 Activation gradients are computed from Y (not Z): sigmoid' = y(1-y),
 tanh' = 1-y^2, relu'/leakyrelu' from sign(y) — valid because all four

@@ -7,9 +7,10 @@ reference's per-layer l2_regularizer(0.1) (ssgd_monitor.py:58-68) is folded
 in as coupled weight decay: g += l2 * w (tf l2_regularizer(s)(w) = s*||w||^2/2
 => d/dw = s*w).
 
-Sparse path: embedding arenas get rowwise updates on coalesced sparse grads
-(unique rows after coalesce -> no atomics needed in the update kernel):
-"sgd" or rowwise "adagrad" (fp32 accumulator per row).
+Sparse path: embedding arenas get rowwise updates on UNCOALESCED sparse
+grads (rows may repeat; the kernels scatter with packed-bf16 atomics so no
+sort/dedup runs on the hot path): "sgd" or rowwise "adagrad" (fp32
+accumulator per row).
 """
 from __future__ import annotations
 
