@@ -53,8 +53,15 @@ class WideDeep(torch.nn.Module):
         self.shifu_output_0 = FusedLinear(prev, 1, activation="none", seed=seed + 999)
 
     def forward(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
-        wide = self.wide_cat(cats).sum(dim=1) + self.wide_dense(dense).reshape(-1)
-        emb = self.embeddings(cats)                     # [B, F*D]
+        from shifu_amd.parallel.ep import ShardedEmbedding, ep_pair_gather
+        if isinstance(self.wide_cat, ShardedEmbedding):
+            # shared routing: one sort/size-exchange/all-to-all serves both
+            # the wide (D=1) and deep (D=embed_dim) arenas
+            wide_e, emb = ep_pair_gather(self.wide_cat, self.embeddings, cats)
+            wide = wide_e.sum(dim=1) + self.wide_dense(dense).reshape(-1)
+        else:
+            wide = self.wide_cat(cats).sum(dim=1) + self.wide_dense(dense).reshape(-1)
+            emb = self.embeddings(cats)                 # [B, F*D]
         x = torch.cat([dense, emb.to(dense.dtype)], dim=1)
         for layer in self.tower:
             x = layer(x)

@@ -122,6 +122,84 @@ class _EPGatherFn(torch.autograd.Function):
         return grad, None, None, None, None, None
 
 
+class _EPGatherPairFn(torch.autograd.Function):
+    """Two arenas (same vocab layout, e.g. Wide&Deep's D=1 wide weights and
+    D=64 deep vectors) share ONE routing: one sort, one split-size exchange,
+    one id all-to-all, and a single combined value all-to-all of [m, D1+D2]."""
+
+    @staticmethod
+    def forward(ctx, arena1, arena2, flat_ids, F, D1, D2, world, rank):
+        n = flat_ids.numel()
+        g = flat_ids.reshape(-1)
+        owner = (g % world).to(torch.int64)
+        perm = torch.argsort(owner, stable=True)
+        ids_sorted = g[perm]
+        send_counts = torch.bincount(owner, minlength=world)
+        all_counts = [torch.zeros_like(send_counts) for _ in range(world)]
+        dist.all_gather(all_counts, send_counts)
+        in_splits = [int(c) for c in send_counts]
+        out_splits = [int(all_counts[src][rank]) for src in range(world)]
+        m = sum(out_splits)
+
+        recv_ids = torch.empty(m, dtype=g.dtype, device=g.device)
+        _all_to_all_single(recv_ids, ids_sorted.contiguous(), out_splits, in_splits)
+        local_rows = torch.div(recv_ids, world, rounding_mode="floor")
+
+        if arena1.dtype == torch.bfloat16 and use_hip(arena1):
+            v1 = hip_ops().embedding_gather(arena1, local_rows.reshape(-1, 1))
+            v2 = hip_ops().embedding_gather(arena2, local_rows.reshape(-1, 1))
+        else:
+            v1 = arena1.index_select(0, local_rows)
+            v2 = arena2.index_select(0, local_rows)
+        vals = torch.cat([v1.reshape(m, D1), v2.reshape(m, D2)], dim=1)
+
+        back = torch.empty(n, D1 + D2, dtype=vals.dtype, device=vals.device)
+        _all_to_all_single(back, vals.contiguous(), in_splits, out_splits)
+        out = torch.empty_like(back)
+        out[perm] = back
+
+        ctx.save_for_backward(perm, local_rows)
+        ctx.splits = (in_splits, out_splits)
+        ctx.dims = (D1, D2)
+        ctx.world = world
+        ctx.shapes = (arena1.shape, arena2.shape)
+        ctx.dtypes = (arena1.dtype, arena2.dtype)
+        B = n // F
+        return (out[:, :D1].reshape(B, F * D1).contiguous(),
+                out[:, D1:].reshape(B, F * D2).contiguous())
+
+    @staticmethod
+    def backward(ctx, dout1, dout2):
+        perm, local_rows = ctx.saved_tensors
+        in_splits, out_splits = ctx.splits
+        D1, D2 = ctx.dims
+        world = ctx.world
+        n = perm.numel()
+        dvals = torch.cat([dout1.reshape(n, D1), dout2.reshape(n, D2)], dim=1)
+        dsorted = dvals[perm].contiguous()
+        grad_rows = torch.empty(local_rows.numel(), D1 + D2, dtype=dsorted.dtype,
+                                device=dsorted.device)
+        _all_to_all_single(grad_rows, dsorted, out_splits, in_splits)
+        grad_rows = grad_rows / world
+        idx = local_rows.reshape(1, -1)
+        g1 = torch.sparse_coo_tensor(idx, grad_rows[:, :D1].contiguous()
+                                     .to(ctx.dtypes[0]), ctx.shapes[0])
+        g2 = torch.sparse_coo_tensor(idx, grad_rows[:, D1:].contiguous()
+                                     .to(ctx.dtypes[1]), ctx.shapes[1])
+        return g1, g2, None, None, None, None, None, None
+
+
+def ep_pair_gather(emb1: "ShardedEmbedding", emb2: "ShardedEmbedding",
+                   ids: torch.Tensor):
+    """Shared-routing lookup for two ShardedEmbeddings over the same vocab.
+    Returns ([B, F*D1], [B, F*D2])."""
+    local = ids.clamp(min=0) % emb1.sizes
+    flat = (local + emb1.offsets).reshape(-1)
+    return _EPGatherPairFn.apply(emb1.arena, emb2.arena, flat,
+                                 emb1.num_features, emb1.dim, emb2.dim,
+                                 emb1.world, emb1.rank)
+
+
 class ShardedEmbedding(torch.nn.Module):
     """MultiEmbedding-compatible module whose arena is sharded row%world.
 

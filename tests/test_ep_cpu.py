@@ -146,3 +146,41 @@ def test_ep_training_step_matches_dp():
     for rank, ok_dense, ok_emb in _run(_ep_train_worker, 29723):
         assert ok_dense, f"rank {rank}: dense params diverged"
         assert ok_emb, f"rank {rank}: embedding shard diverged from DP"
+
+
+def _ep_pair_worker(rank, port, q):
+    """Shared-routing pair gather == two independent ShardedEmbedding lookups."""
+    from shifu_amd.parallel.ep import ShardedEmbedding, ep_pair_gather
+    try:
+        _init(rank, port)
+        vocab = [13, 29]
+        e1 = ShardedEmbedding(vocab, dim=1, seed=4, world=WORLD, rank=rank)
+        e2 = ShardedEmbedding(vocab, dim=6, seed=9, world=WORLD, rank=rank)
+        g = torch.Generator().manual_seed(8 + rank)
+        ids = torch.stack([torch.randint(0, 13, (10,), generator=g),
+                           torch.randint(0, 29, (10,), generator=g)], dim=1)
+        o1p, o2p = ep_pair_gather(e1, e2, ids)
+        o1, o2 = e1(ids), e2(ids)
+        ok_fwd = torch.allclose(o1p, o1, atol=1e-6) and torch.allclose(o2p, o2, atol=1e-6)
+
+        # backward parity: same loss through both paths gives same arena grads
+        (o1p.sum() * 2 + o2p.pow(2).sum()).backward()
+        g1p = e1.arena.grad.coalesce().to_dense().clone()
+        g2p = e2.arena.grad.coalesce().to_dense().clone()
+        e1.arena.grad = None
+        e2.arena.grad = None
+        o1b, o2b = e1(ids), e2(ids)
+        (o1b.sum() * 2 + o2b.pow(2).sum()).backward()
+        ok_bwd = (torch.allclose(g1p, e1.arena.grad.coalesce().to_dense(), atol=1e-5) and
+                  torch.allclose(g2p, e2.arena.grad.coalesce().to_dense(), atol=1e-5))
+        q.put((rank, bool(ok_fwd), bool(ok_bwd)))
+        dist.barrier()
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ep_pair_gather_shared_routing():
+    for rank, ok_fwd, ok_bwd in _run(_ep_pair_worker, 29725):
+        assert ok_fwd, f"rank {rank}: pair forward mismatch"
+        assert ok_bwd, f"rank {rank}: pair backward mismatch"
