@@ -79,7 +79,7 @@ def main():
     use_ep = (args.emb_mode == "ep" or (args.emb_mode == "auto" and world > 1))
     model = WideDeep(N_DENSE, [VOCAB] * N_CAT, args.embed_dim, TOWER, ACTS,
                      seed=777, sharded_embeddings=use_ep, world=world,
-                     rank=rank).to(device)
+                     rank=rank, emb_fast_init=True).to(device)
     if on_gpu:
         # keep embedding arenas bf16 (HBM-resident, gathered by the HIP kernel)
         for p in model.parameters():

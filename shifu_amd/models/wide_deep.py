@@ -24,7 +24,8 @@ from shifu_amd.ops.loss import predict_proba
 class WideDeep(torch.nn.Module):
     def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
                  hidden_nodes: List[int], activations: List[str], seed: int = 1234,
-                 sharded_embeddings: bool = False, world: int = 1, rank: int = 0):
+                 sharded_embeddings: bool = False, world: int = 1, rank: int = 0,
+                 emb_fast_init: bool = False):
         super().__init__()
         self.num_dense = num_dense
         self.embed_dim = embed_dim
@@ -34,7 +35,8 @@ class WideDeep(torch.nn.Module):
         if sharded_embeddings and world > 1:
             from shifu_amd.parallel.ep import ShardedEmbedding
             emb = lambda d, s: ShardedEmbedding(self.vocab_sizes, d, seed=s,
-                                                world=world, rank=rank)
+                                                world=world, rank=rank,
+                                                fast_init=emb_fast_init)
         else:
             emb = lambda d, s: MultiEmbedding(self.vocab_sizes, d, seed=s)
 

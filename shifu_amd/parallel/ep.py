@@ -209,7 +209,7 @@ class ShardedEmbedding(torch.nn.Module):
 
     def __init__(self, vocab_sizes: Sequence[int], dim: int, seed: int = 0,
                  world: int = 1, rank: int = 0,
-                 dtype: torch.dtype = torch.float32):
+                 dtype: torch.dtype = torch.float32, fast_init: bool = False):
         super().__init__()
         self.vocab_sizes = [int(v) for v in vocab_sizes]
         self.dim = int(dim)
@@ -221,22 +221,29 @@ class ShardedEmbedding(torch.nn.Module):
         self.register_buffer("offsets", offsets)
         self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64))
 
-        gen = torch.Generator().manual_seed(seed)
         scale = 1.0 / math.sqrt(max(self.dim, 1))
-        # chunked generation: identical RNG stream to MultiEmbedding's single
-        # torch.rand(total, dim) call, but never materializes the full arena
-        # (8 ranks x multi-GB would blow host RAM on one node)
         shard_rows = len(range(self.rank, self.total_rows, self.world))
-        shard = torch.empty(shard_rows, self.dim)
-        CH = 1 << 20
-        out_off = 0
-        for start in range(0, self.total_rows, CH):
-            n = min(CH, self.total_rows - start)
-            chunk = (torch.rand(n, self.dim, generator=gen) * 2 - 1) * scale
-            first = (self.rank - start) % self.world
-            sel = chunk[first::self.world]
-            shard[out_off:out_off + sel.shape[0]] = sel
-            out_off += sel.shape[0]
+        if fast_init:
+            # per-shard seeded draw (NOT bit-identical to MultiEmbedding's
+            # stream — same distribution; use for benchmarks where multi-GB
+            # replicated CPU generation per rank would dominate startup)
+            gen = torch.Generator().manual_seed(seed * 1000003 + rank)
+            shard = (torch.rand(shard_rows, self.dim, generator=gen) * 2 - 1) * scale
+        else:
+            # chunked generation: identical RNG stream to MultiEmbedding's
+            # single torch.rand(total, dim) call, but never materializes the
+            # full arena (8 ranks x multi-GB would blow host RAM on one node)
+            gen = torch.Generator().manual_seed(seed)
+            shard = torch.empty(shard_rows, self.dim)
+            CH = 1 << 20
+            out_off = 0
+            for start in range(0, self.total_rows, CH):
+                n = min(CH, self.total_rows - start)
+                chunk = (torch.rand(n, self.dim, generator=gen) * 2 - 1) * scale
+                first = (self.rank - start) % self.world
+                sel = chunk[first::self.world]
+                shard[out_off:out_off + sel.shape[0]] = sel
+                out_off += sel.shape[0]
         self.arena = torch.nn.Parameter(shard.to(dtype))
         self.arena._is_embedding_arena = True
         self.arena._is_ep_sharded = True   # GradAggregator: no cross-rank sync
