@@ -1,0 +1,25 @@
+"""bench.py driver-contract test: runs the real script (CPU, tiny) and
+validates the JSON line fields the driver parses."""
+import json
+import subprocess
+import sys
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "32"],
+        capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    d = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+              "dtype", "data", "config"):
+        assert k in d, f"missing field {k}"
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True
+    assert d["scaling"] == "weak"
+    assert d["data"] == "synthetic"
+    assert d["value"] > 0
+    assert "global_batch" in d["config"] and "parallelism" in d["config"]

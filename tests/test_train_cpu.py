@@ -187,3 +187,20 @@ def test_per_epoch_aggregation_semantics(tmp_path):
     tr = Trainer(ShifuMLP(10, [16, 8], ["relu", "relu"]), mc, rc, train, valid)
     tr.fit()
     assert tr.optimizer.step_count == 3  # one aggregated update per epoch
+
+
+def test_reference_default_combo(tmp_path):
+    """The reference's production defaults: weighted-MSE loss on a sigmoid
+    head + Adadelta(lr=1.0) + L2(0.1) (ssgd_monitor.py:129,138,58-68)."""
+    train, valid = _data(n=1500)
+    mc = _mc(epochs=6, loss="weighted_mse", opt="adadelta", lr=1.0)
+    mc.params.l2_reg = 0.1
+    rc = RunConfig(tmp_model_path=str(tmp_path / "ckpt"),
+                   final_model_path=str(tmp_path / "final"))
+    tr = Trainer(ShifuMLP(10, [30, 20], ["tanh", "tanh"], seed=5), mc, rc,
+                 train, valid)
+    first = tr.evaluate(tr.valid_data)
+    tr.fit()
+    last = tr.evaluate(tr.valid_data)
+    assert last["loss"] < first["loss"]
+    assert last["auc"] > 0.55
