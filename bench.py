@@ -43,12 +43,12 @@ PER_GPU_BATCH = 16384
 N_BATCHES = 8  # distinct resident batches cycled through the loop
 
 
-def make_batches(device, batch, rank, dtype):
+def make_batches(device, batch, rank, dtype, vocab=VOCAB, n_cat=N_CAT):
     gen = torch.Generator(device="cpu").manual_seed(1234 + rank)
     batches = []
     for i in range(N_BATCHES):
         dense = torch.randn(batch, N_DENSE, generator=gen).to(device=device, dtype=dtype)
-        cats = torch.randint(0, VOCAB, (batch, N_CAT), generator=gen).to(device)
+        cats = torch.randint(0, vocab, (batch, n_cat), generator=gen).to(device)
         target = (torch.rand(batch, generator=gen) > 0.5).float().to(device)
         weight = torch.ones(batch, device=device)
         batches.append((dense, cats, target, weight))
@@ -63,6 +63,9 @@ def main():
     ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
     ap.add_argument("--bucket-mb", type=int, default=128)
     ap.add_argument("--embed-dim", type=int, default=EMBED_DIM)
+    ap.add_argument("--vocab", type=int, default=VOCAB,
+                    help="per-feature vocab (reduce for CPU smoke only)")
+    ap.add_argument("--n-cat", type=int, default=N_CAT)
     ap.add_argument("--graphs", choices=["auto", "on", "off"], default="auto",
                     help="hipGraph-capture the whole training step")
     ap.add_argument("--emb-mode", choices=["auto", "dp", "ep"], default="auto",
@@ -77,7 +80,7 @@ def main():
 
     torch.manual_seed(777)
     use_ep = (args.emb_mode == "ep" or (args.emb_mode == "auto" and world > 1))
-    model = WideDeep(N_DENSE, [VOCAB] * N_CAT, args.embed_dim, TOWER, ACTS,
+    model = WideDeep(N_DENSE, [args.vocab] * args.n_cat, args.embed_dim, TOWER, ACTS,
                      seed=777, sharded_embeddings=use_ep, world=world,
                      rank=rank, emb_fast_init=True).to(device)
     if on_gpu:
@@ -94,7 +97,7 @@ def main():
     opt = FusedOptimizer(flat, emb_params, optimizer="adam", lr=1e-3,
                          l2_reg=0.0, emb_optimizer="adagrad", emb_lr=0.01)
 
-    batches = make_batches(device, args.batch, rank, dtype)
+    batches = make_batches(device, args.batch, rank, dtype, args.vocab, args.n_cat)
 
     use_graphs = (args.graphs == "on" or
                   (args.graphs == "auto" and on_gpu and world == 1))
@@ -178,7 +181,7 @@ def main():
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"wide_deep[{N_CAT}x{VOCAB}vocab*{args.embed_dim}d+{N_DENSE}dense,tower{TOWER}]",
+                "model": f"wide_deep[{args.n_cat}x{args.vocab}vocab*{args.embed_dim}d+{N_DENSE}dense,tower{TOWER}]",
                 "global_batch": n_gpus * args.batch,
                 "per_gpu_batch": args.batch,
                 "seq_len": None,

@@ -8,7 +8,7 @@ import sys
 def test_bench_json_contract():
     out = subprocess.run(
         [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
-         "--batch", "32"],
+         "--batch", "32", "--vocab", "1000", "--n-cat", "4"],
         capture_output=True, text=True, timeout=600)
     assert out.returncode == 0, out.stderr[-2000:]
     line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
