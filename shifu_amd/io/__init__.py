@@ -2,7 +2,7 @@
 fallback (shifu_amd/data/csv_loader.py) when the extension isn't built."""
 from __future__ import annotations
 
-from typing import Optional, Sequence
+from typing import Sequence
 
 _EXT = None
 _TRIED = False

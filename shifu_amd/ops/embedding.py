@@ -13,12 +13,12 @@ Design (MI355X-first):
   a dense grad would be the full arena (GBs) and its all-reduce would
   dominate the step.  The FusedOptimizer aggregates sparse grads across
   ranks (allgather of rows over xGMI) and applies a rowwise update kernel
-  (scatter-add with fp32 atomics on GPU — SURVEY.md §2.4 embedding row).
+  (sort-free packed-bf16 atomic scatter on GPU — SURVEY.md §2.4 embedding row).
 """
 from __future__ import annotations
 
 import math
-from typing import List, Sequence
+from typing import Sequence
 
 import torch
 
