@@ -140,8 +140,10 @@ class Trainer:
                                  use_gpu_events=(self.device.type == "cuda"))
 
     # ------------------------------------------------------------------ steps
-    def train_step(self, batch: DeviceData, sync: bool = True) -> float:
-        """One forward/backward/(all-reduce)/update step; returns loss."""
+    def train_step(self, batch: DeviceData, sync: bool = True) -> torch.Tensor:
+        """One forward/backward/(all-reduce)/update step; returns the loss as
+        a device scalar (call .item() only at epoch granularity — a per-step
+        host readback would stall the pipeline)."""
         tr = self.tracer
         self.aggregator.set_sync(sync)
         with tr.phase("fwd"):
@@ -157,7 +159,7 @@ class Trainer:
                 self.optimizer.step()
                 self.optimizer.zero_grad()
         self.global_step += 1
-        return float(loss.detach())
+        return loss.detach()  # device scalar: no per-step host sync
 
     @torch.no_grad()
     def evaluate(self, data: DeviceData, batch_size: int = 65536) -> dict:
@@ -204,6 +206,7 @@ class Trainer:
             if self.heartbeat and time.time() - self._hb_last >= self._hb_interval:
                 self.heartbeat()
                 self._hb_last = time.time()
+        mean_loss = float(torch.stack(losses).float().mean()) if losses else 0.0
         if self.device.type == "cuda":
             torch.cuda.synchronize()
         train_time = time.time() - t0
@@ -217,7 +220,7 @@ class Trainer:
             current_epoch=epoch,
             current_epoch_time=train_time,
             current_epoch_valid_time=valid_time,
-            training_error=float(np.mean(losses)) if losses else 0.0,
+            training_error=mean_loss,
             valid_error=val["loss"],
             container_id=f"rank-{self.rank}",
         )
