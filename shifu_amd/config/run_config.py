@@ -64,6 +64,9 @@ class RunConfig:
     log_dir: str = "./logs"
     device: str = "auto"                 # "cuda" | "cpu" | "auto"
     enable_trace: bool = False           # per-phase step timing (utils/trace.py)
+    graphs: str = "auto"                 # hipGraph-captured steps: "auto"|"on"|"off"
+                                         # (auto = on for single-GPU cuda, full
+                                         # batches, no window mode)
 
     def resolved_backend(self) -> str:
         if self.backend != "auto":

@@ -58,6 +58,14 @@ def auc_score(scores: np.ndarray, labels: np.ndarray) -> float:
     return float((ranks[pos].sum() - n_pos * (n_pos + 1) / 2.0) / (n_pos * n_neg))
 
 
+def _arena_offset_rows(p: torch.nn.Parameter, model: torch.nn.Module) -> torch.Tensor:
+    """Rows of an embedding arena hit by all-zero ids: its per-feature offsets."""
+    for mod in model.modules():
+        if getattr(mod, "arena", None) is p:
+            return mod.offsets.to(p.device)
+    return torch.zeros(1, dtype=torch.int64, device=p.device)
+
+
 @dataclass
 class DeviceData:
     dense: torch.Tensor   # [N, Fn] (bf16 on GPU, f32 on CPU)
@@ -138,6 +146,16 @@ class Trainer:
         from shifu_amd.utils.trace import StepTracer
         self.tracer = StepTracer(enabled=rc.enable_trace,
                                  use_gpu_events=(self.device.type == "cuda"))
+        # hipGraph-captured steps (full-size batches only; tail batches and
+        # window mode run eager; capture failure falls back silently)
+        g = getattr(rc, "graphs", "auto")
+        self.use_graphs = (g == "on" or (g == "auto" and
+                                         self.device.type == "cuda" and
+                                         world_size == 1 and
+                                         self.update_window == 1 and
+                                         not rc.enable_trace))
+        self._graph = None
+        self._gb: Optional[DeviceData] = None
 
     # ------------------------------------------------------------------ steps
     def train_step(self, batch: DeviceData, sync: bool = True) -> torch.Tensor:
@@ -160,6 +178,87 @@ class Trainer:
                 self.optimizer.zero_grad()
         self.global_step += 1
         return loss.detach()  # device scalar: no per-step host sync
+
+    # ------------------------------------------------------------- graph mode
+    def _ensure_graph(self) -> bool:
+        if self._graph is not None:
+            return True
+        B = self.batch_size
+        self._gb = DeviceData(
+            dense=torch.empty(B, self.train_data.dense.shape[1],
+                              device=self.device, dtype=self.train_data.dense.dtype),
+            cats=torch.empty(B, self.train_data.cats.shape[1],
+                             device=self.device, dtype=torch.int64),
+            target=torch.empty(B, device=self.device),
+            weight=torch.empty(B, device=self.device),
+        )
+        gb = self._gb
+
+        def body():
+            logits = self.model(gb.dense, gb.cats)
+            loss = weighted_loss(logits, gb.target, gb.weight, self.loss_kind)
+            loss.backward()
+            self.aggregator.finish()
+            self.optimizer.step()
+            self.optimizer.zero_grad()
+            return loss
+
+        from shifu_amd.train.graph import GraphedStep
+        try:
+            # Warmup + capture run REAL steps; training semantics require the
+            # model to be unchanged afterwards.  Seed with all-zero ids so
+            # each arena touches only its F offset rows, snapshot the dense
+            # state + those rows, and restore after capture.
+            gb.dense.normal_()
+            gb.cats.zero_()
+            gb.target.fill_(0.5)
+            gb.weight.fill_(1.0)
+            snap = {
+                "flat": self.flat.flat.clone(),
+                "m": None if self.optimizer.m is None else self.optimizer.m.clone(),
+                "v": None if self.optimizer.v is None else self.optimizer.v.clone(),
+                "step_count": self.optimizer.step_count,
+                "emb": [], "emb_state": [],
+            }
+            for i, p in enumerate(self.emb_params):
+                rows = _arena_offset_rows(p, self.model)
+                snap["emb"].append((rows, p.data[rows].clone()))
+                acc = self.optimizer.emb_state.get(i)
+                snap["emb_state"].append(None if acc is None else acc[rows].clone())
+
+            stepper = GraphedStep(body, warmup=3)
+            stepper.capture()
+
+            # restore pre-capture training state
+            self.flat.flat.copy_(snap["flat"])
+            if snap["m"] is not None:
+                self.optimizer.m.copy_(snap["m"])
+            if snap["v"] is not None:
+                self.optimizer.v.copy_(snap["v"])
+            self.optimizer.step_count = snap["step_count"]
+            if hasattr(self.optimizer, "_step_buf"):
+                self.optimizer._step_buf.fill_(float(snap["step_count"]))
+            for i, p in enumerate(self.emb_params):
+                rows, vals = snap["emb"][i]
+                p.data[rows] = vals
+                if snap["emb_state"][i] is not None:
+                    self.optimizer.emb_state[i][rows] = snap["emb_state"][i]
+            self.flat.refresh_mirror()
+            self._graph = stepper
+            return True
+        except Exception:
+            self.use_graphs = False
+            return False
+
+    def _graphed_step(self, idx: torch.Tensor) -> torch.Tensor:
+        gb = self._gb
+        torch.index_select(self.train_data.dense, 0, idx, out=gb.dense)
+        torch.index_select(self.train_data.cats, 0, idx, out=gb.cats)
+        torch.index_select(self.train_data.target, 0, idx, out=gb.target)
+        torch.index_select(self.train_data.weight, 0, idx, out=gb.weight)
+        loss = self._graph.run()
+        self.global_step += 1
+        return loss.detach().clone()
 
     @torch.no_grad()
     def evaluate(self, data: DeviceData, batch_size: int = 65536) -> dict:
@@ -199,9 +298,14 @@ class Trainer:
         steps = range(0, n, self.batch_size)
         n_steps = len(steps)
         for si, s in enumerate(steps):
-            batch = self.train_data.slice(perm[s:min(s + self.batch_size, n)])
-            # window mode: only every update_window-th (or last) step syncs+updates
+            idx = perm[s:min(s + self.batch_size, n)]
             sync = ((si + 1) % self.update_window == 0) or (si == n_steps - 1)
+            if (self.use_graphs and sync and idx.shape[0] == self.batch_size
+                    and self._ensure_graph()):
+                losses.append(self._graphed_step(idx))
+                continue
+            batch = self.train_data.slice(idx)
+            # window mode: only every update_window-th (or last) step syncs+updates
             losses.append(self.train_step(batch, sync=sync))
             if self.heartbeat and time.time() - self._hb_last >= self._hb_interval:
                 self.heartbeat()
