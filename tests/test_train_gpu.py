@@ -112,3 +112,27 @@ def test_adadelta_reference_default_gpu():
         first = tr.evaluate(tr.valid_data)
         tr.fit()
         assert tr.evaluate(tr.valid_data)["loss"] < first["loss"]
+
+
+def test_graphed_trainer_matches_eager():
+    """Graph-captured epochs must train identically to eager epochs (the
+    capture snapshot/restore must leave no side effects)."""
+    dense, cats, target, weight = synthetic_arrays(4096, 12, (300, 300), seed=7)
+    full = TabularDataset(dense, cats, target, weight)
+    train, valid = full.split(0.2, seed=1)
+
+    def run(graphs):
+        with tempfile.TemporaryDirectory() as td:
+            rc = RunConfig(tmp_model_path=td + "/c", final_model_path=td + "/f",
+                           graphs=graphs)
+            tr = Trainer(WideDeep(12, [300, 300], 8, [64, 32], ["relu", "tanh"],
+                                  seed=6), _mc(epochs=2, batch=512), rc,
+                         train, valid, device=torch.device("cuda"))
+            tr.fit()
+            assert (tr._graph is not None) == (graphs == "on"), \
+                f"graph mode not exercised as expected ({graphs})"
+            return tr.evaluate(tr.valid_data)["loss"]
+
+    eager = run("off")
+    graphed = run("on")
+    assert abs(eager - graphed) < 5e-3, (eager, graphed)
