@@ -58,8 +58,8 @@ def make_batches(device, batch, rank, dtype, vocab=VOCAB, n_cat=N_CAT):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
-    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--steps", type=int, default=50)
+    ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch", type=int, default=PER_GPU_BATCH)
     ap.add_argument("--bucket-mb", type=int, default=128)
     ap.add_argument("--embed-dim", type=int, default=EMBED_DIM)

@@ -136,3 +136,30 @@ def test_graphed_trainer_matches_eager():
     eager = run("off")
     graphed = run("on")
     assert abs(eager - graphed) < 5e-3, (eager, graphed)
+
+
+@pytest.mark.parametrize("cfg", [
+    dict(loss="weighted_mse", opt="adadelta", lr=1.0, acts=["tanh", "tanh"], batch=333),
+    dict(loss="sigmoid_ce", opt="adagrad", lr=0.05, acts=["leakyrelu", "sigmoid"], batch=100),
+    dict(loss="weighted_mse", opt="sgd", lr=0.5, acts=["relu", "relu"], batch=1024),
+])
+def test_config_fuzz_gpu(cfg):
+    """Odd batch sizes x every activation/loss/optimizer family through the
+    full HIP Trainer: must run, stay finite, and not regress the loss."""
+    dense, cats, target, weight = synthetic_arrays(3000, 9, (123, 77), seed=13)
+    full = TabularDataset(dense, cats, target, weight)
+    train, valid = full.split(0.2, seed=1)
+    mc = ModelConfig.from_dict({"train": {"numTrainEpochs": 2, "params": {
+        "NumHiddenLayers": 2, "NumHiddenNodes": [48, 24],
+        "ActivationFunc": cfg["acts"], "LearningRate": cfg["lr"],
+        "Optimizer": cfg["opt"], "Loss": cfg["loss"],
+        "MiniBatchSize": cfg["batch"], "L2Reg": 0.05}}})
+    with tempfile.TemporaryDirectory() as td:
+        rc = RunConfig(tmp_model_path=td + "/c", final_model_path=td + "/f")
+        tr = Trainer(WideDeep(9, [123, 77], 8, [48, 24], cfg["acts"], seed=3),
+                     mc, rc, train, valid, device=torch.device("cuda"))
+        first = tr.evaluate(tr.valid_data)
+        tr.fit()
+        last = tr.evaluate(tr.valid_data)
+        assert np.isfinite(last["loss"])
+        assert last["loss"] <= first["loss"] * 1.05  # no blow-up

@@ -10,7 +10,7 @@ python - <<'PY'
 import json, sys
 sys.path.insert(0, "/root/repo")
 from shifu_amd.data.synthetic import generate_synthetic_csv
-paths = generate_synthetic_csv("gpurun_out/e2e_demo/data", n_rows=1_000_000,
+paths = generate_synthetic_csv("gpurun_out/e2e_demo/data", n_rows=500_000,
                                n_dense=50, vocab_sizes=[100_000]*8, n_files=16, seed=42)
 print(f"generated {len(paths)} files")
 cc = [{"columnNum": 0, "columnFlag": "Target"}, {"columnNum": 1, "columnFlag": "Weight"}]
@@ -28,7 +28,7 @@ json.dump({"num_gpus": 1, "training_data_path": ["gpurun_out/e2e_demo/data"],
            "tmp_model_path": "gpurun_out/e2e_demo/ckpt",
            "final_model_path": "gpurun_out/e2e_demo/final",
            "log_dir": "gpurun_out/e2e_demo/logs",
-           "model_type": "wide_deep", "embed_dim": 16, "enable_trace": True},
+           "model_type": "wide_deep", "embed_dim": 16, "enable_trace": False},
           open("gpurun_out/e2e_demo/run.json", "w"))
 PY
 time python -m shifu_amd.run --run-config $OUT/run.json \
