@@ -59,6 +59,9 @@ class RunConfig:
     max_missed_heartbeats: int = 25      # shifu.task.max-missed-heartbeats default 25
     max_rank_restarts: int = 1           # restart a dead rank from last checkpoint
     checkpoint_every_epochs: int = 1
+    checkpoint_every_secs: float = 0.0   # >0: also time-based mid-epoch saves
+                                         # (reference: Supervisor save_model_secs,
+                                         #  ssgd.py:124-128)
 
     # -- misc --
     log_dir: str = "./logs"

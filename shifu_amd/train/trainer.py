@@ -109,6 +109,8 @@ class Trainer:
         self.heartbeat = heartbeat
         self._hb_interval = float(rc.heartbeat_interval_s)
         self._hb_last = time.time()
+        self._ckpt_secs = float(getattr(rc, "checkpoint_every_secs", 0.0))
+        self._ckpt_last = time.time()
 
         self.model = model.to(self.device)
         dense_dtype = torch.bfloat16 if (self.device.type == "cuda"
@@ -310,6 +312,14 @@ class Trainer:
             if self.heartbeat and time.time() - self._hb_last >= self._hb_interval:
                 self.heartbeat()
                 self._hb_last = time.time()
+            if (self._ckpt_secs > 0 and self.is_chief and epoch > 0
+                    and time.time() - self._ckpt_last >= self._ckpt_secs):
+                # mid-epoch time-based save, tagged as the last COMPLETE epoch
+                # so resume replays the current epoch from its start
+                ckpt.save_checkpoint(self.rc.tmp_model_path, epoch - 1,
+                                     self.global_step, self.model, self.optimizer,
+                                     extra={"np_rng": self._rng.bit_generator.state})
+                self._ckpt_last = time.time()
         mean_loss = float(torch.stack(losses).float().mean()) if losses else 0.0
         if self.device.type == "cuda":
             torch.cuda.synchronize()

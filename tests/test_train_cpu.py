@@ -204,3 +204,16 @@ def test_reference_default_combo(tmp_path):
     last = tr.evaluate(tr.valid_data)
     assert last["loss"] < first["loss"]
     assert last["auc"] > 0.55
+
+
+def test_time_based_checkpointing(tmp_path):
+    """rc.checkpoint_every_secs triggers mid-epoch saves (Supervisor
+    save_model_secs successor, ssgd.py:124-128)."""
+    train, valid = _data(n=3000)
+    mc = _mc(epochs=2, batch=16)      # many small steps
+    rc = RunConfig(tmp_model_path=str(tmp_path / "ckpt"),
+                   final_model_path=str(tmp_path / "final"),
+                   checkpoint_every_secs=0.05)
+    tr = Trainer(ShifuMLP(10, [16, 8], ["relu", "relu"]), mc, rc, train, valid)
+    tr.fit()
+    assert latest_checkpoint(str(tmp_path / "ckpt")) is not None
