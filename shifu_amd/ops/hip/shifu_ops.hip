@@ -836,6 +836,196 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
 }
 
 // ---------------------------------------------------------------------------
+// v5 "TT" wgrad GEMM — C[M,N] f32 = sum_b dz[b,m] * x[b,n], both operands
+// stored reduction-major-ROWS ([B, M] / [B, N]) exactly as the activations
+// come out of forward/act-grad: NO pre-transposes.  The column reads MFMA
+// needs come from gfx950's hardware transpose-read (ds_read_b64_tr_b16):
+// each image is stored in [4(b) x 16(col)] row-major blocks (glds fills the
+// blocked layout straight from coalescing-friendly 16 B global pieces), and
+// a tr read hands each lane column (l&15) of block (base + (l>>4)) — i.e.
+// 4 b-values for its fragment column.  Block order within a 16-col group is
+// f(bblk) = (bblk&1)*8 + bblk/2 so one read pair covers b = 8G..8G+7.
+// Split-K over B via f32 atomics (zero-initialized C).
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(4))) short s16x4t;
+typedef __attribute__((__vector_size__(4 * sizeof(short)))) short __attribute__((address_space(3)))* lds_v4s;
+
+#define V5_BM 128
+#define V5_BN 128
+#define V5_BB 64    // reduction rows per staged tile
+#define V5_IMG_BYTES 16384
+
+// image element index e (0..8191) -> (b_local, col) of the logical [64][128]
+DEVINL void v5_inverse(int e, int& b_local, int& col) {
+  int sb = e >> 6;
+  int nblk = sb >> 4;
+  int fb = sb & 15;
+  int bblk = (fb < 8) ? 2 * fb : 2 * (fb - 8) + 1;
+  int j = (e >> 4) & 3;
+  int c16 = e & 15;
+  b_local = bblk * 4 + j;
+  col = nblk * 16 + c16;
+}
+
+// stage one [64 x 128] blocked image; interior: 4 glds/thread
+DEVINL void v5_stage(const bf16* __restrict__ P, char* img, int b0, int c0,
+                     int R, int ld, int tid, bool interior) {
+  const int lane = tid & 63, wave = tid >> 6;
+  if (interior) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e0 = (tid + c * 256) * 8;
+      int bl, col;
+      v5_inverse(e0, bl, col);
+      const bf16* src = P + (long)(b0 + bl) * ld + c0 + col;
+      char* dst = img + wave * 1024 + c * 4096;
+      __builtin_amdgcn_global_load_lds((gas_ptr)src, (las_ptr)dst, 16, 0, 0);
+    }
+  } else {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int e0 = (tid + c * 256) * 8;
+      int bl, col;
+      v5_inverse(e0, bl, col);
+      s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+      long gb = b0 + bl;
+      if (gb < R) {
+        const bf16* src = P + gb * ld + c0 + col;
+        for (int q = 0; q < 8; ++q)
+          if (c0 + col + q < ld) ((short*)&v)[q] = ((const short*)src)[q];
+      }
+      *(s16x8*)(img + (long)e0 * 2) = v;
+    }
+  }
+}
+
+template <bool SPLITK>
+__global__ __launch_bounds__(256)
+void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
+                    float* __restrict__ C, int M, int N, int R) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* Ai = smem;                      // dz image  (cols = m)
+  char* Bi = smem + V5_IMG_BYTES;       // x image   (cols = n)
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = (wave >> 1) * 64;
+  const int wc = (wave & 1) * 64;
+  const int m0 = blockIdx.y * V5_BM;
+  const int n0 = blockIdx.x * V5_BN;
+  const int r16 = lane & 15;
+  const int kgrp = lane >> 4;
+
+  int b_lo = 0, b_hi = R;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int chunk = ((R + nz - 1) / nz + V5_BB - 1) / V5_BB * V5_BB;
+    b_lo = blockIdx.z * chunk;
+    b_hi = min(R, b_lo + chunk);
+    if (b_lo >= b_hi) return;
+  }
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const bool a_int_mn = (m0 + V5_BM <= M);
+  const bool b_int_mn = (n0 + V5_BN <= N);
+  // tr-read base block offsets per (ks, readhalf): f(2G)=G basis
+  const int roff[2][2] = {{0, 8}, {4, 12}};
+
+  for (int b0 = b_lo; b0 < b_hi; b0 += V5_BB) {
+    bool full_b = (b0 + V5_BB <= R);
+    v5_stage(DZ, Ai, b0, m0, R, M, tid, full_b && a_int_mn);
+    v5_stage(X, Bi, b0, n0, R, N, tid, full_b && b_int_mn);
+    __syncthreads();   // drains the glds (vmcnt0) + lgkm
+
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+      for (int fi = 0; fi < 4; ++fi) {
+        const int nblkA = (wr + fi * 16) >> 4;
+        bf16x8 afrag;
+        {
+          s16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (lds_v4s)(Ai + (nblkA * 16 + roff[ks][0]) * 128 + lane * 8));
+          s16x4t r2 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (lds_v4s)(Ai + (nblkA * 16 + roff[ks][1]) * 128 + lane * 8));
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            ((short*)&afrag)[q] = r1[q];
+            ((short*)&afrag)[q + 4] = r2[q];
+          }
+        }
+#pragma unroll
+        for (int fj = 0; fj < 4; ++fj) {
+          const int nblkB = (wc + fj * 16) >> 4;
+          bf16x8 bfrag;
+          s16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (lds_v4s)(Bi + (nblkB * 16 + roff[ks][0]) * 128 + lane * 8));
+          s16x4t r2 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+              (lds_v4s)(Bi + (nblkB * 16 + roff[ks][1]) * 128 + lane * 8));
+#pragma unroll
+          for (int q = 0; q < 4; ++q) {
+            ((short*)&bfrag)[q] = r1[q];
+            ((short*)&bfrag)[q + 4] = r2[q];
+          }
+          acc[fi][fj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[fi][fj], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int fi = 0; fi < 4; ++fi) {
+#pragma unroll
+    for (int fj = 0; fj < 4; ++fj) {
+      int coln = n0 + wc + fj * 16 + r16;
+      if (coln >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int rowm = m0 + wr + fi * 16 + kgrp * 4 + r;
+        if (rowm >= M) continue;
+        if (SPLITK) atomicAdd(&C[(long)rowm * N + coln], acc[fi][fj][r]);
+        else C[(long)rowm * N + coln] = acc[fi][fj][r];
+      }
+    }
+  }
+}
+
+// dw[M=out, N=in] = dz[B, M]^T @ x[B, N]  — transpose-free wgrad
+at::Tensor gemm_tt_f32(at::Tensor dz, at::Tensor x) {
+  CHECK_GPU(dz); CHECK_CONTIG(dz); CHECK_BF16(dz);
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
+  long R = dz.size(0), M = dz.size(1), N = x.size(1);
+  TORCH_CHECK(x.size(0) == R, "reduction mismatch dz^T@x");
+  auto c = at::empty({M, N}, x.options().dtype(at::kFloat));
+  hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
+  long gx = (N + V5_BN - 1) / V5_BN, gy = (M + V5_BM - 1) / V5_BM;
+  long max_z = (R + V5_BB - 1) / V5_BB;
+  long z = std::min<long>(
+      std::max<long>(splitk_target_blocks() / std::max<long>(gx * gy, 1), 1), max_z);
+  auto s = cur_stream();
+  if (z <= 1) {
+    hipLaunchKernelGGL((gemm_tt_kernel<false>), dim3((unsigned)gx, (unsigned)gy),
+                       dim3(256), 2 * V5_IMG_BYTES, s,
+                       (const bf16*)dz.data_ptr(), (const bf16*)x.data_ptr(),
+                       (float*)c.data_ptr(), (int)M, (int)N, (int)R);
+  } else {
+    hipLaunchKernelGGL((gemm_tt_kernel<true>), dim3((unsigned)gx, (unsigned)gy, (unsigned)z),
+                       dim3(256), 2 * V5_IMG_BYTES, s,
+                       (const bf16*)dz.data_ptr(), (const bf16*)x.data_ptr(),
+                       (float*)c.data_ptr(), (int)M, (int)N, (int)R);
+  }
+  return c;
+}
+
+// ---------------------------------------------------------------------------
 // tiled bf16 transpose: out[C,R] = in[R,C]^T.  64x64 LDS tiles (+8 pad),
 // 16B coalesced global loads AND stores; scalar traffic stays inside LDS.
 // ---------------------------------------------------------------------------
@@ -1655,6 +1845,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_ntv3_bf16", &gemm_ntv3_bf16);
   m.def("gemm_ntv3_f32", &gemm_ntv3_f32);
   m.def("transpose_bf16", &transpose_bf16);
+  m.def("gemm_tt_f32", &gemm_tt_f32);
   m.def("gemv_fwd", &gemv_fwd);
   m.def("gemv_bwd", &gemv_bwd);
   m.def("fm2_fwd", &fm2_fwd);

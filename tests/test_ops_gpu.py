@@ -174,6 +174,18 @@ def test_fused_linear_head_autograd_gpu():
         assert rel < 0.08, f"{name} rel {rel}"
 
 
+@pytest.mark.parametrize("R,M,N", [(64, 128, 128), (512, 256, 384),
+                                   (16384, 1024, 1864), (200, 100, 130),
+                                   (1000, 37, 64)])
+def test_gemm_tt_wgrad(R, M, N):
+    """Transpose-free wgrad (hardware tr_b16 reads): dw = dz^T @ x."""
+    dz = _rand_bf16(R, M, seed=R + 21, scale=0.5)
+    x = _rand_bf16(R, N, seed=N + 22, scale=0.5)
+    c = hip_ops().gemm_tt_f32(dz, x)
+    ok, err = _rel_close(c, dz.float().t() @ x.float())
+    assert ok, f"gemm_tt {R}x{M}x{N} maxdiff={err}"
+
+
 def test_gemm_nt():
     dz = _rand_bf16(320, 96, seed=8)
     w = _rand_bf16(130, 96, seed=9)

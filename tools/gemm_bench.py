@@ -20,6 +20,8 @@ SHAPES = [
     ("l1_dgrd3", 8192, 1864, 1024, "v3"),
     ("l1_wgrad", 1864, 1024, 8192, "tn"),
     ("l1_wgrd3", 1024, 1864, 8192, "v3f"),
+    ("l1_wgrdT", 1024, 1864, 8192, "tt"),
+    ("l1wgT16k", 1024, 1864, 16384, "tt"),
     ("l1f16k", 16384, 1024, 1864, "v3"),
     ("l1d16k", 16384, 1864, 1024, "v3"),
     ("sq4096", 4096, 4096, 4096, "nn"),
@@ -69,6 +71,12 @@ def main():
             b = torch.randn(N, K, device="cuda").to(torch.bfloat16)
             ours = lambda: ext.gemm_ntv3_bf16(a, b)
             ref = lambda: a @ b.t()
+        elif kind == "tt":  # transpose-free wgrad: dz[B,M], x[B,N]
+            a0 = torch.randn(K, M, device="cuda").to(torch.bfloat16)
+            b0 = torch.randn(K, N, device="cuda").to(torch.bfloat16)
+            ours = lambda: ext.gemm_tt_f32(a0, b0)
+            ref = lambda: a0.t().float() @ b0.float()
+            a, b = ext.transpose_bf16(a0), ext.transpose_bf16(b0)
         else:  # v3f: wgrad incl. the two activation transposes
             a0 = torch.randn(K, M, device="cuda").to(torch.bfloat16)  # dz [B,N']
             b0 = torch.randn(K, N, device="cuda").to(torch.bfloat16)  # x [B,K']
