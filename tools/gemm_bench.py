@@ -86,7 +86,7 @@ def main():
 
         c1 = ours().float()
         c2 = (a.float() @ b.float()) if kind == "nn" else \
-             (a.float() @ b.float().t()) if kind in ("nt", "v3", "v3f") else \
+             (a.float() @ b.float().t()) if kind in ("nt", "v3", "v3f", "tt") else \
              (a.float().t() @ b.float())
         rel = float((c1 - c2).abs().max() / c2.abs().max().clamp_min(1e-3))
 
