@@ -904,8 +904,8 @@ __global__ __launch_bounds__(256)
 void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
                     float* __restrict__ C, int M, int N, int R) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* Ai = smem;                      // dz image  (cols = m)
-  char* Bi = smem + V5_IMG_BYTES;       // x image   (cols = n)
+  char* Ai = smem;                      // buffer0: dz image (cols = m)
+  char* Bi = smem + V5_IMG_BYTES;       // buffer0: x image  (cols = n)
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -937,11 +937,29 @@ void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
   // tr-read base block offsets per (ks, readhalf): f(2G)=G basis
   const int roff[2][2] = {{0, 8}, {4, 12}};
 
+  // double-buffered: stage tile t+1 while computing tile t; ONE counted
+  // drain + raw barrier per tile (a __syncthreads per tile would stall on
+  // the in-flight glds before every compute phase)
+  {
+    bool fb = (b_lo + V5_BB <= R);
+    v5_stage(DZ, Ai, b_lo, m0, R, M, tid, fb && a_int_mn);
+    v5_stage(X, Bi, b_lo, n0, R, N, tid, fb && b_int_mn);
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+  int cur = 0;
   for (int b0 = b_lo; b0 < b_hi; b0 += V5_BB) {
-    bool full_b = (b0 + V5_BB <= R);
-    v5_stage(DZ, Ai, b0, m0, R, M, tid, full_b && a_int_mn);
-    v5_stage(X, Bi, b0, n0, R, N, tid, full_b && b_int_mn);
-    __syncthreads();   // drains the glds (vmcnt0) + lgkm
+    char* Ac = smem + cur * 2 * V5_IMG_BYTES;
+    char* Bc = Ac + V5_IMG_BYTES;
+    int bn = b0 + V5_BB;
+    if (bn < b_hi) {
+      char* An = smem + (cur ^ 1) * 2 * V5_IMG_BYTES;
+      bool fb = (bn + V5_BB <= R);
+      v5_stage(DZ, An, bn, m0, R, M, tid, fb && a_int_mn);
+      v5_stage(X, An + V5_IMG_BYTES, bn, n0, R, N, tid, fb && b_int_mn);
+    }
+    const char* Ai2 = Ac;
+    const char* Bi2 = Bc;
 
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -951,9 +969,9 @@ void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
         bf16x8 afrag;
         {
           s16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (lds_v4s)(Ai + (nblkA * 16 + roff[ks][0]) * 128 + lane * 8));
+              (lds_v4s)(Ai2 + (nblkA * 16 + roff[ks][0]) * 128 + lane * 8));
           s16x4t r2 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (lds_v4s)(Ai + (nblkA * 16 + roff[ks][1]) * 128 + lane * 8));
+              (lds_v4s)(Ai2 + (nblkA * 16 + roff[ks][1]) * 128 + lane * 8));
 #pragma unroll
           for (int q = 0; q < 4; ++q) {
             ((short*)&afrag)[q] = r1[q];
@@ -965,9 +983,9 @@ void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
           const int nblkB = (wc + fj * 16) >> 4;
           bf16x8 bfrag;
           s16x4t r1 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (lds_v4s)(Bi + (nblkB * 16 + roff[ks][0]) * 128 + lane * 8));
+              (lds_v4s)(Bi2 + (nblkB * 16 + roff[ks][0]) * 128 + lane * 8));
           s16x4t r2 = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
-              (lds_v4s)(Bi + (nblkB * 16 + roff[ks][1]) * 128 + lane * 8));
+              (lds_v4s)(Bi2 + (nblkB * 16 + roff[ks][1]) * 128 + lane * 8));
 #pragma unroll
           for (int q = 0; q < 4; ++q) {
             ((short*)&bfrag)[q] = r1[q];
@@ -978,7 +996,9 @@ void gemm_tt_kernel(const bf16* __restrict__ DZ, const bf16* __restrict__ X,
         }
       }
     }
-    __syncthreads();
+    asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
   }
 
 #pragma unroll
@@ -1013,12 +1033,12 @@ at::Tensor gemm_tt_f32(at::Tensor dz, at::Tensor x) {
   auto s = cur_stream();
   if (z <= 1) {
     hipLaunchKernelGGL((gemm_tt_kernel<false>), dim3((unsigned)gx, (unsigned)gy),
-                       dim3(256), 2 * V5_IMG_BYTES, s,
+                       dim3(256), 4 * V5_IMG_BYTES, s,
                        (const bf16*)dz.data_ptr(), (const bf16*)x.data_ptr(),
                        (float*)c.data_ptr(), (int)M, (int)N, (int)R);
   } else {
     hipLaunchKernelGGL((gemm_tt_kernel<true>), dim3((unsigned)gx, (unsigned)gy, (unsigned)z),
-                       dim3(256), 2 * V5_IMG_BYTES, s,
+                       dim3(256), 4 * V5_IMG_BYTES, s,
                        (const bf16*)dz.data_ptr(), (const bf16*)x.data_ptr(),
                        (float*)c.data_ptr(), (int)M, (int)N, (int)R);
   }
