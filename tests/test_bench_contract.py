@@ -23,3 +23,21 @@ def test_bench_json_contract():
     assert d["data"] == "synthetic"
     assert d["value"] > 0
     assert "global_batch" in d["config"] and "parallelism" in d["config"]
+
+
+def test_bench_torchrun_2rank_contract():
+    """The driver's multi-GPU launch path: torchrun world=2 (gloo on CPU),
+    EP embeddings, JSON from rank 0 only."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29651", "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "32", "--vocab", "1000", "--n-cat", "4"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, f"expected exactly one JSON line, got {len(lines)}"
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["global_batch"] == 64
+    assert "ep2" in d["config"]["parallelism"]
