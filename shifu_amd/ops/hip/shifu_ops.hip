@@ -627,37 +627,36 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       }
 
 #pragma unroll
-    for (int q = 0; q < 4; ++q) {
-      // A fragments for accumulator rows 2q, 2q+1
-      bf16x8 afr[2][2];
+    for (int q = 0; q < 2; ++q) {
+      // A fragments for accumulator rows 4q..4q+3
+      bf16x8 afr[4][2];
 #pragma unroll
-      for (int m2 = 0; m2 < 2; ++m2)
+      for (int m2 = 0; m2 < 4; ++m2)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
-          int row = wm * 128 + (q * 2 + m2) * 16 + r16;
+          int row = wm * 128 + (q * 4 + m2) * 16 + r16;
           int g = ks * 4 + kgrp;
           afr[m2][ks] = *(const bf16x8*)(Ar + row * 128 + ((g ^ rx) * 16));
         }
-      // stage schedule: q0 -> A half0 (t+1); q1 -> A half1 (t+1) + B half0
-      // (t+2); q2 -> B half1 (t+2)
-      if (q == 0 && t + 1 < n_kt)
+      // stage schedule: q0 -> A halves (t+1); q1 -> B halves (t+2)
+      if (q == 0 && t + 1 < n_kt) {
         v4_stage_slot(A, B, smem, 0, kt + 1, m0, n0, M, N, K, tid);
-      if (q == 1) {
-        if (t + 1 < n_kt) v4_stage_slot(A, B, smem, 1, kt + 1, m0, n0, M, N, K, tid);
-        if (t + 2 < n_kt) v4_stage_slot(A, B, smem, 2, kt + 2, m0, n0, M, N, K, tid);
+        v4_stage_slot(A, B, smem, 1, kt + 1, m0, n0, M, N, K, tid);
       }
-      if (q == 2 && t + 2 < n_kt)
+      if (q == 1 && t + 2 < n_kt) {
+        v4_stage_slot(A, B, smem, 2, kt + 2, m0, n0, M, N, K, tid);
         v4_stage_slot(A, B, smem, 3, kt + 2, m0, n0, M, N, K, tid);
+      }
 
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int m2 = 0; m2 < 2; ++m2)
+      for (int m2 = 0; m2 < 4; ++m2)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
           for (int nf = 0; nf < 4; ++nf)
-            acc[q * 2 + m2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afr[m2][ks], bfr[nf][ks], acc[q * 2 + m2][nf], 0, 0, 0);
+            acc[q * 4 + m2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[m2][ks], bfr[nf][ks], acc[q * 4 + m2][nf], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
       // phase barrier: bound wave skew so the next phase's staging never
       // overwrites a region a lagging wave still reads.  lgkmcnt(0) flushes
