@@ -44,3 +44,20 @@ def test_bad_input_rejected(exported):
     client = TestClient(create_app(path))
     resp = client.post("/score", json={"rows": [[1.0]]})  # too few features
     assert resp.status_code == 400
+
+
+def test_reference_eval_fixture_equivalent(tmp_path):
+    """Mirror of the reference's ONLY real test (TensorflowModelTest.java:35-60):
+    score a 1522-feature random row through an exported model and assert the
+    sigmoid output lies in [0,1]."""
+    import numpy as np
+    from shifu_amd.models.mlp import ShifuMLP
+    from shifu_amd.serve import ShifuScorer
+    model = ShifuMLP(1522, [50, 20], ["tanh", "tanh"], seed=9)
+    export_model(model, str(tmp_path / "final"))
+    sc = ShifuScorer()
+    sc.init(str(tmp_path / "final" / "GenericModelConfig.json"))
+    rng = np.random.default_rng(0)
+    for _ in range(5):
+        p = sc.compute(list(rng.standard_normal(1522)))
+        assert 0.0 <= p <= 1.0
