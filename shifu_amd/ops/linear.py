@@ -40,7 +40,11 @@ from shifu_amd.ops.dispatch import use_hip, hip_ops
 # Consumers (optimizer / aggregator) call drain_wgrad_events() before
 # touching the flat grads.
 # ---------------------------------------------------------------------------
-_ASYNC_WGRAD = os.environ.get("SHIFU_ASYNC_WGRAD", "1") == "1"
+# OFF by default: measured -30% on deep dense towers (the side stream
+# becomes the critical path and competes for CUs) and ~neutral on Wide&Deep;
+# with it off, per-bucket all-reduce/backward overlap via the accumulate-grad
+# hooks applies on multi-GPU.  Opt in with SHIFU_ASYNC_WGRAD=1.
+_ASYNC_WGRAD = os.environ.get("SHIFU_ASYNC_WGRAD", "0") == "1"
 _WGRAD_STREAM = None
 _WGRAD_EVENTS: list = []
 
