@@ -39,7 +39,7 @@ VOCAB = 1_000_000
 EMBED_DIM = 64
 TOWER = [1024, 512, 256]
 ACTS = ["relu", "relu", "relu"]
-PER_GPU_BATCH = 16384
+PER_GPU_BATCH = 32768
 N_BATCHES = 8  # distinct resident batches cycled through the loop
 
 
