@@ -61,10 +61,11 @@ class WideDeep(torch.nn.Module):
             # the wide (D=1) and deep (D=embed_dim) arenas
             wide_e, emb = ep_pair_gather(self.wide_cat, self.embeddings, cats)
             wide = wide_e.sum(dim=1) + self.wide_dense(dense).reshape(-1)
+            x = torch.cat([dense, emb.to(dense.dtype)], dim=1)
         else:
             wide = self.wide_cat(cats).sum(dim=1) + self.wide_dense(dense).reshape(-1)
-            emb = self.embeddings(cats)                 # [B, F*D]
-        x = torch.cat([dense, emb.to(dense.dtype)], dim=1)
+            from shifu_amd.ops.embedding import gather_concat
+            x = gather_concat(self.embeddings, cats, dense)  # [B, nd+F*D] fused
         for layer in self.tower:
             x = layer(x)
         deep = self.shifu_output_0(x).reshape(-1)

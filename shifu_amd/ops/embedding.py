@@ -67,6 +67,49 @@ class _EmbGatherFn(torch.autograd.Function):
         return grad, None, None, None
 
 
+class _EmbGatherConcatFn(torch.autograd.Function):
+    """Fused [dense | embeddings] tower-input build: the gather kernel writes
+    straight into the concat buffer (a separate torch.cat pass over the
+    [B, nd+F*D] activations measured ~4% of the Wide&Deep step)."""
+
+    @staticmethod
+    def forward(ctx, arena, flat_ids, dense, F, D):
+        B, nd = dense.shape
+        ext = hip_ops()
+        out = torch.empty(B, nd + F * D, device=dense.device, dtype=dense.dtype)
+        out[:, :nd].copy_(dense)
+        ext.embedding_gather_into(arena, flat_ids.contiguous(), out, nd)
+        ctx.save_for_backward(flat_ids)
+        ctx.arena_shape = arena.shape
+        ctx.F, ctx.D, ctx.nd = F, D, nd
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        (flat_ids,) = ctx.saved_tensors
+        B = flat_ids.shape[0]
+        values = dout[:, ctx.nd:].reshape(B * ctx.F, ctx.D)
+        grad = torch.sparse_coo_tensor(flat_ids.reshape(1, B * ctx.F),
+                                       values, ctx.arena_shape)
+        # dense input carries no grad in training (raw features)
+        return grad, None, None, None, None
+
+
+def gather_concat(emb: "MultiEmbedding", ids: torch.Tensor,
+                  dense: torch.Tensor) -> torch.Tensor:
+    """[dense | emb(ids)] in one pass when the fused kernel applies;
+    falls back to gather + torch.cat."""
+    if (emb.arena.dtype == torch.bfloat16 and dense.dtype == torch.bfloat16
+            and emb.dim % 8 == 0 and use_hip(emb.arena)
+            and not dense.requires_grad):
+        local = ids.clamp(min=0) % emb.sizes
+        flat = local + emb.offsets
+        return _EmbGatherConcatFn.apply(emb.arena, flat, dense.contiguous(),
+                                        emb.num_features, emb.dim)
+    out = emb(ids)
+    return torch.cat([dense, out.to(dense.dtype)], dim=1)
+
+
 class MultiEmbedding(torch.nn.Module):
     """F categorical features -> concatenated [B, F*D] embeddings from one arena."""
 

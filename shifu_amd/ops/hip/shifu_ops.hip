@@ -1473,10 +1473,13 @@ void adagrad_step(at::Tensor w, at::Tensor g, at::Tensor m,
 // embedding arena gather: out[b, f*D+d] = arena[ids[b,f], d]  (bf16)
 // one 16B (8 bf16) chunk per thread; out writes coalesced.
 // ---------------------------------------------------------------------------
+// out may be a STRIDED view: out[b] row starts at out + b*out_stride + col0
+// (gather-into-concat: writes land directly inside the tower input buffer)
 __global__ void emb_gather_kernel(const bf16* __restrict__ arena,
                                   const long* __restrict__ ids,
                                   bf16* __restrict__ out,
-                                  long rows_bf, long D) {
+                                  long rows_bf, long D, long F,
+                                  long out_stride, long col0) {
   long chunks_per_row = D / 8;
   long total = rows_bf * chunks_per_row;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -1485,7 +1488,9 @@ __global__ void emb_gather_kernel(const bf16* __restrict__ arena,
     long rf = t / chunks_per_row;      // (b*F + f)
     long c = t % chunks_per_row;
     long row = ids[rf];
-    ((s16x8*)out)[t] = *(const s16x8*)(arena + row * D + c * 8);
+    long b = rf / F, f = rf % F;
+    *(s16x8*)(out + b * out_stride + col0 + f * D + c * 8) =
+        *(const s16x8*)(arena + row * D + c * 8);
   }
 }
 
@@ -1514,7 +1519,7 @@ at::Tensor embedding_gather(at::Tensor arena, at::Tensor ids) {
     int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
     hipLaunchKernelGGL(emb_gather_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
                        (const bf16*)arena.data_ptr(), (const long*)ids.data_ptr(),
-                       (bf16*)out.data_ptr(), rows, D);
+                       (bf16*)out.data_ptr(), rows, D, F, F * D, 0L);
   } else {
     long total = rows * D;
     int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
@@ -1637,6 +1642,26 @@ __global__ void emb_denom_kernel(const float* __restrict__ acc, const long* __re
 
 static int scat_blocks(long total) {
   return (int)std::min((total + 255) / 256 + 1, (long)4096);
+}
+
+// gather arena rows for ids [B,F] directly into out[:, col0:col0+F*D]
+// (out is the [B, n_dense + F*D] tower-input buffer; avoids a separate
+// torch.cat pass over the concatenated activations)
+void embedding_gather_into(at::Tensor arena, at::Tensor ids, at::Tensor out,
+                           long col0) {
+  CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
+  CHECK_GPU(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64");
+  long B = ids.size(0), F = ids.size(1), D = arena.size(1);
+  TORCH_CHECK(D % 8 == 0, "gather_into needs D%8==0");
+  long out_stride = out.size(1);
+  TORCH_CHECK(col0 + F * D <= out_stride, "slice out of range");
+  long rows = B * F;
+  long total = rows * (D / 8);
+  int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
+  hipLaunchKernelGGL(emb_gather_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const bf16*)arena.data_ptr(), (const long*)ids.contiguous().data_ptr(),
+                     (bf16*)out.data_ptr(), rows, D, F, out_stride, col0);
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
@@ -1881,6 +1906,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adadelta_step", &adadelta_step);
   m.def("adagrad_step", &adagrad_step);
   m.def("embedding_gather", &embedding_gather);
+  m.def("embedding_gather_into", &embedding_gather_into);
   m.def("emb_sgd_step", &emb_sgd_step);
   m.def("emb_adagrad_step", &emb_adagrad_step);
 }

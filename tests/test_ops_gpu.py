@@ -402,3 +402,33 @@ def test_fm2_fused_gpu():
     dref = ((s.unsqueeze(1) - v) * g.reshape(B, 1, 1)).reshape(B, F * D)
     ok, err = _rel_close(emb.grad, dref, 4e-2)
     assert ok, f"fm2 bwd maxdiff={err}"
+
+
+def test_gather_concat_fused():
+    """Fused [dense|emb] build == gather + cat, forward and arena grads."""
+    from shifu_amd.ops.embedding import MultiEmbedding, gather_concat
+    emb = MultiEmbedding([50, 70], dim=8, seed=1)
+    emb.cuda()
+    emb.arena.data = emb.arena.data.to(torch.bfloat16)
+    g = torch.Generator().manual_seed(2)
+    ids = torch.stack([torch.randint(0, 50, (64,), generator=g),
+                       torch.randint(0, 70, (64,), generator=g)], 1).cuda()
+    dense = torch.randn(64, 10, generator=g).to(torch.bfloat16).cuda()
+
+    x1 = gather_concat(emb, ids, dense)
+    (x1.float() * torch.arange(x1.numel(), device="cuda").reshape(x1.shape)) \
+        .sum().backward()
+    from shifu_amd.ops.embedding import sparse_rows_values
+    r1, v1 = sparse_rows_values(emb.arena.grad)
+    emb.arena.grad = None
+
+    x2 = torch.cat([dense, emb(ids).to(dense.dtype)], dim=1)
+    assert torch.equal(x1.float().cpu(), x2.float().cpu())
+    (x2.float() * torch.arange(x2.numel(), device="cuda").reshape(x2.shape)) \
+        .sum().backward()
+    r2, v2 = sparse_rows_values(emb.arena.grad)
+    d1 = torch.zeros(120, 8, device="cuda")
+    d1.index_add_(0, r1, v1.float())
+    d2 = torch.zeros(120, 8, device="cuda")
+    d2.index_add_(0, r2, v2.float())
+    assert torch.allclose(d1.cpu(), d2.cpu(), atol=1e-2)
