@@ -52,16 +52,18 @@ class DeepFM(torch.nn.Module):
         B = dense.shape[0]
         F = len(self.vocab_sizes)
         from shifu_amd.parallel.ep import ShardedEmbedding, ep_pair_gather
+        from shifu_amd.ops.fm import fm_second_order
         if isinstance(self.fm_first, ShardedEmbedding):
             fm1_e, emb_flat = ep_pair_gather(self.fm_first, self.embeddings, cats)
             fm1 = fm1_e.sum(dim=1) + self.fm_dense(dense).reshape(-1)
+            fm2 = fm_second_order(emb_flat, F, self.embed_dim)
+            x = torch.cat([dense, emb_flat.to(dense.dtype)], dim=1)
         else:
             fm1 = self.fm_first(cats).sum(dim=1) + self.fm_dense(dense).reshape(-1)
-            emb_flat = self.embeddings(cats)               # [B, F*D]
-        from shifu_amd.ops.fm import fm_second_order
-        fm2 = fm_second_order(emb_flat, F, self.embed_dim)  # [B] (fused on GPU)
-
-        x = torch.cat([dense, emb_flat.to(dense.dtype)], dim=1)
+            from shifu_amd.ops.embedding import gather_concat
+            x = gather_concat(self.embeddings, cats, dense)  # [B, nd+F*D] fused
+            emb_view = x[:, self.num_dense:]                 # strided view
+            fm2 = fm_second_order(emb_view, F, self.embed_dim)
         for layer in self.tower:
             x = layer(x)
         deep = self.shifu_output_0(x).reshape(-1)

@@ -16,8 +16,9 @@ class _FM2Fn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, emb_flat: torch.Tensor, F: int, D: int):
         B = emb_flat.shape[0]
-        if emb_flat.dtype == torch.bfloat16 and use_hip(emb_flat):
-            fm2, s = hip_ops().fm2_fwd(emb_flat.contiguous(), F, D)
+        if (emb_flat.dtype == torch.bfloat16 and use_hip(emb_flat)
+                and emb_flat.stride(1) == 1):
+            fm2, s = hip_ops().fm2_fwd(emb_flat, F, D)   # strided view OK
             ctx.save_for_backward(emb_flat, s)
         else:
             v = emb_flat.reshape(B, F, D).float()
@@ -25,7 +26,8 @@ class _FM2Fn(torch.autograd.Function):
             fm2 = 0.5 * (s * s - (v * v).sum(dim=1)).sum(dim=1)
             ctx.save_for_backward(emb_flat, s)
         ctx.F, ctx.D = F, D
-        ctx.hip = emb_flat.dtype == torch.bfloat16 and use_hip(emb_flat)
+        ctx.hip = (emb_flat.dtype == torch.bfloat16 and use_hip(emb_flat)
+                   and emb_flat.stride(1) == 1)
         return fm2
 
     @staticmethod
@@ -34,8 +36,7 @@ class _FM2Fn(torch.autograd.Function):
         F, D = ctx.F, ctx.D
         B = emb_flat.shape[0]
         if ctx.hip:
-            demb = hip_ops().fm2_bwd(emb_flat.contiguous(), s,
-                                     dout.float().contiguous(), F, D)
+            demb = hip_ops().fm2_bwd(emb_flat, s, dout.float().contiguous(), F, D)
         else:
             v = emb_flat.reshape(B, F, D).float()
             demb = ((s.unsqueeze(1) - v) * dout.reshape(B, 1, 1).float()) \

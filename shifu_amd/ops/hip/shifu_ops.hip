@@ -1818,7 +1818,8 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
 // bwd: dv[b,f,d] = (s[b,d] - v[b,f,d]) * dout[b]
 // ---------------------------------------------------------------------------
 __global__ void fm2_fwd_kernel(const bf16* __restrict__ emb, float* __restrict__ fm2,
-                               float* __restrict__ s, long B, long F, long D) {
+                               float* __restrict__ s, long B, long F, long D,
+                               long row_stride) {
   long b = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (b >= B) return;
   int lane = threadIdx.x & 63;
@@ -1826,7 +1827,7 @@ __global__ void fm2_fwd_kernel(const bf16* __restrict__ emb, float* __restrict__
   for (long d = lane; d < D; d += 64) {
     float sum = 0.0f, sq = 0.0f;
     for (long f = 0; f < F; ++f) {
-      float v = __bfloat162float(emb[b * F * D + f * D + d]);
+      float v = __bfloat162float(emb[b * row_stride + f * D + d]);
       sum += v;
       sq += v * v;
     }
@@ -1840,41 +1841,46 @@ __global__ void fm2_fwd_kernel(const bf16* __restrict__ emb, float* __restrict__
 
 __global__ void fm2_bwd_kernel(const bf16* __restrict__ emb, const float* __restrict__ s,
                                const float* __restrict__ dout, bf16* __restrict__ demb,
-                               long B, long F, long D) {
+                               long B, long F, long D, long row_stride) {
   long total = B * F * D;
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long t = i; t < total; t += stride) {
     long b = t / (F * D);
-    long d = t % D;
-    float v = __bfloat162float(emb[t]);
+    long r = t % (F * D);
+    long d = r % D;
+    float v = __bfloat162float(emb[b * row_stride + r]);
     demb[t] = __float2bfloat16((s[b * D + d] - v) * dout[b]);
   }
 }
 
 std::vector<at::Tensor> fm2_fwd(at::Tensor emb, long F, long D) {
-  CHECK_GPU(emb); CHECK_CONTIG(emb); CHECK_BF16(emb);
-  long B = emb.size(0);
+  // emb: [B, F*D] contiguous, or a row-strided view [B, F*D] of a wider
+  // contiguous buffer (the fused tower-input concat)
+  CHECK_GPU(emb); CHECK_BF16(emb);
+  TORCH_CHECK(emb.stride(1) == 1, "inner dim must be contiguous");
+  long B = emb.size(0), rs = emb.stride(0);
   auto fm2 = at::empty({B}, emb.options().dtype(at::kFloat));
   auto sum = at::empty({B, D}, emb.options().dtype(at::kFloat));
   int wpb = 4;
   hipLaunchKernelGGL(fm2_fwd_kernel, dim3((unsigned)((B + wpb - 1) / wpb)),
                      dim3(64 * wpb), 0, cur_stream(),
                      (const bf16*)emb.data_ptr(), (float*)fm2.data_ptr(),
-                     (float*)sum.data_ptr(), B, F, D);
+                     (float*)sum.data_ptr(), B, F, D, rs);
   return {fm2, sum};
 }
 
 at::Tensor fm2_bwd(at::Tensor emb, at::Tensor s, at::Tensor dout, long F, long D) {
-  CHECK_GPU(emb); CHECK_CONTIG(emb); CHECK_BF16(emb);
+  CHECK_GPU(emb); CHECK_BF16(emb);
+  TORCH_CHECK(emb.stride(1) == 1, "inner dim must be contiguous");
   CHECK_F32(s); CHECK_F32(dout);
-  long B = emb.size(0);
-  auto demb = at::empty_like(emb);
+  long B = emb.size(0), rs = emb.stride(0);
+  auto demb = at::empty({B, F * D}, emb.options());
   long total = B * F * D;
   hipLaunchKernelGGL(fm2_bwd_kernel, dim3(scat_blocks(total)), dim3(256), 0, cur_stream(),
                      (const bf16*)emb.data_ptr(), (const float*)s.data_ptr(),
                      (const float*)dout.contiguous().data_ptr(),
-                     (bf16*)demb.data_ptr(), B, F, D);
+                     (bf16*)demb.data_ptr(), B, F, D, rs);
   return demb;
 }
 
