@@ -821,12 +821,17 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
   long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
   long kt4 = (K + V4_BK - 1) / V4_BK;
-  // v4 only where its grid alone fills the chip (1 block/CU); split-K parts
-  // under ~16 K-tiles spend too long in the ring prologue — measured slower
-  // than the 128^2 v3 split-K on the tabular wgrad shapes.
+  // v4 where its grid fills the chip; v4 split-K only when each part keeps
+  // >=24 K-tiles (the ring prologue is 3 tiles — short parts measured
+  // slower than the 128^2 v3 split-K at reduction 8192, faster at 32768).
+  long z24 = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
+                            std::max<long>(kt4 / 24, 1));
   if (use_v4(M, N, K) && gx4 * gy4 >= 200) {
     launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                                  (float*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
+  } else if (M >= 256 && N >= 256 && z24 > 1 && gx4 * gy4 * z24 >= 200) {
+    launch_nt_v4_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                            (float*)c.data_ptr(), M, N, K, cur_stream(), z24);
   } else {
     launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                          (float*)c.data_ptr(), M, N, K, cur_stream());

@@ -22,6 +22,8 @@ SHAPES = [
     ("l1_wgrd3", 1024, 1864, 8192, "v3f"),
     ("l1_wgrdT", 1024, 1864, 8192, "tt"),
     ("l1wgT16k", 1024, 1864, 16384, "tt"),
+    ("l1wg32k", 1024, 1864, 32768, "v3f"),
+    ("l2wg32k", 512, 1024, 32768, "v3f"),
     ("l1f16k", 16384, 1024, 1864, "v3"),
     ("l1d16k", 16384, 1864, 1024, "v3"),
     ("sq4096", 4096, 4096, 4096, "nn"),
