@@ -750,6 +750,18 @@ static long splitk_target_blocks() {
   return t;
 }
 
+// Minimum K-tiles per split-K part before the v4 ring pipeline is allowed:
+// its prologue stages 3 tiles, so short parts waste a large fraction of the
+// loop in fill/drain (measured: slower than v3 at reduction 8192 with
+// 21-tile parts, faster at 32768 with >=24).
+static long splitk_min_kt() {
+  static long t = [] {
+    const char* e = getenv("SHIFU_SPLITK_MINKT");
+    return e ? atol(e) : 24L;
+  }();
+  return t;
+}
+
 static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
                                  long M, long N, long K, hipStream_t s) {
   long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
@@ -822,10 +834,10 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
   long kt4 = (K + V4_BK - 1) / V4_BK;
   // v4 where its grid fills the chip; v4 split-K only when each part keeps
-  // >=24 K-tiles (the ring prologue is 3 tiles — short parts measured
-  // slower than the 128^2 v3 split-K at reduction 8192, faster at 32768).
-  long z24 = std::min<long>(std::max<long>(512 / std::max<long>(gx4 * gy4, 1), 1),
-                            std::max<long>(kt4 / 24, 1));
+  // >=splitk_min_kt() K-tiles (see splitk_min_kt above).
+  long z24 = std::min<long>(
+      std::max<long>(splitk_target_blocks() / std::max<long>(gx4 * gy4, 1), 1),
+      std::max<long>(kt4 / splitk_min_kt(), 1));
   if (use_v4(M, N, K) && gx4 * gy4 >= 200) {
     launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                                  (float*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
