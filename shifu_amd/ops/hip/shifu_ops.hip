@@ -386,11 +386,15 @@ DEVINL void nt_stage(const bf16* __restrict__ P, bf16* lds, int r0, int k0,
   }
 }
 
+// slab != 0 (split-K only): instead of atomicAdd into C[M,N], each z-part
+// plain-stores its partial into C + blockIdx.z*slab (slab = M*N) and a
+// separate splitk_reduce_kernel sums the parts — no f32 atomics, bitwise
+// deterministic.  The host guarantees every z-part is non-empty in slab mode.
 template <int EPI, typename OUT_T, bool SPLITK = false>
 __global__ __launch_bounds__(256)
 void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                     OUT_T* __restrict__ C, const bf16* __restrict__ bias,
-                    int M, int N, int K, int act) {
+                    int M, int N, int K, int act, long slab) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   bf16* As = (bf16*)smem;                          // [128][64] swizzled
   bf16* Bs = (bf16*)(smem + NT_BM * NT_BK * 2);    // [128][64] swizzled
@@ -468,8 +472,10 @@ void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
         float v = acc[fi][fj][r];
         if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
         if (EPI == EPI_F32) {
-          if (SPLITK) atomicAdd(&((float*)C)[(long)row * N + col], v);
-          else ((float*)C)[(long)row * N + col] = v;
+          if (SPLITK) {
+            if (slab) ((float*)C)[(long)blockIdx.z * slab + (long)row * N + col] = v;
+            else atomicAdd(&((float*)C)[(long)row * N + col], v);
+          } else ((float*)C)[(long)row * N + col] = v;
         } else {
           ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
         }
@@ -562,7 +568,7 @@ template <int EPI, typename OUT_T, bool SPLITK = false>
 __global__ __launch_bounds__(512, 1)
 void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                        OUT_T* __restrict__ C, const bf16* __restrict__ bias,
-                       int M, int N, int K, int act) {
+                       int M, int N, int K, int act, long slab) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -700,8 +706,10 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
         float v = acc[mf][nf][r];
         if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bv, act);
         if (EPI == EPI_F32) {
-          if (SPLITK) atomicAdd(&((float*)C)[(long)row * N + col], v);
-          else ((float*)C)[(long)row * N + col] = v;
+          if (SPLITK) {
+            if (slab) ((float*)C)[(long)blockIdx.z * slab + (long)row * N + col] = v;
+            else atomicAdd(&((float*)C)[(long)row * N + col], v);
+          } else ((float*)C)[(long)row * N + col] = v;
         } else {
           ((bf16*)C)[(long)row * N + col] = __float2bfloat16(v);
         }
@@ -720,18 +728,7 @@ static void launch_nt_v4(const bf16* A, const bf16* B, OUT_T* C, const bf16* bia
   hipFuncSetAttribute((const void*)kern,
                       hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
   hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
-                     A, B, C, bias, (int)M, (int)N, (int)K, act);
-}
-
-static void launch_nt_v4_splitk_f32(const bf16* A, const bf16* B, float* C,
-                                    long M, long N, long K, hipStream_t s, long z) {
-  dim3 grid((unsigned)((N + V4_BN - 1) / V4_BN),
-            (unsigned)((M + V4_BM - 1) / V4_BM), (unsigned)z);
-  auto kern = gemm_nt_v4_kernel<EPI_F32, float, true>;
-  hipFuncSetAttribute((const void*)kern,
-                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
-  hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
-                     A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
+                     A, B, C, bias, (int)M, (int)N, (int)K, act, 0L);
 }
 
 template <int EPI, typename OUT_T>
@@ -739,8 +736,9 @@ static void launch_nt(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
                       long M, long N, long K, int act, hipStream_t s) {
   dim3 grid((N + NT_BN - 1) / NT_BN, (M + NT_BM - 1) / NT_BM);
   hipLaunchKernelGGL((gemm_nt_kernel<EPI, OUT_T, false>), grid, dim3(256),
-                     NT_LDS_BYTES, s, A, B, C, bias, (int)M, (int)N, (int)K, act);
+                     NT_LDS_BYTES, s, A, B, C, bias, (int)M, (int)N, (int)K, act, 0L);
 }
+
 
 static long splitk_target_blocks() {
   static long t = [] {
@@ -748,6 +746,61 @@ static long splitk_target_blocks() {
     return e ? atol(e) : 512L;
   }();
   return t;
+}
+
+// C[i] = sum_z W[z*MN + i] — the slab-mode reduction (f32x4 grid-stride).
+__global__ __launch_bounds__(256)
+void splitk_reduce_kernel(const float* __restrict__ W, float* __restrict__ C,
+                          long MN, int z) {
+  long t0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long nthr = (long)gridDim.x * blockDim.x;
+  if ((MN & 3) == 0) {
+    long nv = MN >> 2;
+    for (long i = t0; i < nv; i += nthr) {
+      f32x4 s = ((const f32x4*)W)[i];
+      for (int zz = 1; zz < z; ++zz) s += ((const f32x4*)(W + (long)zz * MN))[i];
+      ((f32x4*)C)[i] = s;
+    }
+  } else {
+    for (long i = t0; i < MN; i += nthr) {
+      float s = W[i];
+      for (int zz = 1; zz < z; ++zz) s += W[(long)zz * MN + i];
+      C[i] = s;
+    }
+  }
+}
+
+static void launch_splitk_reduce(const float* W, float* C, long MN, long z,
+                                 hipStream_t s) {
+  long work = (MN & 3) == 0 ? MN >> 2 : MN;
+  int blocks = (int)std::min<long>((work + 255) / 256, 2048);
+  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0, s,
+                     W, C, MN, (int)z);
+}
+
+// Split-K accumulation mode (default slab): each z-part plain-stores its
+// partial into a [z, M*N] workspace and splitk_reduce_kernel sums it —
+// bitwise deterministic and measured 11-22% faster than f32 atomicAdd on the
+// bench wgrad shapes (atomics to the same lines serialize in L2; see
+// profiles/r1_measurements.md).  SHIFU_SPLITK_SLAB=0 restores atomics.
+static bool splitk_slab_mode() {
+  static int v = [] {
+    const char* e = getenv("SHIFU_SPLITK_SLAB");
+    return e ? atoi(e) : 1;
+  }();
+  return v != 0;
+}
+
+// Shrink z until the kernels' ceil-chunking leaves no empty z-part (slab mode
+// requires every part to store its slab; fixpoint of per=ceil(t/z), z=ceil(t/per)).
+static long splitk_no_empty_z(long tiles, long z) {
+  z = std::min(std::max(z, 1L), std::max(tiles, 1L));
+  for (;;) {
+    long per = (tiles + z - 1) / z;
+    long z2 = (tiles + per - 1) / per;
+    if (z2 == z) return z;
+    z = z2;
+  }
 }
 
 // Minimum K-tiles per split-K part before the v4 ring pipeline is allowed:
@@ -762,16 +815,58 @@ static long splitk_min_kt() {
   return t;
 }
 
-static void launch_nt_splitk_f32(const bf16* A, const bf16* B, float* C,
-                                 long M, long N, long K, hipStream_t s) {
+// v4 split-K into c[M,N] f32 (handles its own zero/workspace; c may be uninit).
+static void run_nt_v4_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
+                                 long M, long N, long K, hipStream_t s, long z) {
+  dim3 grid((unsigned)((N + V4_BN - 1) / V4_BN),
+            (unsigned)((M + V4_BM - 1) / V4_BM), (unsigned)z);
+  auto kern = gemm_nt_v4_kernel<EPI_F32, float, true>;
+  hipFuncSetAttribute((const void*)kern,
+                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
+  float* cp = (float*)c.data_ptr();
+  if (splitk_slab_mode()) {
+    long kt = (K + V4_BK - 1) / V4_BK;
+    long zs = splitk_no_empty_z(kt, z);
+    grid.z = (unsigned)zs;
+    auto w = at::empty({zs, M * N}, c.options());
+    hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
+                       A, B, (float*)w.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K, 0, M * N);
+    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, zs, s);
+  } else {
+    hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
+    hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
+                       A, B, cp, nullptr, (int)M, (int)N, (int)K, 0, 0L);
+  }
+}
+
+// v3 split-K into c[M,N] f32 (handles its own zero/workspace; c may be uninit).
+static void run_nt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
+                              long M, long N, long K, hipStream_t s) {
   long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
   long max_z = (K + NT_BK - 1) / NT_BK;
   long z = std::min<long>(
       std::max<long>(splitk_target_blocks() / std::max<long>(gx * gy, 1), 1), max_z);
-  if (z <= 1) { launch_nt<EPI_F32, float>(A, B, C, nullptr, M, N, K, 0, s); return; }
-  dim3 grid((unsigned)gx, (unsigned)gy, (unsigned)z);
-  hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>), grid, dim3(256),
-                     NT_LDS_BYTES, s, A, B, C, nullptr, (int)M, (int)N, (int)K, 0);
+  float* cp = (float*)c.data_ptr();
+  if (z <= 1) {
+    launch_nt<EPI_F32, float>(A, B, cp, nullptr, M, N, K, 0, s);
+    return;
+  }
+  if (splitk_slab_mode()) {
+    z = splitk_no_empty_z(max_z, z);
+    auto w = at::empty({z, M * N}, c.options());
+    hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>),
+                       dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
+                       NT_LDS_BYTES, s, A, B, (float*)w.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K, 0, M * N);
+    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, z, s);
+  } else {
+    hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
+    hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>),
+                       dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
+                       NT_LDS_BYTES, s, A, B, cp, nullptr,
+                       (int)M, (int)N, (int)K, 0, 0L);
+  }
 }
 
 // v4 (256^2 pipelined) pays when both tile dims fill 256 rows and the K loop
@@ -830,11 +925,11 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   long M = a.size(0), K = a.size(1), N = b.size(0);
   TORCH_CHECK(b.size(1) == K, "shape mismatch a@b^T");
   auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
-  hipMemsetAsync(c.data_ptr(), 0, (size_t)M * N * 4, cur_stream());
   long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
   long kt4 = (K + V4_BK - 1) / V4_BK;
   // v4 where its grid fills the chip; v4 split-K only when each part keeps
-  // >=splitk_min_kt() K-tiles (see splitk_min_kt above).
+  // >=splitk_min_kt() K-tiles (see splitk_min_kt above).  Each route covers
+  // every in-range element of c, so no zero-fill is needed here.
   long z24 = std::min<long>(
       std::max<long>(splitk_target_blocks() / std::max<long>(gx4 * gy4, 1), 1),
       std::max<long>(kt4 / splitk_min_kt(), 1));
@@ -842,11 +937,11 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
     launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                                  (float*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
   } else if (M >= 256 && N >= 256 && z24 > 1 && gx4 * gy4 * z24 >= 200) {
-    launch_nt_v4_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-                            (float*)c.data_ptr(), M, N, K, cur_stream(), z24);
+    run_nt_v4_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                         c, M, N, K, cur_stream(), z24);
   } else {
-    launch_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
-                         (float*)c.data_ptr(), M, N, K, cur_stream());
+    run_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                      c, M, N, K, cur_stream());
   }
   return c;
 }
