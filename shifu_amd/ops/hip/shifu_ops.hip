@@ -748,6 +748,17 @@ static long splitk_target_blocks() {
   return t;
 }
 
+// v4 runs 1 block/CU (128 KB LDS), so its split-K sweet spot is ~256 blocks
+// (one per CU) — measured 188us vs 223us at 512 blocks on [1024,1864,32768].
+// v3 (256 threads, 32 KB LDS) needs 2+ blocks/CU, hence the 512 above.
+static long splitk_target_blocks_v4() {
+  static long t = [] {
+    const char* e = getenv("SHIFU_SPLITK_TARGET_V4");
+    return e ? atol(e) : 256L;
+  }();
+  return t;
+}
+
 // C[i] = sum_z W[z*MN + i] — the slab-mode reduction (f32x4 grid-stride).
 __global__ __launch_bounds__(256)
 void splitk_reduce_kernel(const float* __restrict__ W, float* __restrict__ C,
@@ -847,6 +858,18 @@ static void run_nt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
   long max_z = (K + NT_BK - 1) / NT_BK;
   long z = std::min<long>(
       std::max<long>(splitk_target_blocks() / std::max<long>(gx * gy, 1), 1), max_z);
+  // Prefer parts >=16 K-tiles when that still leaves >=256 blocks (measured:
+  // [256,512,32768] 37.6us at z=32/256 blocks vs 43.6us at z=64/512 blocks;
+  // shapes where the cap would under-fill the chip keep the 512-block z).
+  // SHIFU_SPLITK_V3MINKT=1 disables the cap.
+  static long v3minkt = [] {
+    const char* e = getenv("SHIFU_SPLITK_V3MINKT");
+    return e ? atol(e) : 16L;
+  }();
+  // Only for deep reductions (>=256 K-tiles): at K=8192 the same cap cost 5%
+  // end-to-end on DeepFM (parts got too few blocks-in-flight per period).
+  long z16 = std::max<long>(max_z / std::max(v3minkt, 1L), 1);
+  if (max_z >= 256 && gx * gy * z16 >= 256) z = std::min(z, z16);
   float* cp = (float*)c.data_ptr();
   if (z <= 1) {
     launch_nt<EPI_F32, float>(A, B, cp, nullptr, M, N, K, 0, s);
@@ -931,7 +954,7 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
   // >=splitk_min_kt() K-tiles (see splitk_min_kt above).  Each route covers
   // every in-range element of c, so no zero-fill is needed here.
   long z24 = std::min<long>(
-      std::max<long>(splitk_target_blocks() / std::max<long>(gx4 * gy4, 1), 1),
+      std::max<long>(splitk_target_blocks_v4() / std::max<long>(gx4 * gy4, 1), 1),
       std::max<long>(kt4 / splitk_min_kt(), 1));
   if (use_v4(M, N, K) && gx4 * gy4 >= 200) {
     launch_nt_v4<EPI_F32, float>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
