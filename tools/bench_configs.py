@@ -31,7 +31,7 @@ def _train_throughput(model, batches, steps, warmup, optimizer="adam", lr=1e-3):
     bind_mirrors(model, flat)
     opt = FusedOptimizer(flat, emb_params, optimizer=optimizer, lr=lr)
 
-    def step(i):
+    def eager_step(i):
         dense, cats, target, weight = batches[i % len(batches)]
         loss = weighted_loss(model(dense, cats), target, weight, "sigmoid_ce")
         loss.backward()
@@ -39,6 +39,45 @@ def _train_throughput(model, batches, steps, warmup, optimizer="adam", lr=1e-3):
         opt.step()
         opt.zero_grad()
         return loss
+
+    step = eager_step
+    if torch.cuda.is_available():
+        # hipGraph the step (same machinery as bench.py): the eager loop is
+        # host-launch-bound on these sub-ms models and measured 12-14M
+        # samples/s box-to-box noise on identical code; graphed runs are stable.
+        try:
+            sd, sc, st, sw = [x.clone() for x in batches[0]]
+
+            def body():
+                loss = weighted_loss(model(sd, sc), st, sw, "sigmoid_ce")
+                loss.backward()
+                flat.sync_grads()
+                opt.step()
+                opt.zero_grad()
+
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(3):
+                    body()
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                body()
+
+            def graphed_step(i):
+                dense, cats, target, weight = batches[i % len(batches)]
+                sd.copy_(dense)
+                st.copy_(target)
+                sw.copy_(weight)
+                if sc.numel():
+                    sc.copy_(cats)
+                graph.replay()
+
+            step = graphed_step
+        except Exception as e:  # capture unsupported -> eager numbers
+            print(f"# hipGraph capture failed ({e}); falling back to eager",
+                  flush=True)
 
     for i in range(warmup):
         step(i)
