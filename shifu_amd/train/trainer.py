@@ -349,6 +349,13 @@ class Trainer:
             np_rng = (info.get("extra") or {}).get("np_rng")
             if np_rng is not None and self.rank == 0:
                 self._rng.bit_generator.state = np_rng
+            else:
+                # Non-chief ranks (the checkpoint stores only rank 0's stream)
+                # and old checkpoints: fast-forward by the permutations already
+                # consumed, so the resumed run draws the same epoch orderings
+                # an uninterrupted run would have.
+                for _ in range(self.start_epoch):
+                    self._rng.permutation(len(self.train_data))
 
     def fit(self) -> List[TrainingIntermediateResult]:
         self.maybe_resume()

@@ -217,3 +217,57 @@ def test_time_based_checkpointing(tmp_path):
     tr = Trainer(ShifuMLP(10, [16, 8], ["relu", "relu"]), mc, rc, train, valid)
     tr.fit()
     assert latest_checkpoint(str(tmp_path / "ckpt")) is not None
+
+
+def test_resume_rng_fast_forward(tmp_path):
+    """A checkpoint without a saved numpy RNG stream (non-chief ranks; legacy
+    checkpoints) must fast-forward the permutation stream so the resumed run
+    draws the same epoch orderings an uninterrupted run would."""
+    train, valid = _data(n=400)
+    rc = RunConfig(tmp_model_path=str(tmp_path / "ckpt"),
+                   final_model_path=str(tmp_path / "final"))
+    model = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+    tr = Trainer(model, _mc(epochs=2), rc, train, valid)
+    tr.fit()
+
+    path = latest_checkpoint(str(tmp_path / "ckpt"))
+    blob = torch.load(path, weights_only=False)
+    blob["extra"].pop("np_rng")
+    torch.save(blob, path)
+
+    model2 = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+    tr2 = Trainer(model2, _mc(epochs=4), rc, train, valid)
+    tr2.maybe_resume()
+    assert tr2.start_epoch == 2
+
+    ref = np.random.default_rng(rc.seed + 0)
+    n = len(tr2.train_data)
+    for _ in range(2):
+        ref.permutation(n)
+    assert np.array_equal(tr2._rng.permutation(n), ref.permutation(n))
+
+
+def test_resume_equivalence_cpu(tmp_path):
+    """4 epochs straight vs 2 + resume + 2 on CPU fp32: identical final
+    valid loss (exact determinism — no atomics on the CPU path)."""
+    train, valid = _data(n=400)
+
+    def run(epochs, sub):
+        rc = RunConfig(tmp_model_path=str(tmp_path / sub / "ckpt"),
+                       final_model_path=str(tmp_path / sub / "final"))
+        model = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+        tr = Trainer(model, _mc(epochs=epochs), rc, train, valid)
+        tr.fit()
+        return tr
+
+    straight = run(4, "a").evaluate(run(4, "a2").valid_data)["loss"]
+
+    rc_b = RunConfig(tmp_model_path=str(tmp_path / "b" / "ckpt"),
+                     final_model_path=str(tmp_path / "b" / "final"))
+    m1 = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+    Trainer(m1, _mc(epochs=2), rc_b, train, valid).fit()
+    m2 = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+    tr_res = Trainer(m2, _mc(epochs=4), rc_b, train, valid)
+    tr_res.fit()
+    resumed = tr_res.evaluate(tr_res.valid_data)["loss"]
+    assert abs(straight - resumed) < 1e-7, (straight, resumed)
