@@ -58,6 +58,11 @@ class RunConfig:
     heartbeat_interval_s: float = 1.0    # shifu.task.heartbeat-interval default 1000ms
     max_missed_heartbeats: int = 25      # shifu.task.max-missed-heartbeats default 25
     max_rank_restarts: int = 1           # restart a dead rank from last checkpoint
+    startup_grace_s: float = 360.0       # heartbeat budget before a rank's FIRST
+                                         # message: model/arena init can take
+                                         # minutes (successor of the reference's
+                                         # 6-min cluster-registration cutover,
+                                         # util/Constants.java:92-94)
     checkpoint_every_epochs: int = 1
     checkpoint_every_secs: float = 0.0   # >0: also time-based mid-epoch saves
                                          # (reference: Supervisor save_model_secs,

@@ -113,15 +113,20 @@ class Launcher:
         procs = self._spawn(ctx, q)
         world = self.rc.num_gpus
         last_hb = {r: time.time() for r in range(world)}
+        started = {r: False for r in range(world)}   # first message received?
         done = set()
         err: Optional[str] = None
         hb_budget = self.rc.heartbeat_interval_s * self.rc.max_missed_heartbeats
+        # until a rank's first message, allow the (long) init grace: model /
+        # multi-GB arena construction runs before any heartbeat can be sent
+        grace = max(hb_budget, float(getattr(self.rc, "startup_grace_s", 360.0)))
 
         try:
             while len(done) < world and err is None:
                 try:
                     msg, rank, payload = q.get(timeout=self.rc.heartbeat_interval_s)
                     last_hb[rank] = time.time()
+                    started[rank] = True
                     if msg == MSG_METRIC:
                         self._on_metric(payload)
                     elif msg == MSG_DONE:
@@ -134,11 +139,12 @@ class Launcher:
                 for r, p in enumerate(procs):
                     if r in done:
                         continue
+                    budget = hb_budget if started[r] else grace
                     if not p.is_alive() and p.exitcode not in (0, None):
                         err = f"rank {r} exited with code {p.exitcode}"
-                    elif now - last_hb[r] > hb_budget:
+                    elif now - last_hb[r] > budget:
                         err = (f"rank {r} missed heartbeats for {now - last_hb[r]:.0f}s "
-                               f"(budget {hb_budget:.0f}s)")
+                               f"(budget {budget:.0f}s)")
         finally:
             for p in procs:
                 if err is not None and p.is_alive():

@@ -63,3 +63,38 @@ def test_launcher_gives_up_after_max_restarts(tmp_path):
     la = Launcher(rc, ModelConfig(), _always_fail_entry)
     with pytest.raises(RuntimeError, match="failed after"):
         la.run()
+
+
+def _slow_start_entry(rank, world, rc, mc, sink, heartbeat):
+    time.sleep(3.0)   # longer than the steady-state budget, within startup grace
+    heartbeat()
+    sink(TrainingIntermediateResult(worker_index=rank, current_epoch=0,
+                                    training_error=0.1, valid_error=0.1,
+                                    current_epoch_time=0.05))
+
+
+def _hb_then_hang_entry(rank, world, rc, mc, sink, heartbeat):
+    heartbeat()       # ends the startup grace for this rank
+    time.sleep(60)
+
+
+def test_startup_grace_allows_slow_init(tmp_path):
+    """Model/arena construction may exceed the heartbeat budget; before a
+    rank's first message the (long) startup grace applies (successor of the
+    reference's 6-min registration cutover, Constants.java:92-94)."""
+    rc = RunConfig(num_gpus=1, log_dir=str(tmp_path), heartbeat_interval_s=0.2,
+                   max_missed_heartbeats=5, startup_grace_s=30.0)
+    stats = Launcher(rc, ModelConfig(), _slow_start_entry).run()
+    assert len(stats) == 1
+
+
+def test_heartbeat_timeout_after_start(tmp_path):
+    """Once a rank has spoken, the normal heartbeat budget applies and a hung
+    rank is detected quickly."""
+    rc = RunConfig(num_gpus=1, log_dir=str(tmp_path), heartbeat_interval_s=0.2,
+                   max_missed_heartbeats=5, startup_grace_s=30.0,
+                   max_rank_restarts=0)
+    t0 = time.time()
+    with pytest.raises(RuntimeError, match="missed heartbeats"):
+        Launcher(rc, ModelConfig(), _hb_then_hang_entry).run()
+    assert time.time() - t0 < 25.0   # detected via budget, not the 60s sleep
