@@ -37,6 +37,10 @@ class RunConfig:
     # -- model family --
     model_type: str = "auto"   # "mlp" | "wide_deep" | "deepfm" | "auto"
     embed_dim: int = 16
+    emb_mode: str = "auto"     # "ep" shards arenas row%world with all-to-all
+                               # routing (parallel/ep.py); "dp" replicates +
+                               # sparse-allgathers grads; "auto" = ep at
+                               # world>1 (the xGMI-native choice)
 
     # -- model / training --
     model_config_path: Optional[str] = None

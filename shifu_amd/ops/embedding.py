@@ -114,7 +114,8 @@ class MultiEmbedding(torch.nn.Module):
     """F categorical features -> concatenated [B, F*D] embeddings from one arena."""
 
     def __init__(self, vocab_sizes: Sequence[int], dim: int,
-                 seed: int = 0, device: str = "cpu", dtype: torch.dtype = torch.float32):
+                 seed: int = 0, device: str = "cpu", dtype: torch.dtype = torch.float32,
+                 empty_init: bool = False):
         super().__init__()
         self.vocab_sizes = [int(v) for v in vocab_sizes]
         self.dim = int(dim)
@@ -125,9 +126,14 @@ class MultiEmbedding(torch.nn.Module):
         self.register_buffer("offsets", offsets)
         # device-resident so forward makes no host->device copies (hipGraph-safe)
         self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64))
-        gen = torch.Generator().manual_seed(seed)
-        scale = 1.0 / math.sqrt(max(self.dim, 1))
-        arena = (torch.rand(self.total_rows, self.dim, generator=gen) * 2 - 1) * scale
+        if empty_init:
+            # caller overwrites (e.g. EP export consolidation) — skip the
+            # multi-GB random draw
+            arena = torch.empty(self.total_rows, self.dim)
+        else:
+            gen = torch.Generator().manual_seed(seed)
+            scale = 1.0 / math.sqrt(max(self.dim, 1))
+            arena = (torch.rand(self.total_rows, self.dim, generator=gen) * 2 - 1) * scale
         self.arena = torch.nn.Parameter(arena.to(dtype))
         self.arena._is_embedding_arena = True  # FusedOptimizer routes this to the sparse path
 

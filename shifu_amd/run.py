@@ -59,9 +59,12 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
             if rc.resolved_device() == "cuda" else torch.device("cpu")
 
         vocab = rc.vocab_sizes or [1000] * len(rc.selected_categorical_columns)
+        emb_mode = getattr(rc, "emb_mode", "auto")
+        use_ep = (emb_mode == "ep" or (emb_mode == "auto" and world > 1))
         model = build_model(mc, len(rc.selected_numeric_columns), vocab,
                             model_type=rc.resolved_model_type(),
-                            embed_dim=rc.embed_dim, seed=rc.seed)
+                            embed_dim=rc.embed_dim, seed=rc.seed,
+                            sharded_embeddings=use_ep, world=world, rank=rank)
         trainer = Trainer(model, mc, rc, train, valid, rank=rank,
                           world_size=world, device=device, metric_sink=metric_sink,
                           heartbeat=heartbeat)

@@ -376,9 +376,56 @@ class Trainer:
                 os.path.join(self.rc.log_dir, f"trace-rank{self.rank}.json"))
             if self.is_chief:
                 print(self.tracer.summary_line(), flush=True)
+        swapped = self._consolidate_sharded_for_export()  # collective when EP
         if self.is_chief:
-            export_model(self.model, self.rc.final_model_path,
-                         model_name=self.mc.model_name, algorithm=self.mc.algorithm,
-                         selected_columns=(self.rc.selected_numeric_columns +
-                                           self.rc.selected_categorical_columns))
+            try:
+                export_model(self.model, self.rc.final_model_path,
+                             model_name=self.mc.model_name, algorithm=self.mc.algorithm,
+                             selected_columns=(self.rc.selected_numeric_columns +
+                                               self.rc.selected_categorical_columns))
+            finally:
+                for name, orig in swapped.items():
+                    parent = self.model
+                    parts = name.split(".")
+                    for part in parts[:-1]:
+                        parent = getattr(parent, part)
+                    setattr(parent, parts[-1], orig)
         return results
+
+    def _consolidate_sharded_for_export(self) -> dict:
+        """EP models: all-gather each sharded arena (row r lives on rank
+        r % world) and temporarily swap a full replicated MultiEmbedding into
+        the chief's model so the export is self-contained.  A collective —
+        EVERY rank must call this; returns {module_name: original} on the
+        chief (for restore), {} elsewhere."""
+        from shifu_amd.parallel.ep import ShardedEmbedding
+        sharded = [(n, m) for n, m in self.model.named_modules()
+                   if isinstance(m, ShardedEmbedding)]
+        if not sharded:
+            return {}
+        import torch.distributed as dist
+        from shifu_amd.ops.embedding import MultiEmbedding
+        swapped = {}
+        for name, mod in sharded:
+            shard = mod.arena.data
+            world = mod.world
+            counts = [len(range(r, mod.total_rows, world)) for r in range(world)]
+            pad = shard.new_zeros(max(counts), mod.dim)
+            pad[:shard.shape[0]] = shard
+            outs = [torch.empty_like(pad) for _ in range(world)]
+            dist.all_gather(outs, pad)
+            if not self.is_chief:
+                continue
+            full = shard.new_empty(mod.total_rows, mod.dim)
+            for r in range(world):
+                full[r::world] = outs[r][:counts[r]]
+            rep = MultiEmbedding(mod.vocab_sizes, mod.dim, empty_init=True,
+                                 dtype=shard.dtype).to(shard.device)
+            rep.arena.data = full
+            parent = self.model
+            parts = name.split(".")
+            for part in parts[:-1]:
+                parent = getattr(parent, part)
+            swapped[name] = getattr(parent, parts[-1])
+            setattr(parent, parts[-1], rep)
+        return swapped

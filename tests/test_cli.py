@@ -65,3 +65,10 @@ def test_run_main_end_to_end(tmp_path):
     # progress board written
     board = (tmp_path / "logs" / "progress.board").read_text()
     assert "epoch 0:" in board and "epoch 1:" in board
+    # the export is self-contained: at world=2 the run used EP-sharded
+    # arenas, which must have been consolidated to full arenas for export
+    import torch
+    from shifu_amd.train.export import load_exported
+    m = load_exported(str(final))
+    p = m.predict(torch.zeros(1, 4), torch.tensor([[3, 7]]))
+    assert 0.0 <= float(p[0]) <= 1.0
