@@ -184,3 +184,65 @@ def test_ep_pair_gather_shared_routing():
     for rank, ok_fwd, ok_bwd in _run(_ep_pair_worker, 29725):
         assert ok_fwd, f"rank {rank}: pair forward mismatch"
         assert ok_bwd, f"rank {rank}: pair backward mismatch"
+
+
+def _ep_window_worker(rank, port, q):
+    """Window (local-SGD) mode with EP arenas: 4 steps syncing every 2nd must
+    produce the same model as the replicated-DP path — sparse grads accumulate
+    across the window on both paths."""
+    from shifu_amd.models.wide_deep import WideDeep
+    from shifu_amd.ops.flat import FlatParams, split_params
+    from shifu_amd.ops.loss import weighted_loss
+    from shifu_amd.ops.optim import FusedOptimizer
+    from shifu_amd.parallel.dist import GradAggregator
+    try:
+        _init(rank, port)
+        torch.manual_seed(0)
+        vocab = [50, 70]
+
+        def run(sharded):
+            model = WideDeep(4, vocab, 4, [8], ["relu"], seed=5,
+                             sharded_embeddings=sharded, world=WORLD, rank=rank)
+            dense_params, emb_params = split_params(model)
+            flat = FlatParams(dense_params)
+            agg = GradAggregator(flat, emb_params, bucket_mb=1)
+            opt = FusedOptimizer(flat, emb_params, optimizer="sgd", lr=0.05,
+                                 l2_reg=0.0, emb_optimizer="sgd", emb_lr=0.05)
+            for si in range(4):
+                sync = (si % 2 == 1)
+                agg.set_sync(sync)
+                g = torch.Generator().manual_seed(200 + 10 * rank + si)
+                dense = torch.randn(8, 4, generator=g)
+                cats = torch.randint(0, 50, (8, 2), generator=g)
+                y = (torch.rand(8, generator=g) > 0.5).float()
+                w = torch.ones(8)
+                loss = weighted_loss(model(dense, cats), y, w, "sigmoid_ce")
+                loss.backward()
+                if sync:
+                    agg.finish()
+                    opt.step()
+                    opt.zero_grad()
+            return model
+
+        m_ep = run(True)
+        m_dp = run(False)
+        ok_dense = all(torch.allclose(a.detach(), b.detach(), atol=1e-5)
+                       for (na, a), (nb, b) in zip(
+                           sorted(m_ep.named_parameters(), key=lambda kv: kv[0]),
+                           sorted(m_dp.named_parameters(), key=lambda kv: kv[0]))
+                       if "arena" not in na)
+        ok_emb = (torch.allclose(m_ep.embeddings.arena.data,
+                                 m_dp.embeddings.arena.data[rank::WORLD], atol=1e-5)
+                  and torch.allclose(m_ep.wide_cat.arena.data,
+                                     m_dp.wide_cat.arena.data[rank::WORLD], atol=1e-5))
+        q.put((rank, bool(ok_dense), bool(ok_emb)))
+        dist.barrier()
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ep_window_mode_matches_dp():
+    for rank, ok_dense, ok_emb in _run(_ep_window_worker, 29727):
+        assert ok_dense, f"rank {rank}: dense params diverged in window mode"
+        assert ok_emb, f"rank {rank}: embedding shards diverged in window mode"
