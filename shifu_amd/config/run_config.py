@@ -50,6 +50,11 @@ class RunConfig:
     epochs: Optional[int] = None             # overrides ModelConfig numTrainEpochs
     batch_size: Optional[int] = None         # per-rank batch size override
     dtype: str = "bf16"                      # compute dtype on GPU ("bf16"|"fp32")
+    data_residency: str = "auto"             # "auto"/"device": whole shard in
+                                             # HBM (fastest); "stream": shard
+                                             # stays in pinned host RAM, only
+                                             # batches cross PCIe (shards
+                                             # beyond-HBM datasets)
 
     # -- distributed knobs (xGMI-tuned) --
     bucket_mb: int = 128       # all-reduce bucket size; ring all-reduce over

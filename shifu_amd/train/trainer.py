@@ -91,6 +91,46 @@ class DeviceData:
                           self.target[idx], self.weight[idx])
 
 
+class StreamingData:
+    """Out-of-core residency (ROADMAP item 5): the shard stays in host RAM
+    (pinned when the target is a GPU) and only the current batch crosses PCIe.
+    Use when the per-rank shard exceeds what you want resident in the 288 GB
+    HBM (RunConfig.data_residency="stream").  Same interface as DeviceData;
+    `slice(idx)` takes HOST indices and returns a device-resident batch
+    (async H2D from pinned memory)."""
+
+    index_device = "cpu"     # epoch permutations index host-side
+
+    def __init__(self, ds: TabularDataset, device: torch.device,
+                 dense_dtype: torch.dtype):
+        self.device = device
+        pin = device.type == "cuda"
+
+        def host(t: torch.Tensor) -> torch.Tensor:
+            return t.pin_memory() if pin and t.numel() else t
+
+        self.dense = host(torch.from_numpy(ds.dense).to(dense_dtype))
+        self.cats = host(torch.from_numpy(ds.cats))
+        self.target = host(torch.from_numpy(ds.target))
+        self.weight = host(torch.from_numpy(ds.weight))
+
+    @classmethod
+    def from_dataset(cls, ds: TabularDataset, device: torch.device,
+                     dense_dtype: torch.dtype) -> "StreamingData":
+        return cls(ds, device, dense_dtype)
+
+    def __len__(self):
+        return self.target.shape[0]
+
+    def slice(self, idx) -> DeviceData:
+        dev = self.device
+        nb = dev.type == "cuda"
+        return DeviceData(self.dense[idx].to(dev, non_blocking=nb),
+                          self.cats[idx].to(dev, non_blocking=nb),
+                          self.target[idx].to(dev, non_blocking=nb),
+                          self.weight[idx].to(dev, non_blocking=nb))
+
+
 class Trainer:
     def __init__(self, model: torch.nn.Module, mc: ModelConfig, rc: RunConfig,
                  train_data: TabularDataset, valid_data: TabularDataset,
@@ -122,8 +162,10 @@ class Trainer:
                 if getattr(p, "_is_embedding_arena", False):
                     p.data = p.data.to(torch.bfloat16)
         self.dense_dtype = dense_dtype
-        self.train_data = DeviceData.from_dataset(train_data, self.device, dense_dtype)
-        self.valid_data = DeviceData.from_dataset(valid_data, self.device, dense_dtype)
+        residency = getattr(rc, "data_residency", "auto")
+        container = (StreamingData if residency == "stream" else DeviceData)
+        self.train_data = container.from_dataset(train_data, self.device, dense_dtype)
+        self.valid_data = container.from_dataset(valid_data, self.device, dense_dtype)
 
         dense_params, emb_params = split_params(self.model)
         from shifu_amd.ops.flat import bind_mirrors
@@ -151,11 +193,12 @@ class Trainer:
         # hipGraph-captured steps (full-size batches only; tail batches and
         # window mode run eager; capture failure falls back silently)
         g = getattr(rc, "graphs", "auto")
-        self.use_graphs = (g == "on" or (g == "auto" and
-                                         self.device.type == "cuda" and
-                                         world_size == 1 and
-                                         self.update_window == 1 and
-                                         not rc.enable_trace))
+        self.use_graphs = ((g == "on" or (g == "auto" and
+                                          self.device.type == "cuda" and
+                                          world_size == 1 and
+                                          self.update_window == 1 and
+                                          not rc.enable_trace))
+                           and not isinstance(self.train_data, StreamingData))
         self._graph = None
         self._gb: Optional[DeviceData] = None
 
@@ -295,7 +338,9 @@ class Trainer:
     def run_epoch(self, epoch: int) -> TrainingIntermediateResult:
         t0 = time.time()
         n = len(self.train_data)
-        perm = torch.from_numpy(self._rng.permutation(n)).to(self.device)
+        perm = torch.from_numpy(self._rng.permutation(n))
+        if getattr(self.train_data, "index_device", "device") != "cpu":
+            perm = perm.to(self.device)
         losses = []
         steps = range(0, n, self.batch_size)
         n_steps = len(steps)

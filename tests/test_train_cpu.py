@@ -271,3 +271,23 @@ def test_resume_equivalence_cpu(tmp_path):
     tr_res.fit()
     resumed = tr_res.evaluate(tr_res.valid_data)["loss"]
     assert abs(straight - resumed) < 1e-7, (straight, resumed)
+
+
+def test_streaming_residency_matches_device(tmp_path):
+    """data_residency="stream" (host-resident shard, per-batch transfer) must
+    train identically to the default device-resident path."""
+    train, valid = _data(n=400)
+    mc = _mc(epochs=2)
+
+    def run(residency, sub):
+        rc = RunConfig(tmp_model_path=str(tmp_path / sub / "ckpt"),
+                       final_model_path=str(tmp_path / sub / "final"),
+                       data_residency=residency)
+        model = ShifuMLP(10, [16, 8], ["relu", "relu"], seed=3)
+        tr = Trainer(model, mc, rc, train, valid)
+        tr.fit()
+        return tr.evaluate(tr.valid_data)["loss"]
+
+    resident = run("auto", "a")
+    streamed = run("stream", "b")
+    assert abs(resident - streamed) < 1e-7, (resident, streamed)

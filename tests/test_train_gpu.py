@@ -168,3 +168,30 @@ def test_config_fuzz_gpu(cfg):
         # correctness gate is "no blow-up" — short Adadelta/wMSE runs may
         # wobble a few % before descending
         assert last["loss"] <= first["loss"] * 1.25
+
+
+def test_streaming_residency_gpu():
+    """Host-pinned streaming residency trains like the HBM-resident path
+    (dense MLP: fully deterministic kernels -> tight tolerance)."""
+    dense, cats, target, weight = synthetic_arrays(2000, 9, (), seed=17)
+    full = TabularDataset(dense, cats, target, weight)
+    train, valid = full.split(0.2, seed=1)
+    mc = ModelConfig.from_dict({"train": {"numTrainEpochs": 2, "params": {
+        "NumHiddenLayers": 2, "NumHiddenNodes": [48, 24],
+        "ActivationFunc": ["relu", "relu"], "LearningRate": 0.01,
+        "Optimizer": "adam", "Loss": "sigmoid_ce",
+        "MiniBatchSize": 256, "L2Reg": 0.0}}})
+
+    def run(residency, td):
+        rc = RunConfig(tmp_model_path=td + "/c", final_model_path=td + "/f",
+                       data_residency=residency, graphs="off")
+        tr = Trainer(ShifuMLP(9, [48, 24], ["relu", "relu"], seed=3), mc, rc,
+                     train, valid, device=torch.device("cuda"))
+        tr.fit()
+        return tr.evaluate(tr.valid_data)["loss"]
+
+    with tempfile.TemporaryDirectory() as td:
+        resident = run("auto", td + "/a")
+        streamed = run("stream", td + "/b")
+    assert np.isfinite(streamed)
+    assert abs(resident - streamed) < 1e-3, (resident, streamed)
