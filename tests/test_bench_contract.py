@@ -41,3 +41,21 @@ def test_bench_torchrun_2rank_contract():
     assert d["n_gpus"] == 2
     assert d["config"]["global_batch"] == 64
     assert "ep2" in d["config"]["parallelism"]
+
+
+def test_bench_torchrun_2rank_dp_mode():
+    """--emb-mode dp at world=2: replicated arenas + sparse allgather
+    aggregation through the real bench script."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29653", "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "32", "--vocab", "1000", "--n-cat", "4",
+         "--emb-mode", "dp"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
