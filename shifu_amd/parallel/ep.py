@@ -69,15 +69,17 @@ class _EPGatherFn(torch.autograd.Function):
         n = flat_ids.numel()
         g = flat_ids.reshape(-1)
         owner = (g % world).to(torch.int64)
-        perm = torch.argsort(owner, stable=True)
-        ids_sorted = g[perm]
+        # launch the split-size exchange FIRST so it overlaps the sort; read
+        # the result back with ONE host transfer (per-element int() on device
+        # tensors would be one sync each — 2*world syncs per lookup)
         send_counts = torch.bincount(owner, minlength=world)
-
-        # exchange split sizes
         all_counts = [torch.zeros_like(send_counts) for _ in range(world)]
         dist.all_gather(all_counts, send_counts)
-        in_splits = [int(c) for c in send_counts]                 # what I send
-        out_splits = [int(all_counts[src][rank]) for src in range(world)]
+        perm = torch.argsort(owner, stable=True)
+        ids_sorted = g[perm]
+        cnt = torch.stack(all_counts).cpu()
+        in_splits = [int(c) for c in cnt[rank]]                   # what I send
+        out_splits = [int(cnt[src][rank]) for src in range(world)]
         m = sum(out_splits)
 
         recv_ids = torch.empty(m, dtype=g.dtype, device=g.device)
@@ -132,13 +134,15 @@ class _EPGatherPairFn(torch.autograd.Function):
         n = flat_ids.numel()
         g = flat_ids.reshape(-1)
         owner = (g % world).to(torch.int64)
-        perm = torch.argsort(owner, stable=True)
-        ids_sorted = g[perm]
+        # size exchange first (overlaps the sort); one host readback total
         send_counts = torch.bincount(owner, minlength=world)
         all_counts = [torch.zeros_like(send_counts) for _ in range(world)]
         dist.all_gather(all_counts, send_counts)
-        in_splits = [int(c) for c in send_counts]
-        out_splits = [int(all_counts[src][rank]) for src in range(world)]
+        perm = torch.argsort(owner, stable=True)
+        ids_sorted = g[perm]
+        cnt = torch.stack(all_counts).cpu()
+        in_splits = [int(c) for c in cnt[rank]]
+        out_splits = [int(cnt[src][rank]) for src in range(world)]
         m = sum(out_splits)
 
         recv_ids = torch.empty(m, dtype=g.dtype, device=g.device)
