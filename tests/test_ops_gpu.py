@@ -432,3 +432,30 @@ def test_gather_concat_fused():
     d2 = torch.zeros(120, 8, device="cuda")
     d2.index_add_(0, r2, v2.float())
     assert torch.allclose(d1.cpu(), d2.cpu(), atol=1e-2)
+
+
+def test_splitk_atomic_fallback_mode():
+    """SHIFU_SPLITK_SLAB=0 restores the f32-atomic split-K epilogue; it must
+    stay numerically correct (the knob is read once per process, hence the
+    subprocess)."""
+    import os
+    import subprocess
+    import sys
+    code = (
+        "import torch\n"
+        "from shifu_amd.ops.dispatch import hip_ops\n"
+        "ext = hip_ops(); torch.manual_seed(0)\n"
+        "for (M, N, K) in [(300, 520, 10000), (512, 1024, 8192)]:\n"
+        "    a = torch.randn(M, K, device='cuda').to(torch.bfloat16)\n"
+        "    b = torch.randn(N, K, device='cuda').to(torch.bfloat16)\n"
+        "    want = a.float() @ b.float().t()\n"
+        "    got = ext.gemm_ntv3_f32(a, b)\n"
+        "    rel = float((got - want).abs().max() / want.abs().max())\n"
+        "    assert rel < 1e-4, (M, N, K, rel)\n"
+        "print('ATOMIC_OK')\n"
+    )
+    env = dict(os.environ, SHIFU_SPLITK_SLAB="0")
+    r = subprocess.run([sys.executable, "-c", code], env=env,
+                       capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "ATOMIC_OK" in r.stdout
