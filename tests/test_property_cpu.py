@@ -65,3 +65,54 @@ def test_weighted_loss_properties(n, kind, seed, wzero):
     # increase the loss (dL/dz <= 0 where y=1) for sigmoid_ce
     if kind == "sigmoid_ce" and not wzero:
         assert torch.all(z.grad[y == 1.0] <= 1e-7)
+
+
+@settings(max_examples=20, deadline=None)
+@given(b=st.integers(1, 17), f=st.integers(1, 6), d=st.integers(1, 9),
+       seed=st.integers(0, 10_000))
+def test_fm_second_order_equals_pairwise(b, f, d, seed):
+    """fm2[b] must equal the brute-force sum of pairwise dot products
+    sum_{i<j} <v_i, v_j> (the identity the fused form exploits)."""
+    from shifu_amd.ops.fm import fm_second_order
+    torch.manual_seed(seed)
+    v = torch.randn(b, f, d, requires_grad=True)
+    out = fm_second_order(v.reshape(b, f * d), f, d)
+    brute = torch.zeros(b)
+    for i in range(f):
+        for j in range(i + 1, f):
+            brute += (v[:, i] * v[:, j]).sum(dim=1)
+    assert torch.allclose(out, brute, atol=1e-4), (b, f, d)
+    out.sum().backward()
+    v2 = v.detach().clone().requires_grad_(True)
+    brute2 = torch.zeros(b)
+    for i in range(f):
+        for j in range(i + 1, f):
+            brute2 += (v2[:, i] * v2[:, j]).sum(dim=1)
+    brute2.sum().backward()
+    assert torch.allclose(v.grad, v2.grad, atol=1e-4)
+
+
+@settings(max_examples=20, deadline=None)
+@given(b=st.integers(1, 20), d=st.integers(1, 12), seed=st.integers(0, 10_000))
+def test_multi_embedding_matches_torch_embedding(b, d, seed):
+    from shifu_amd.ops.embedding import MultiEmbedding
+    torch.manual_seed(seed)
+    vocab = [7, 13]
+    emb = MultiEmbedding(vocab, d, seed=seed)
+    t0 = torch.nn.Embedding(7, d)
+    t1 = torch.nn.Embedding(13, d)
+    with torch.no_grad():
+        t0.weight.copy_(emb.arena[:7])
+        t1.weight.copy_(emb.arena[7:])
+    g = torch.Generator().manual_seed(seed + 1)
+    ids = torch.stack([torch.randint(0, 7, (b,), generator=g),
+                       torch.randint(0, 13, (b,), generator=g)], dim=1)
+    out = emb(ids)
+    ref = torch.cat([t0(ids[:, 0]), t1(ids[:, 1])], dim=1)
+    assert torch.allclose(out, ref, atol=1e-6)
+    gr = torch.randn_like(out)
+    out.backward(gr)
+    ref.backward(gr)
+    dense = emb.arena.grad.coalesce().to_dense()
+    assert torch.allclose(dense[:7], t0.weight.grad, atol=1e-6)
+    assert torch.allclose(dense[7:], t1.weight.grad, atol=1e-6)
