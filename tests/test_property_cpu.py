@@ -83,6 +83,10 @@ def test_fm_second_order_equals_pairwise(b, f, d, seed):
             brute += (v[:, i] * v[:, j]).sum(dim=1)
     assert torch.allclose(out, brute, atol=1e-4), (b, f, d)
     out.sum().backward()
+    if f == 1:
+        # no pairs: fm2 == 0 and its gradient vanishes (s - v == 0)
+        assert torch.allclose(v.grad, torch.zeros_like(v.grad), atol=1e-5)
+        return
     v2 = v.detach().clone().requires_grad_(True)
     brute2 = torch.zeros(b)
     for i in range(f):
