@@ -64,3 +64,25 @@ def test_native_throughput(tmp_path):
     t_python = time.time() - t0
     assert len(a) == len(b) == 60000
     assert t_native < t_python / 3, f"native {t_native:.2f}s vs python {t_python:.2f}s"
+
+
+def test_python_fallback_when_ext_missing(tmp_path, monkeypatch):
+    """With the extension forced absent, load_csv_native must transparently
+    fall back to the pure-Python parser with identical results."""
+    import shifu_amd.io as io_mod
+    from shifu_amd.data.synthetic import generate_synthetic_csv
+    paths = generate_synthetic_csv(str(tmp_path), n_rows=200, n_dense=3,
+                                   vocab_sizes=[9], n_files=2, seed=4)
+    cols = dict(selected_numeric=[2, 3, 4], selected_categorical=[5],
+                target_column=0, weight_column=1)
+    native = load_csv_native(paths, **cols)
+    monkeypatch.setattr(io_mod, "_EXT", None)
+    monkeypatch.setattr(io_mod, "_TRIED", True)
+    fallback = io_mod.load_csv_native(paths, **cols)
+    assert io_mod.native_io() is None
+    assert len(fallback) == len(native)
+    import numpy as np
+    np.testing.assert_allclose(fallback.dense, native.dense, atol=1e-6)
+    np.testing.assert_array_equal(fallback.cats, native.cats)
+    np.testing.assert_allclose(fallback.target, native.target, atol=1e-6)
+    np.testing.assert_allclose(fallback.weight, native.weight, atol=1e-6)
