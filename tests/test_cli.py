@@ -72,3 +72,42 @@ def test_run_main_end_to_end(tmp_path):
     m = load_exported(str(final))
     p = m.predict(torch.zeros(1, 4), torch.tensor([[3, 7]]))
     assert 0.0 <= float(p[0]) <= 1.0
+
+
+def test_score_cli_end_to_end(tmp_path, capfd):
+    """Batch-scoring CLI over an exported model: scores every row, valid
+    probabilities, AUC reported against the target column."""
+    import torch
+    from shifu_amd.models.wide_deep import WideDeep
+    from shifu_amd.train.export import export_model
+    from shifu_amd.score import main as score_main
+
+    data_dir = str(tmp_path / "data")
+    generate_synthetic_csv(data_dir, n_rows=400, n_dense=4, vocab_sizes=[20, 30],
+                           n_files=2, seed=11)
+    cc = [{"columnNum": 0, "columnName": "target", "columnFlag": "Target"},
+          {"columnNum": 1, "columnName": "w", "columnFlag": "Weight"}]
+    cc += [{"columnNum": i, "columnName": f"d{i}", "finalSelect": True,
+            "columnType": "N"} for i in range(2, 6)]
+    cc += [{"columnNum": 6, "columnName": "c0", "finalSelect": True,
+            "columnType": "C", "vocabSize": 20},
+           {"columnNum": 7, "columnName": "c1", "finalSelect": True,
+            "columnType": "C", "vocabSize": 30}]
+    cc_path = str(tmp_path / "ColumnConfig.json")
+    with open(cc_path, "w") as f:
+        json.dump(cc, f)
+
+    model = WideDeep(4, [20, 30], 4, [8], ["relu"], seed=2)
+    export_model(model, str(tmp_path / "final"))
+
+    out_path = str(tmp_path / "scores.csv")
+    rc = score_main(["--model", str(tmp_path / "final"), "--data", data_dir,
+                     "--column-config", cc_path, "--output", out_path, "--auc"])
+    assert rc == 0
+    lines = open(out_path).read().strip().splitlines()
+    assert len(lines) == 400
+    vals = [float(l) for l in lines]
+    assert all(0.0 <= v <= 1.0 for v in vals)
+    err = capfd.readouterr().err
+    summary = json.loads(err.strip().splitlines()[-1])
+    assert summary["rows"] == 400 and "auc" in summary
