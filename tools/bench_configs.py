@@ -115,7 +115,7 @@ def _emit(name, samples, elapsed, steps, extra):
 
 def main():
     ap = argparse.ArgumentParser()
-    ap.add_argument("--config", required=True, choices=["2", "4", "5", "5io"])
+    ap.add_argument("--config", required=True, choices=["1", "2", "4", "5", "5io"])
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=8192)
@@ -126,6 +126,42 @@ def main():
     device = "cuda" if on_gpu else "cpu"
     dtype = torch.bfloat16 if on_gpu else torch.float32
     torch.manual_seed(1)
+
+    if args.config == "1":
+        # config 1: 3-layer MLP on 1k-row synthetic CSV, CPU world_size=1 —
+        # exercises the FULL plumbing (CSV parse, ModelConfig, Trainer,
+        # checkpoint, export) and reports end-to-end wall time.
+        import tempfile
+        from shifu_amd.config.model_config import ModelConfig
+        from shifu_amd.config.run_config import RunConfig
+        from shifu_amd.data.synthetic import generate_synthetic_csv
+        from shifu_amd.io import load_csv_native
+        from shifu_amd.models.mlp import ShifuMLP
+        from shifu_amd.train.trainer import Trainer
+        mc = ModelConfig.from_dict({"train": {"numTrainEpochs": 3, "params": {
+            "NumHiddenLayers": 3, "NumHiddenNodes": [64, 32, 16],
+            "ActivationFunc": ["relu"] * 3, "LearningRate": 0.01,
+            "Optimizer": "adam", "Loss": "sigmoid_ce", "MiniBatchSize": 100,
+            "L2Reg": 0.0}}})
+        with tempfile.TemporaryDirectory() as td:
+            paths = generate_synthetic_csv(td + "/data", n_rows=1000,
+                                           n_dense=30, n_files=1, seed=1)
+            t0 = time.time()
+            ds = load_csv_native(paths, list(range(2, 32)), [], 0, 1)
+            train, valid = ds.split(0.2, seed=1)
+            rc = RunConfig(tmp_model_path=td + "/ckpt",
+                           final_model_path=td + "/final", device="cpu")
+            tr = Trainer(ShifuMLP(30, [64, 32, 16], ["relu"] * 3, seed=1),
+                         mc, rc, train, valid)
+            tr.fit()
+            el = time.time() - t0
+            auc = tr.evaluate(tr.valid_data)["auc"]
+        print(json.dumps({
+            "metric": "config1_cpu_e2e_seconds", "value": el, "unit": "s",
+            "higher_is_better": False, "data": "synthetic",
+            "config": {"model": "mlp[30->64,32,16]", "rows": 1000,
+                       "epochs": 3, "auc": auc}}), flush=True)
+        return
 
     if args.config == "2":
         from shifu_amd.models.mlp import ShifuMLP
