@@ -120,3 +120,40 @@ def test_multi_embedding_matches_torch_embedding(b, d, seed):
     dense = emb.arena.grad.coalesce().to_dense()
     assert torch.allclose(dense[:7], t0.weight.grad, atol=1e-6)
     assert torch.allclose(dense[7:], t1.weight.grad, atol=1e-6)
+
+
+@settings(max_examples=15, deadline=None)
+@given(n=st.integers(1, 60), nd=st.integers(1, 5), seed=st.integers(0, 10_000),
+       gz=st.booleans())
+def test_csv_parsers_agree(tmp_path_factory, n, nd, seed, gz):
+    """Fuzz: the native C++ reader and the pure-Python parser must agree on
+    arbitrary well-formed rows (values spanning magnitudes and signs)."""
+    import os
+    from shifu_amd.data.csv_loader import load_csv_files
+    from shifu_amd.io import load_csv_native, native_io
+    rng = np.random.default_rng(seed)
+    rows = []
+    for _ in range(n):
+        t = float(rng.integers(0, 2))
+        w = float(abs(rng.standard_normal()))
+        feats = rng.standard_normal(nd) * (10.0 ** rng.integers(-6, 7, nd))
+        cat = int(rng.integers(0, 1000))
+        rows.append([t, w] + [float(x) for x in feats] + [cat])
+    td = tmp_path_factory.mktemp("csv")
+    path = os.path.join(str(td), "part.csv" + (".gz" if gz else ""))
+    import gzip as gzmod
+    op = (lambda p: gzmod.open(p, "wt")) if gz else (lambda p: open(p, "w"))
+    with op(path) as f:
+        for r in rows:
+            f.write("|".join(repr(v) for v in r) + "\n")
+    cols = dict(selected_numeric=list(range(2, 2 + nd)),
+                selected_categorical=[2 + nd], target_column=0, weight_column=1)
+    py = load_csv_files([path], **cols)
+    nat = load_csv_native([path], **cols)
+    if native_io() is None:
+        pytest.skip("native reader not built")
+    assert len(py) == len(nat) == n
+    np.testing.assert_allclose(nat.dense, py.dense, rtol=1e-6, atol=1e-30)
+    np.testing.assert_array_equal(nat.cats, py.cats)
+    np.testing.assert_allclose(nat.target, py.target, atol=1e-6)
+    np.testing.assert_allclose(nat.weight, py.weight, rtol=1e-6)
