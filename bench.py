@@ -74,7 +74,15 @@ def main():
     args = ap.parse_args()
 
     rank, world, device = init_distributed()
-    n_gpus = world if world > 1 else args.gpus
+    if args.gpus != 1 and args.gpus != world:
+        # never over-claim GPUs: a single process with --gpus 8 would report
+        # 8x inflated samples/s.  The driver launches N>1 via torchrun, so
+        # world (the rank count actually doing work) is the only truth.
+        print(f"bench.py: --gpus {args.gpus} but world_size is {world}; "
+              f"launch N>1 via torch.distributed.run (one rank per GPU)",
+              file=sys.stderr, flush=True)
+        sys.exit(2)
+    n_gpus = world
     on_gpu = device.type == "cuda"
     dtype = torch.bfloat16 if on_gpu else torch.float32
 
@@ -159,11 +167,14 @@ def main():
         torch.distributed.barrier()
     elapsed = time.time() - t0
 
-    # max elapsed over ranks
+    # max elapsed over ranks (+ min, for the straggler spread)
+    elapsed_min = elapsed
     if is_distributed():
         t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
-        elapsed = float(t)
+        tmin = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        torch.distributed.all_reduce(tmin, op=torch.distributed.ReduceOp.MIN)
+        elapsed, elapsed_min = float(t), float(tmin)
 
     if rank == 0:
         samples = n_gpus * args.batch * args.steps
@@ -175,6 +186,7 @@ def main():
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": elapsed / args.steps * 1000.0,
+            "rank_spread_ms": (elapsed - elapsed_min) * 1000.0,
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no numbers

@@ -60,8 +60,11 @@ class RunConfig:
     bucket_mb: int = 128       # all-reduce bucket size; ring all-reduce over
                                # 7x153GB/s p2p links is per-link bound -> big buckets
     overlap_allreduce: bool = True
-    quorum_ratio: float = 1.0  # REPLICAS_TO_AGGREGATE_RATIO analog (ssgd.py:19) — kept
-                               # for API parity; sync all-reduce uses all ranks.
+    quorum_ratio: float = 1.0  # REPLICAS_TO_AGGREGATE_RATIO analog (ssgd.py:19;
+                               # SAGN.py:161 uses 0.9): q < 1 aggregates each
+                               # sync step from only ceil(q*world) ranks on a
+                               # deterministic rotation (parallel/dist.py
+                               # GradAggregator); requires emb_mode="dp".
 
     # -- robustness (successor of heartbeat/backup machinery) --
     heartbeat_interval_s: float = 1.0    # shifu.task.heartbeat-interval default 1000ms

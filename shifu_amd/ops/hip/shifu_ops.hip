@@ -1404,7 +1404,10 @@ __global__ void loss_fwd_kernel(const bf16* __restrict__ z, const float* __restr
       per = wf * (fmaxf(zf, 0.0f) - zf * yf + log1pf(__expf(-fabsf(zf))));
     }
     ls += per;
-    ws += wf;
+    // TF loss-reduction SUM_BY_NONZERO_WEIGHTS (the reference's
+    // tf.losses.mean_squared_error default): normalize by the COUNT of
+    // nonzero-weight samples, not sum(w)
+    ws += (wf != 0.0f) ? 1.0f : 0.0f;
   }
   // wave reduce
 #pragma unroll

@@ -8,12 +8,14 @@ the loss kernel, which is both faster (one pass over
 [B]-sized data) and numerically stable for the cross-entropy variant.
 
 Kinds:
-* "weighted_mse": L = sum_i w_i (sigmoid(z_i) - y_i)^2 / sum_i w_i
-* "sigmoid_ce" : L = sum_i w_i BCE(sigmoid(z_i), y_i) / sum_i w_i
+* "weighted_mse": L = sum_i w_i (sigmoid(z_i) - y_i)^2 / #{i: w_i != 0}
+* "sigmoid_ce" : L = sum_i w_i BCE(sigmoid(z_i), y_i) / #{i: w_i != 0}
 
-Both are normalized by sum(w) (weighted mean).  The reference's TF default
-(SUM_BY_NONZERO_WEIGHTS) divides by the count of nonzero weights instead;
-with the all-ones default weights the two coincide.
+Both normalize by the COUNT of nonzero-weight samples — TF's
+SUM_BY_NONZERO_WEIGHTS, the reference's tf.losses.mean_squared_error
+default reduction — so training with a real weight column keeps the same
+loss scale and effective learning rate as the reference.  With the all-ones
+default weights this equals the plain mean.
 """
 from __future__ import annotations
 
@@ -33,6 +35,7 @@ class _WeightedLossFn(torch.autograd.Function):
         z1 = z.reshape(-1)
         if use_hip(z1):
             ext = hip_ops()
+            # third output = count of nonzero weights (SUM_BY_NONZERO_WEIGHTS)
             p, loss_sum, wsum = ext.weighted_loss_fwd(
                 z1.contiguous(), y.contiguous(), w.contiguous(), kind)
             loss = loss_sum / wsum.clamp_min(1e-12)
@@ -45,7 +48,7 @@ class _WeightedLossFn(torch.autograd.Function):
             else:
                 per = wf * torch.nn.functional.binary_cross_entropy_with_logits(
                     zf, yf, reduction="none")
-            wsum = wf.sum()
+            wsum = (wf != 0).float().sum()
             loss = per.sum() / wsum.clamp_min(1e-12)
         ctx.save_for_backward(p, y, w, wsum if torch.is_tensor(wsum) else torch.tensor(wsum))
         ctx.kind = kind

@@ -151,6 +151,7 @@ class Trainer:
         self._hb_last = time.time()
         self._ckpt_secs = float(getattr(rc, "checkpoint_every_secs", 0.0))
         self._ckpt_last = time.time()
+        self._has_sharded = False  # set after model construction below
 
         self.model = model.to(self.device)
         dense_dtype = torch.bfloat16 if (self.device.type == "cuda"
@@ -167,6 +168,18 @@ class Trainer:
         self.train_data = container.from_dataset(train_data, self.device, dense_dtype)
         self.valid_data = container.from_dataset(valid_data, self.device, dense_dtype)
 
+        self._has_sharded = any(getattr(p, "_is_ep_sharded", False)
+                                for p in self.model.parameters())
+        if self._ckpt_secs > 0 and self._has_sharded:
+            # time-based triggers are rank-local clocks; EP shards saved at
+            # different steps would be mutually inconsistent.  Epoch-cadence
+            # saves (rank-synchronized) still run.
+            if self.is_chief:
+                print("shifu_amd: checkpoint_every_secs is disabled for "
+                      "EP-sharded embeddings (epoch-cadence saves only)",
+                      flush=True)
+            self._ckpt_secs = 0.0
+
         dense_params, emb_params = split_params(self.model)
         from shifu_amd.ops.flat import bind_mirrors
         self.flat = FlatParams(dense_params,
@@ -175,7 +188,8 @@ class Trainer:
         self.emb_params = emb_params
         self.aggregator = GradAggregator(self.flat, emb_params,
                                          bucket_mb=rc.bucket_mb,
-                                         overlap=rc.overlap_allreduce)
+                                         overlap=rc.overlap_allreduce,
+                                         quorum_ratio=rc.quorum_ratio)
         p = mc.params
         self.optimizer = FusedOptimizer(
             self.flat, emb_params, optimizer=p.optimizer, lr=p.learning_rate,
@@ -327,7 +341,7 @@ class Trainer:
             else:
                 per = w * (p - b.target.float()) ** 2
             total += float(per.sum())
-            wtot += float(w.sum())
+            wtot += float((w != 0).sum())  # SUM_BY_NONZERO_WEIGHTS (ops/loss.py)
             scores.append(p.cpu().numpy())
         self.model.train()
         scores = np.concatenate(scores)
@@ -338,6 +352,12 @@ class Trainer:
     def run_epoch(self, epoch: int) -> TrainingIntermediateResult:
         t0 = time.time()
         n = len(self.train_data)
+        # snapshot the RNG state and step count BEFORE drawing this epoch's
+        # permutation: a mid-epoch checkpoint (tagged epoch-1) must let the
+        # resumed run redraw the SAME permutation and recount from the
+        # epoch-start step
+        rng_pre = self._rng.bit_generator.state
+        step_pre = self.global_step
         perm = torch.from_numpy(self._rng.permutation(n))
         if getattr(self.train_data, "index_device", "device") != "cpu":
             perm = perm.to(self.device)
@@ -350,20 +370,21 @@ class Trainer:
             if (self.use_graphs and sync and idx.shape[0] == self.batch_size
                     and self._ensure_graph()):
                 losses.append(self._graphed_step(idx))
-                continue
-            batch = self.train_data.slice(idx)
-            # window mode: only every update_window-th (or last) step syncs+updates
-            losses.append(self.train_step(batch, sync=sync))
+            else:
+                batch = self.train_data.slice(idx)
+                # window mode: only every update_window-th (or last) step
+                # syncs+updates
+                losses.append(self.train_step(batch, sync=sync))
             if self.heartbeat and time.time() - self._hb_last >= self._hb_interval:
                 self.heartbeat()
                 self._hb_last = time.time()
-            if (self._ckpt_secs > 0 and self.is_chief and epoch > 0
+            if (self._ckpt_secs > 0
                     and time.time() - self._ckpt_last >= self._ckpt_secs):
                 # mid-epoch time-based save, tagged as the last COMPLETE epoch
-                # so resume replays the current epoch from its start
-                ckpt.save_checkpoint(self.rc.tmp_model_path, epoch - 1,
-                                     self.global_step, self.model, self.optimizer,
-                                     extra={"np_rng": self._rng.bit_generator.state})
+                # (epoch-1 == -1 resumes epoch 0 correctly) with the pre-epoch
+                # RNG/step snapshot so resume replays this epoch from its start
+                self._save_checkpoint(epoch - 1, global_step=step_pre,
+                                      np_rng=rng_pre)
                 self._ckpt_last = time.time()
         mean_loss = float(torch.stack(losses).float().mean()) if losses else 0.0
         if self.device.type == "cuda":
@@ -384,10 +405,23 @@ class Trainer:
             container_id=f"rank-{self.rank}",
         )
 
+    def _save_checkpoint(self, epoch: int, global_step: Optional[int] = None,
+                         np_rng=None) -> None:
+        """EP-aware save: every rank writes its shard, rank 0 the main file
+        (train/checkpoint.py layout)."""
+        gs = self.global_step if global_step is None else global_step
+        rng = self._rng.bit_generator.state if np_rng is None else np_rng
+        ckpt.save_checkpoint(self.rc.tmp_model_path, epoch, gs,
+                             self.model, self.optimizer,
+                             extra={"np_rng": rng},
+                             rank=self.rank, world=self.world)
+
     def maybe_resume(self) -> None:
-        path = ckpt.latest_checkpoint(self.rc.tmp_model_path)
+        path = ckpt.latest_checkpoint(self.rc.tmp_model_path, world=self.world)
         if path:
-            info = ckpt.load_checkpoint(path, self.model, self.optimizer, self.device)
+            info = ckpt.load_checkpoint(path, self.model, self.optimizer,
+                                        self.device, rank=self.rank,
+                                        world=self.world)
             self.flat.refresh_mirror()
             self.start_epoch = int(info["epoch"]) + 1
             self.global_step = int(info["global_step"])
@@ -410,10 +444,10 @@ class Trainer:
             results.append(r)
             if self.metric_sink:
                 self.metric_sink(r)
-            if self.is_chief and (epoch + 1) % self.rc.checkpoint_every_epochs == 0:
-                ckpt.save_checkpoint(self.rc.tmp_model_path, epoch,
-                                     self.global_step, self.model, self.optimizer,
-                                     extra={"np_rng": self._rng.bit_generator.state})
+            if (epoch + 1) % self.rc.checkpoint_every_epochs == 0:
+                # all ranks: EP shards are per-rank; non-chief without shards
+                # is a no-op inside save_checkpoint
+                self._save_checkpoint(epoch)
             if is_distributed():
                 torch.distributed.barrier()
         if self.tracer.enabled:

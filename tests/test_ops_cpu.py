@@ -62,13 +62,14 @@ def test_weighted_loss_value_and_grad(kind):
     loss.backward()
 
     p = torch.sigmoid(z.detach())
+    nnz = (w != 0).float().sum()  # TF SUM_BY_NONZERO_WEIGHTS normalization
     if kind == "weighted_mse":
-        expect = (w * (p - y) ** 2).sum() / w.sum()
-        dz = w * 2 * (p - y) * p * (1 - p) / w.sum()
+        expect = (w * (p - y) ** 2).sum() / nnz
+        dz = w * 2 * (p - y) * p * (1 - p) / nnz
     else:
         expect = (w * torch.nn.functional.binary_cross_entropy_with_logits(
-            z.detach(), y, reduction="none")).sum() / w.sum()
-        dz = w * (p - y) / w.sum()
+            z.detach(), y, reduction="none")).sum() / nnz
+        dz = w * (p - y) / nnz
     assert torch.allclose(loss, expect, atol=1e-6)
     assert torch.allclose(z.grad, dz, atol=1e-6)
 

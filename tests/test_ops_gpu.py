@@ -253,7 +253,8 @@ def test_weighted_loss(kind):
         per = w * torch.nn.functional.binary_cross_entropy_with_logits(zf, y, reduction="none")
     assert torch.allclose(p, pref, atol=1e-5)
     assert abs(float(ls) - float(per.sum())) < 1e-2 * max(float(per.sum()), 1.0)
-    assert abs(float(ws) - float(w.sum())) < 1e-3 * float(w.sum())
+    nnz = float((w != 0).sum())  # TF SUM_BY_NONZERO_WEIGHTS count
+    assert abs(float(ws) - nnz) < 1e-3 * nnz
 
     dz = hip_ops().weighted_loss_bwd(p, y, w, kind,
                                      torch.tensor([0.125], device="cuda"))
