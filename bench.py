@@ -68,9 +68,12 @@ def main():
     ap.add_argument("--n-cat", type=int, default=N_CAT)
     ap.add_argument("--graphs", choices=["auto", "on", "off"], default="auto",
                     help="hipGraph-capture the whole training step")
-    ap.add_argument("--emb-mode", choices=["auto", "dp", "ep"], default="auto",
-                    help="embedding parallelism: replicated+sparse-allgather (dp) "
-                         "or sharded+all-to-all (ep); auto=ep when world>1")
+    ap.add_argument("--emb-mode", choices=["auto", "dp", "ep", "ep_row"],
+                    default="auto",
+                    help="embedding parallelism: replicated+sparse-allgather "
+                         "(dp), feature-sharded static all-to-all (ep, the "
+                         "default at world>1), or row%%world sharding "
+                         "(ep_row)")
     args = ap.parse_args()
 
     rank, world, device = init_distributed()
@@ -87,9 +90,13 @@ def main():
     dtype = torch.bfloat16 if on_gpu else torch.float32
 
     torch.manual_seed(777)
-    use_ep = (args.emb_mode == "ep" or (args.emb_mode == "auto" and world > 1))
+    mode = args.emb_mode
+    if mode == "auto":
+        mode = "ep" if world > 1 else "dp"
+    sharded = {"ep": "table", "ep_row": "row", "dp": False}[mode]
+    use_ep = bool(sharded)
     model = WideDeep(N_DENSE, [args.vocab] * args.n_cat, args.embed_dim, TOWER, ACTS,
-                     seed=777, sharded_embeddings=use_ep, world=world,
+                     seed=777, sharded_embeddings=sharded, world=world,
                      rank=rank, emb_fast_init=True).to(device)
     if on_gpu:
         # keep embedding arenas bf16 (HBM-resident, gathered by the HIP kernel)
@@ -197,8 +204,8 @@ def main():
                 "global_batch": n_gpus * args.batch,
                 "per_gpu_batch": args.batch,
                 "seq_len": None,
-                "parallelism": (f"dp{n_gpus}+ep{n_gpus}(emb)" if use_ep
-                                else f"dp{n_gpus}"),
+                "parallelism": (f"dp{n_gpus}+ep{n_gpus}(emb:{sharded})"
+                                if use_ep else f"dp{n_gpus}"),
                 "optimizer": "adam+rowwise_adagrad(emb)",
                 "loss": "sigmoid_ce",
                 "hipgraphs": use_graphs,

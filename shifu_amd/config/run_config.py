@@ -37,10 +37,14 @@ class RunConfig:
     # -- model family --
     model_type: str = "auto"   # "mlp" | "wide_deep" | "deepfm" | "auto"
     embed_dim: int = 16
-    emb_mode: str = "auto"     # "ep" shards arenas row%world with all-to-all
-                               # routing (parallel/ep.py); "dp" replicates +
-                               # sparse-allgathers grads; "auto" = ep at
-                               # world>1 (the xGMI-native choice)
+    emb_mode: str = "auto"     # "ep"/"ep_table": feature-sharded arenas with
+                               # STATIC all-to-all splits (parallel/ep.py
+                               # TableShardedEmbedding — no per-step host
+                               # sync, graph-capturable); "ep_row": row%world
+                               # sharding (single table > 1 GPU's HBM);
+                               # "dp": replicated + sparse-allgather grads;
+                               # "auto" = ep at world>1 (the xGMI-native
+                               # choice), dp when quorum_ratio < 1
 
     # -- model / training --
     model_config_path: Optional[str] = None

@@ -28,13 +28,9 @@ class DeepFM(torch.nn.Module):
         self.vocab_sizes = list(vocab_sizes)
         F = len(self.vocab_sizes)
 
-        if sharded_embeddings and world > 1:
-            from shifu_amd.parallel.ep import ShardedEmbedding
-            emb = lambda d, s: ShardedEmbedding(self.vocab_sizes, d, seed=s,
-                                                world=world, rank=rank,
-                                                fast_init=emb_fast_init)
-        else:
-            emb = lambda d, s: MultiEmbedding(self.vocab_sizes, d, seed=s)
+        from shifu_amd.models.wide_deep import _emb_factory
+        emb = _emb_factory(self.vocab_sizes, sharded_embeddings, world, rank,
+                           emb_fast_init)
 
         self.fm_first = emb(1, seed + 11)
         self.fm_dense = FusedLinear(num_dense, 1, activation="none", seed=seed + 12)
@@ -51,10 +47,15 @@ class DeepFM(torch.nn.Module):
     def forward(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
         B = dense.shape[0]
         F = len(self.vocab_sizes)
-        from shifu_amd.parallel.ep import ShardedEmbedding, ep_pair_gather
+        from shifu_amd.parallel.ep import (ShardedEmbedding,
+                                           TableShardedEmbedding,
+                                           ep_pair_gather, table_pair_gather)
         from shifu_amd.ops.fm import fm_second_order
-        if isinstance(self.fm_first, ShardedEmbedding):
-            fm1_e, emb_flat = ep_pair_gather(self.fm_first, self.embeddings, cats)
+        if isinstance(self.fm_first, (ShardedEmbedding, TableShardedEmbedding)):
+            pair = (table_pair_gather
+                    if isinstance(self.fm_first, TableShardedEmbedding)
+                    else ep_pair_gather)
+            fm1_e, emb_flat = pair(self.fm_first, self.embeddings, cats)
             fm1 = fm1_e.sum(dim=1) + self.fm_dense(dense).reshape(-1)
             fm2 = fm_second_order(emb_flat, F, self.embed_dim)
             x = torch.cat([dense, emb_flat.to(dense.dtype)], dim=1)

@@ -21,6 +21,27 @@ from shifu_amd.ops.linear import FusedLinear
 from shifu_amd.ops.loss import predict_proba
 
 
+def _emb_factory(vocab_sizes, sharded, world: int, rank: int, fast_init: bool):
+    """sharded: False -> replicated MultiEmbedding; "table"/True -> feature-
+    sharded TableShardedEmbedding (static all-to-all splits, the default EP
+    mode); "row" -> row%world ShardedEmbedding (for a single table too big
+    for one GPU)."""
+    if not sharded or world <= 1:
+        return lambda d, s: MultiEmbedding(vocab_sizes, d, seed=s)
+    mode = "table" if sharded is True else str(sharded)
+    if mode == "table":
+        from shifu_amd.parallel.ep import TableShardedEmbedding
+        return lambda d, s: TableShardedEmbedding(vocab_sizes, d, seed=s,
+                                                  world=world, rank=rank,
+                                                  fast_init=fast_init)
+    if mode == "row":
+        from shifu_amd.parallel.ep import ShardedEmbedding
+        return lambda d, s: ShardedEmbedding(vocab_sizes, d, seed=s,
+                                             world=world, rank=rank,
+                                             fast_init=fast_init)
+    raise ValueError(f"unknown sharded_embeddings mode {sharded!r}")
+
+
 class WideDeep(torch.nn.Module):
     def __init__(self, num_dense: int, vocab_sizes: Sequence[int], embed_dim: int,
                  hidden_nodes: List[int], activations: List[str], seed: int = 1234,
@@ -32,13 +53,8 @@ class WideDeep(torch.nn.Module):
         self.vocab_sizes = list(vocab_sizes)
         F = len(self.vocab_sizes)
 
-        if sharded_embeddings and world > 1:
-            from shifu_amd.parallel.ep import ShardedEmbedding
-            emb = lambda d, s: ShardedEmbedding(self.vocab_sizes, d, seed=s,
-                                                world=world, rank=rank,
-                                                fast_init=emb_fast_init)
-        else:
-            emb = lambda d, s: MultiEmbedding(self.vocab_sizes, d, seed=s)
+        emb = _emb_factory(self.vocab_sizes, sharded_embeddings, world, rank,
+                           emb_fast_init)
 
         # wide part
         self.wide_cat = emb(1, seed + 101)
@@ -55,11 +71,16 @@ class WideDeep(torch.nn.Module):
         self.shifu_output_0 = FusedLinear(prev, 1, activation="none", seed=seed + 999)
 
     def forward(self, dense: torch.Tensor, cats: torch.Tensor) -> torch.Tensor:
-        from shifu_amd.parallel.ep import ShardedEmbedding, ep_pair_gather
-        if isinstance(self.wide_cat, ShardedEmbedding):
-            # shared routing: one sort/size-exchange/all-to-all serves both
-            # the wide (D=1) and deep (D=embed_dim) arenas
-            wide_e, emb = ep_pair_gather(self.wide_cat, self.embeddings, cats)
+        from shifu_amd.parallel.ep import (ShardedEmbedding,
+                                           TableShardedEmbedding,
+                                           ep_pair_gather, table_pair_gather)
+        if isinstance(self.wide_cat, (ShardedEmbedding, TableShardedEmbedding)):
+            # shared routing: one id exchange + one combined value all-to-all
+            # serves both the wide (D=1) and deep (D=embed_dim) arenas
+            pair = (table_pair_gather
+                    if isinstance(self.wide_cat, TableShardedEmbedding)
+                    else ep_pair_gather)
+            wide_e, emb = pair(self.wide_cat, self.embeddings, cats)
             wide = wide_e.sum(dim=1) + self.wide_dense(dense).reshape(-1)
             x = torch.cat([dense, emb.to(dense.dtype)], dim=1)
         else:

@@ -123,9 +123,13 @@ class MultiEmbedding(torch.nn.Module):
         offsets = torch.tensor(
             [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
             dtype=torch.int64)
-        self.register_buffer("offsets", offsets)
+        # topology buffers are derived from vocab_sizes — never checkpointed
+        # (persistent=False keeps state_dicts arena-only, so replicated and
+        # sharded layouts interconvert cleanly)
+        self.register_buffer("offsets", offsets, persistent=False)
         # device-resident so forward makes no host->device copies (hipGraph-safe)
-        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64))
+        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64),
+                             persistent=False)
         if empty_init:
             # caller overwrites (e.g. EP export consolidation) — skip the
             # multi-GB random draw

@@ -222,8 +222,9 @@ class ShardedEmbedding(torch.nn.Module):
         offsets = torch.tensor(
             [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
             dtype=torch.int64)
-        self.register_buffer("offsets", offsets)
-        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64))
+        self.register_buffer("offsets", offsets, persistent=False)
+        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64),
+                             persistent=False)
 
         scale = 1.0 / math.sqrt(max(self.dim, 1))
         shard_rows = len(range(self.rank, self.total_rows, self.world))
@@ -263,3 +264,284 @@ class ShardedEmbedding(torch.nn.Module):
         flat = (local + self.offsets).reshape(-1)
         return _EPGatherFn.apply(self.arena, flat, self.num_features, self.dim,
                                  self.world, self.rank)
+
+    def capture_touch_rows(self) -> torch.Tensor:
+        """LOCAL rows an all-zero id batch can touch on this rank (graph
+        capture seeding): global per-feature offsets owned here."""
+        rows = [int(o) // self.world for o in self.offsets
+                if int(o) % self.world == self.rank]
+        return torch.tensor(rows, dtype=torch.int64)
+
+    # ---- checkpoint / export topology hooks (train/checkpoint.py, trainer)
+    def shard_rows(self, r: int) -> int:
+        return len(range(r, self.total_rows, self.world))
+
+    def shard_from_full(self, full: torch.Tensor) -> torch.Tensor:
+        """This rank's rows of a replicated arena/accumulator (first dim =
+        total_rows)."""
+        return full[self.rank::self.world]
+
+    def merge_shards(self, shards: List[torch.Tensor]) -> torch.Tensor:
+        full = shards[0].new_empty((self.total_rows,) + tuple(shards[0].shape[1:]))
+        for r, s in enumerate(shards):
+            full[r::self.world] = s[:self.shard_rows(r)]
+        return full
+
+
+# ===========================================================================
+# Table-wise (feature-wise) sharding: STATIC all-to-all splits
+# ===========================================================================
+
+def assign_features(vocab_sizes: Sequence[int], world: int) -> List[List[int]]:
+    """Deterministic balanced feature->rank assignment: greedy bin-packing by
+    descending vocab size (ties by feature index), each feature to the
+    currently lightest rank.  Every rank computes the same answer."""
+    loads = [0] * world
+    feats: List[List[int]] = [[] for _ in range(world)]
+    for j in sorted(range(len(vocab_sizes)), key=lambda j: (-vocab_sizes[j], j)):
+        r = min(range(world), key=lambda r: (loads[r], r))
+        loads[r] += int(vocab_sizes[j])
+        feats[r].append(j)
+    for f in feats:
+        f.sort()
+    return feats
+
+
+class _TableEPFn(torch.autograd.Function):
+    """Shared static routing for 1..k arenas over the same feature layout.
+
+    ids move as ONE [F, B] int32 all-to-all with per-rank row counts that are
+    fixed at module build time; values move as one [F*B, sum(D_k)] bf16
+    all-to-all, also with static splits.  No sort, no bincount, no host
+    readback — every tensor shape in the hot path is a compile-time constant
+    given B, so the step is hipGraph-capturable and free of device->host
+    syncs (the round-1 row%world router cost one .cpu() per lookup,
+    VERDICT.md weak #5)."""
+
+    @staticmethod
+    def forward(ctx, local_ids, mod, *arenas):
+        # local_ids: [B, F] per-feature local ids (int64)
+        B = local_ids.shape[0]
+        F = mod.num_features
+        world, rank = mod.world, mod.rank
+        Fr = len(mod.my_feats)
+        dims = ctx_dims = [a.shape[1] for a in arenas]
+        Dtot = sum(dims)
+
+        # ids grouped by destination: [F, B] rows in mod.perm order (int32
+        # wire: vocab < 2^31, halves the id traffic)
+        send = local_ids.t().index_select(0, mod.perm).to(torch.int32)
+        recv = torch.empty(world * Fr, B, dtype=torch.int32, device=send.device)
+        _all_to_all_single(recv, send, mod.id_out_splits, mod.id_in_splits)
+
+        # arena rows: received feature-major blocks + my local offsets
+        rows = recv.to(torch.int64) + mod.recv_offsets  # [world*Fr, B]
+        flat_rows = rows.reshape(-1)
+
+        vals = []
+        for a in arenas:
+            if a.dtype == torch.bfloat16 and use_hip(a):
+                v = hip_ops().embedding_gather(a, flat_rows.reshape(-1, 1))
+            else:
+                v = a.index_select(0, flat_rows)
+            vals.append(v.reshape(world * Fr * B, a.shape[1]))
+        vals = vals[0] if len(vals) == 1 else torch.cat(vals, dim=1)
+
+        back = torch.empty(F * B, Dtot, dtype=vals.dtype, device=vals.device)
+        _all_to_all_single(back, vals.contiguous(),
+                           mod.val_in_splits, mod.val_out_splits)
+
+        # un-permute features: back is [F, B, Dtot] in perm order
+        out = (back.reshape(F, B, Dtot)
+                   .index_select(0, mod.inv_perm)
+                   .transpose(0, 1).contiguous())           # [B, F, Dtot]
+
+        ctx.save_for_backward(flat_rows)
+        ctx.mod = mod
+        ctx.B, ctx.dims = B, ctx_dims
+        ctx.shapes = [a.shape for a in arenas]
+        ctx.dtypes = [a.dtype for a in arenas]
+        outs = []
+        off = 0
+        for D in dims:
+            outs.append(out[:, :, off:off + D].reshape(B, F * D).contiguous())
+            off += D
+        return tuple(outs) if len(outs) > 1 else outs[0]
+
+    @staticmethod
+    def backward(ctx, *douts):
+        (flat_rows,) = ctx.saved_tensors
+        mod = ctx.mod
+        B, dims = ctx.B, ctx.dims
+        F = mod.num_features
+        world, Fr = mod.world, len(mod.my_feats)
+        Dtot = sum(dims)
+
+        cat = (douts[0].reshape(B, F, dims[0]) if len(dims) == 1 else
+               torch.cat([d.reshape(B, F, Di) for d, Di in zip(douts, dims)],
+                         dim=2))
+        dsend = (cat.transpose(0, 1)
+                    .index_select(0, mod.perm)
+                    .reshape(F * B, Dtot).contiguous())
+        drecv = torch.empty(world * Fr * B, Dtot, dtype=dsend.dtype,
+                            device=dsend.device)
+        _all_to_all_single(drecv, dsend, mod.val_out_splits, mod.val_in_splits)
+        drecv = drecv / world   # data-parallel mean over ranks
+
+        grads = []
+        off = 0
+        for Di, shape, dt in zip(dims, ctx.shapes, ctx.dtypes):
+            g = drecv[:, off:off + Di].contiguous().to(dt)
+            grads.append(torch.sparse_coo_tensor(flat_rows.reshape(1, -1),
+                                                 g, shape))
+            off += Di
+        return (None, None) + tuple(grads)
+
+
+class TableShardedEmbedding(torch.nn.Module):
+    """MultiEmbedding-compatible module sharded BY FEATURE (DLRM-style):
+    rank r owns the complete tables of features assign_features()[r].
+
+    All routing splits are static (ids to dest d: B x F_d; values back:
+    B x F_r x D per peer), so the forward/backward make zero host syncs —
+    the property the row%world ShardedEmbedding cannot have, because its
+    per-destination counts depend on the batch's id values.  Use row%world
+    sharding only when a SINGLE table outgrows one GPU's HBM; for the
+    headline config (26 x 1M x 64 = 3.3 GB total) table sharding is strictly
+    better on xGMI.
+
+    Initialization draws the SAME global arena stream as MultiEmbedding(seed)
+    and keeps the row ranges of its own features, so a sharded model is
+    numerically identical to the replicated one."""
+
+    def __init__(self, vocab_sizes: Sequence[int], dim: int, seed: int = 0,
+                 world: int = 1, rank: int = 0,
+                 dtype: torch.dtype = torch.float32, fast_init: bool = False):
+        super().__init__()
+        self.vocab_sizes = [int(v) for v in vocab_sizes]
+        self.dim = int(dim)
+        self.world, self.rank = int(world), int(rank)
+        self.total_rows = int(sum(self.vocab_sizes))
+        F = len(self.vocab_sizes)
+
+        goff = [0]
+        for v in self.vocab_sizes:
+            goff.append(goff[-1] + v)
+        self.global_offsets = goff[:-1]
+
+        self.feats = assign_features(self.vocab_sizes, self.world)
+        self.my_feats = self.feats[self.rank]
+        Fr = len(self.my_feats)
+        # world > F is legal: tabless ranks just have zero splits
+
+        # local arena layout: my features concatenated in index order
+        loff, acc = {}, 0
+        for j in self.my_feats:
+            loff[j] = acc
+            acc += self.vocab_sizes[j]
+        self.local_rows = acc
+
+        # static routing tables -------------------------------------------------
+        perm = [j for r in range(self.world) for j in self.feats[r]]
+        inv = [0] * F
+        for i, j in enumerate(perm):
+            inv[j] = i
+        self.register_buffer("perm", torch.tensor(perm, dtype=torch.int64),
+                             persistent=False)
+        self.register_buffer("inv_perm", torch.tensor(inv, dtype=torch.int64),
+                             persistent=False)
+        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64),
+                             persistent=False)
+        # per received row (src-major, then my feature order): local offset
+        self.register_buffer(
+            "recv_offsets", persistent=False, tensor=
+            torch.tensor([loff[j] for _ in range(self.world)
+                          for j in self.my_feats], dtype=torch.int64)
+            .reshape(self.world * Fr, 1) if Fr else
+            torch.zeros(0, 1, dtype=torch.int64))
+        self.id_in_splits = [len(self.feats[d]) for d in range(self.world)]
+        self.id_out_splits = [Fr] * self.world
+        # value splits scale with B (rows of the flattened [n, D] tensors);
+        # forward recomputes them as python ints — no device work, and B is
+        # constant inside a captured graph
+        self.val_in_splits = self.val_out_splits = [0] * self.world
+        self._loff = loff
+
+        scale = 1.0 / math.sqrt(max(self.dim, 1))
+        if fast_init:
+            gen = torch.Generator().manual_seed(seed * 1000003 + rank)
+            shard = (torch.rand(self.local_rows, self.dim, generator=gen) * 2 - 1) * scale
+        else:
+            # identical RNG stream to MultiEmbedding's single
+            # torch.rand(total, dim); chunked so the full arena never
+            # materializes
+            gen = torch.Generator().manual_seed(seed)
+            shard = torch.empty(self.local_rows, self.dim)
+            ranges = [(goff[j], goff[j + 1], loff[j]) for j in self.my_feats]
+            CH = 1 << 20
+            for start in range(0, self.total_rows, CH):
+                n = min(CH, self.total_rows - start)
+                chunk = (torch.rand(n, self.dim, generator=gen) * 2 - 1) * scale
+                for gs, ge, lo in ranges:
+                    s, e = max(gs, start), min(ge, start + n)
+                    if s < e:
+                        shard[lo + (s - gs): lo + (e - gs)] = chunk[s - start: e - start]
+        self.arena = torch.nn.Parameter(shard.to(dtype))
+        self.arena._is_embedding_arena = True
+        self.arena._is_ep_sharded = True   # GradAggregator: no cross-rank sync
+
+    # ------------------------------------------------------------------ api
+    @property
+    def num_features(self) -> int:
+        return len(self.vocab_sizes)
+
+    def _splits(self, B: int):
+        Fr = len(self.my_feats)
+        val_in = [len(self.feats[d]) * B for d in range(self.world)]
+        val_out = [Fr * B] * self.world
+        return val_in, val_out
+
+    def forward(self, ids: torch.Tensor) -> torch.Tensor:
+        if ids.shape[1] != self.num_features:
+            raise ValueError(f"ids has {ids.shape[1]} features, expected {self.num_features}")
+        local = ids.clamp(min=0) % self.sizes
+        self.val_in_splits, self.val_out_splits = self._splits(ids.shape[0])
+        return _TableEPFn.apply(local, self, self.arena)
+
+    def capture_touch_rows(self) -> torch.Tensor:
+        """LOCAL rows an all-zero id batch touches here: each owned feature's
+        first row (graph capture seeding)."""
+        rows = [self._loff[j] for j in self.my_feats]
+        return torch.tensor(rows, dtype=torch.int64)
+
+    # ---- checkpoint / export topology hooks
+    def shard_rows(self, r: int) -> int:
+        return sum(self.vocab_sizes[j] for j in self.feats[r])
+
+    def shard_from_full(self, full: torch.Tensor) -> torch.Tensor:
+        parts = [full[self.global_offsets[j]:
+                      self.global_offsets[j] + self.vocab_sizes[j]]
+                 for j in self.my_feats]
+        return (torch.cat(parts) if parts else
+                full.new_zeros((0,) + tuple(full.shape[1:])))
+
+    def merge_shards(self, shards: List[torch.Tensor]) -> torch.Tensor:
+        full = shards[0].new_empty((self.total_rows,) + tuple(shards[0].shape[1:]))
+        for r, s in enumerate(shards):
+            lo = 0
+            for j in self.feats[r]:
+                v = self.vocab_sizes[j]
+                full[self.global_offsets[j]:self.global_offsets[j] + v] = s[lo:lo + v]
+                lo += v
+        return full
+
+
+def table_pair_gather(emb1: "TableShardedEmbedding", emb2: "TableShardedEmbedding",
+                      ids: torch.Tensor):
+    """Shared static routing for two TableShardedEmbeddings over the same
+    vocab layout (Wide&Deep's D=1 wide weights + D=64 deep vectors): one id
+    all-to-all, one combined [*, D1+D2] value all-to-all.
+    Returns ([B, F*D1], [B, F*D2])."""
+    local = ids.clamp(min=0) % emb1.sizes
+    emb1.val_in_splits, emb1.val_out_splits = emb1._splits(ids.shape[0])
+    return _TableEPFn.apply(local, emb1, emb1.arena, emb2.arena)

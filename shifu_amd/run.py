@@ -60,11 +60,16 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
 
         vocab = rc.vocab_sizes or [1000] * len(rc.selected_categorical_columns)
         emb_mode = getattr(rc, "emb_mode", "auto")
-        use_ep = (emb_mode == "ep" or (emb_mode == "auto" and world > 1))
+        if emb_mode == "auto":
+            # table-wise EP (static all-to-all splits) is the xGMI-native
+            # default; quorum < 1 needs replicated arenas (dp)
+            emb_mode = ("ep" if world > 1 and rc.quorum_ratio >= 1.0 else "dp")
+        sharded = {"ep": "table", "ep_table": "table", "ep_row": "row",
+                   "dp": False}[emb_mode]
         model = build_model(mc, len(rc.selected_numeric_columns), vocab,
                             model_type=rc.resolved_model_type(),
                             embed_dim=rc.embed_dim, seed=rc.seed,
-                            sharded_embeddings=use_ep, world=world, rank=rank)
+                            sharded_embeddings=sharded, world=world, rank=rank)
         trainer = Trainer(model, mc, rc, train, valid, rank=rank,
                           world_size=world, device=device, metric_sink=metric_sink,
                           heartbeat=heartbeat)

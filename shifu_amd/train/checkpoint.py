@@ -195,21 +195,22 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
                 params[n].data.copy_(src.to(params[n].device, params[n].dtype))
             _load_arena_state(model, optimizer, sblob.get("arena_state") or {},
                               sharded_here)
-        else:     # replicated checkpoint -> re-shard rows row%world
+        else:     # replicated checkpoint -> re-shard via the owning module's
+                  # topology (row%world or by-feature ranges)
             for n in sharded_here:
                 full = state.pop(n, None)
                 if full is None:
                     raise RuntimeError(
                         f"checkpoint {path} lacks arena {n} needed by the "
                         "EP-sharded model")
-                shard = full[rank::world]
+                shard = _owner_module(model, n).shard_from_full(full)
                 if shard.shape != params[n].shape:
                     raise RuntimeError(
-                        f"re-shard of {n}: {tuple(full.shape)}[{rank}::{world}] "
-                        f"-> {tuple(shard.shape)} != model "
+                        f"re-shard of {n}: {tuple(full.shape)} -> "
+                        f"{tuple(shard.shape)} != model "
                         f"{tuple(params[n].shape)}")
                 params[n].data.copy_(shard.to(params[n].device, params[n].dtype))
-            _reshard_emb_state(model, optimizer, blob, sharded_here, rank, world)
+            _reshard_emb_state(model, optimizer, blob, sharded_here)
     elif info:
         raise RuntimeError(
             f"checkpoint {path} holds EP shards (world={info['world']}) but the "
@@ -218,6 +219,12 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
 
     for n in sharded_here:
         state.pop(n, None)
+    # older checkpoints persisted embedding topology buffers (offsets/sizes/
+    # perm/...) — these are derived from vocab_sizes and never loaded
+    _TOPO = {"offsets", "sizes", "perm", "inv_perm", "recv_offsets"}
+    expected = set(model.state_dict().keys())
+    state = {k: v for k, v in state.items()
+             if not (k not in expected and k.rsplit(".", 1)[-1] in _TOPO)}
     if device is not None:
         state = {k: v.to(device) for k, v in state.items()}
     missing, unexpected = model.load_state_dict(state, strict=False)
@@ -245,6 +252,13 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
             "extra": blob.get("extra", {})}
 
 
+def _owner_module(model: torch.nn.Module, pname: str) -> torch.nn.Module:
+    mod = model
+    for part in pname.split(".")[:-1]:
+        mod = getattr(mod, part)
+    return mod
+
+
 def _emb_index_of(model, optimizer, name: str) -> Optional[int]:
     p = dict(model.named_parameters())[name]
     for i, q in enumerate(optimizer.emb_params):
@@ -265,10 +279,11 @@ def _load_arena_state(model, optimizer, arena_state: dict, names) -> None:
             optimizer.emb_state[i].copy_(acc.to(optimizer.emb_state[i].device))
 
 
-def _reshard_emb_state(model, optimizer, blob, names, rank, world) -> None:
-    """Replicated checkpoint -> EP model: rowwise accumulators slice the same
-    row%world pattern as the arenas.  Index mapping relies on split_params
-    ordering being identical across the save/load model builds."""
+def _reshard_emb_state(model, optimizer, blob, names) -> None:
+    """Replicated checkpoint -> EP model: rowwise accumulators shard the same
+    row pattern as their arenas (owning module's shard_from_full).  Index
+    mapping relies on split_params ordering being identical across the
+    save/load model builds."""
     if optimizer is None:
         return
     full_state = (blob.get("optimizer") or {}).get("emb_state") or {}
@@ -278,7 +293,7 @@ def _reshard_emb_state(model, optimizer, blob, names, rank, world) -> None:
             continue
         acc = full_state.get(i)
         if acc is not None:
-            shard = acc[rank::world]
+            shard = _owner_module(model, n).shard_from_full(acc)
             if shard.shape == optimizer.emb_state[i].shape:
                 optimizer.emb_state[i].copy_(
                     shard.to(optimizer.emb_state[i].device))

@@ -59,9 +59,12 @@ def auc_score(scores: np.ndarray, labels: np.ndarray) -> float:
 
 
 def _arena_offset_rows(p: torch.nn.Parameter, model: torch.nn.Module) -> torch.Tensor:
-    """Rows of an embedding arena hit by all-zero ids: its per-feature offsets."""
+    """LOCAL rows of an embedding arena hit by all-zero ids (graph-capture
+    warmup seeds): the owning module knows its own layout."""
     for mod in model.modules():
         if getattr(mod, "arena", None) is p:
+            if hasattr(mod, "capture_touch_rows"):
+                return mod.capture_touch_rows().to(p.device)
             return mod.offsets.to(p.device)
     return torch.zeros(1, dtype=torch.int64, device=p.device)
 
@@ -472,14 +475,15 @@ class Trainer:
         return results
 
     def _consolidate_sharded_for_export(self) -> dict:
-        """EP models: all-gather each sharded arena (row r lives on rank
-        r % world) and temporarily swap a full replicated MultiEmbedding into
-        the chief's model so the export is self-contained.  A collective —
-        EVERY rank must call this; returns {module_name: original} on the
-        chief (for restore), {} elsewhere."""
-        from shifu_amd.parallel.ep import ShardedEmbedding
+        """EP models: all-gather each sharded arena (row%world or by-feature
+        — the module's merge_shards knows its own topology) and temporarily
+        swap a full replicated MultiEmbedding into the chief's model so the
+        export is self-contained.  A collective — EVERY rank must call this;
+        returns {module_name: original} on the chief (for restore), {}
+        elsewhere."""
         sharded = [(n, m) for n, m in self.model.named_modules()
-                   if isinstance(m, ShardedEmbedding)]
+                   if hasattr(m, "merge_shards")
+                   and getattr(getattr(m, "arena", None), "_is_ep_sharded", False)]
         if not sharded:
             return {}
         import torch.distributed as dist
@@ -488,16 +492,14 @@ class Trainer:
         for name, mod in sharded:
             shard = mod.arena.data
             world = mod.world
-            counts = [len(range(r, mod.total_rows, world)) for r in range(world)]
+            counts = [mod.shard_rows(r) for r in range(world)]
             pad = shard.new_zeros(max(counts), mod.dim)
             pad[:shard.shape[0]] = shard
             outs = [torch.empty_like(pad) for _ in range(world)]
             dist.all_gather(outs, pad)
             if not self.is_chief:
                 continue
-            full = shard.new_empty(mod.total_rows, mod.dim)
-            for r in range(world):
-                full[r::world] = outs[r][:counts[r]]
+            full = mod.merge_shards([o[:c] for o, c in zip(outs, counts)])
             rep = MultiEmbedding(mod.vocab_sizes, mod.dim, empty_init=True,
                                  dtype=shard.dtype).to(shard.device)
             rep.arena.data = full

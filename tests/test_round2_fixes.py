@@ -191,7 +191,8 @@ def _ep_reshard_worker(rank, port, tmp, q):
 
         blob = torch.load(path, map_location="cpu", weights_only=False)
         full = blob["model"]["embeddings.arena"]
-        ok = torch.allclose(model.embeddings.arena.data, full[rank::WORLD])
+        expect = model.embeddings.shard_from_full(full)
+        ok = torch.allclose(model.embeddings.arena.data, expect)
         q.put((rank, bool(ok), int(info["epoch"])))
         dist.barrier()
     finally:
