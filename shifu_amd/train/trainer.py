@@ -204,6 +204,22 @@ class Trainer:
         self.global_step = 0
         self.start_epoch = 0
         self._rng = np.random.default_rng(rc.seed + rank)
+        # Uniform stepping at world>1: every rank takes the SAME number of
+        # steps per epoch with FULL-SIZE batches, padding tails with
+        # weight-0 rows.  Shards differ by up to one row, so without this
+        # (a) ranks could take different step counts (mismatched collectives
+        # = deadlock) and (b) tail batch sizes could differ across ranks,
+        # which the static-split table-EP all-to-all cannot tolerate.
+        # Weight-0 rows are exact no-ops: the loss normalizes by the
+        # nonzero-weight count and their gradients are identically zero.
+        self._uniform_steps = None
+        if world_size > 1 and is_distributed():
+            import math as _math
+            t = torch.tensor([_math.ceil(len(self.train_data) / self.batch_size)],
+                             device=self.device if self.device.type == "cuda"
+                             else "cpu")
+            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+            self._uniform_steps = int(t)
         from shifu_amd.utils.trace import StepTracer
         self.tracer = StepTracer(enabled=rc.enable_trace,
                                  use_gpu_events=(self.device.type == "cuda"))
@@ -312,12 +328,14 @@ class Trainer:
             self.use_graphs = False
             return False
 
-    def _graphed_step(self, idx: torch.Tensor) -> torch.Tensor:
+    def _graphed_step(self, idx: torch.Tensor, k_valid: int) -> torch.Tensor:
         gb = self._gb
         torch.index_select(self.train_data.dense, 0, idx, out=gb.dense)
         torch.index_select(self.train_data.cats, 0, idx, out=gb.cats)
         torch.index_select(self.train_data.target, 0, idx, out=gb.target)
         torch.index_select(self.train_data.weight, 0, idx, out=gb.weight)
+        if k_valid < idx.shape[0]:          # outside the captured graph
+            gb.weight[k_valid:] = 0.0
         loss = self._graph.run()
         self.global_step += 1
         return loss.detach().clone()
@@ -325,31 +343,65 @@ class Trainer:
     @torch.no_grad()
     def evaluate(self, data: DeviceData, batch_size: int = 65536) -> dict:
         """Weighted loss + AUC over a dataset (the reference's per-epoch valid
-        pass, ssgd_monitor.py:281-284)."""
+        pass, ssgd_monitor.py:281-284).
+
+        At world>1 the forward may contain EP collectives, so every rank runs
+        the SAME number of batches with the SAME batch size — ranks with
+        fewer local rows pad with weight-0 repeats of row 0 (no-ops in the
+        metrics)."""
         self.model.eval()
         n = len(data)
-        if n == 0:
+        steps = (n + batch_size - 1) // batch_size
+        eb = batch_size
+        if self._uniform_steps is not None:
+            t = torch.tensor([n], device=self.device
+                             if self.device.type == "cuda" else "cpu")
+            torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+            gmax = int(t)
+            eb = max(min(batch_size, gmax), 1)
+            steps = max((gmax + eb - 1) // eb, 1)
+        if steps == 0 or (n == 0 and self._uniform_steps is None):
             self.model.train()
             return {"loss": 0.0, "auc": 0.5, "n": 0}
+        idx_dev = ("cpu" if getattr(data, "index_device", "device") == "cpu"
+                   else self.device)
         total, wtot = 0.0, 0.0
         scores = []
-        for s in range(0, n, batch_size):
-            b = data.slice(slice(s, min(s + batch_size, n)))
+        for si in range(steps):
+            s = si * eb
+            k = max(min(s + eb, n) - s, 0)
+            if n == 0:
+                # empty local shard but peers still run collectives: join
+                # with an all-padding batch
+                b = DeviceData(
+                    dense=torch.zeros(eb, self.train_data.dense.shape[1],
+                                      device=self.device, dtype=self.dense_dtype),
+                    cats=torch.zeros(eb, self.train_data.cats.shape[1],
+                                     device=self.device, dtype=torch.int64),
+                    target=torch.zeros(eb, device=self.device),
+                    weight=torch.zeros(eb, device=self.device))
+            else:
+                idx = torch.arange(s, s + k, device=idx_dev)
+                if self._uniform_steps is not None and k < eb:
+                    idx = torch.cat([idx, idx.new_zeros(eb - k)])
+                b = data.slice(idx)
             logits = self.model(b.dense, b.cats).float().reshape(-1)
-            p = torch.sigmoid(logits)
-            w = b.weight.float()
+            p = torch.sigmoid(logits)[:k]
+            w = b.weight.float()[:k]
+            y = b.target.float()[:k]
             if self.loss_kind in ("sigmoid_ce", "ce", "bce"):
                 per = w * torch.nn.functional.binary_cross_entropy_with_logits(
-                    logits, b.target.float(), reduction="none")
+                    logits[:k], y, reduction="none")
             else:
-                per = w * (p - b.target.float()) ** 2
+                per = w * (p - y) ** 2
             total += float(per.sum())
             wtot += float((w != 0).sum())  # SUM_BY_NONZERO_WEIGHTS (ops/loss.py)
             scores.append(p.cpu().numpy())
         self.model.train()
         scores = np.concatenate(scores)
-        labels = data.target.cpu().numpy()
-        return {"loss": total / max(wtot, 1e-12), "auc": auc_score(scores, labels), "n": n}
+        labels = data.target.cpu().numpy()[:len(scores)]
+        return {"loss": total / max(wtot, 1e-12),
+                "auc": auc_score(scores, labels), "n": n}
 
     # ------------------------------------------------------------------ epochs
     def run_epoch(self, epoch: int) -> TrainingIntermediateResult:
@@ -365,16 +417,23 @@ class Trainer:
         if getattr(self.train_data, "index_device", "device") != "cpu":
             perm = perm.to(self.device)
         losses = []
-        steps = range(0, n, self.batch_size)
-        n_steps = len(steps)
-        for si, s in enumerate(steps):
-            idx = perm[s:min(s + self.batch_size, n)]
+        B = self.batch_size
+        n_steps = self._uniform_steps or max((n + B - 1) // B, 1)
+        for si in range(n_steps):
+            s = si * B
+            idx = perm[s:min(s + B, n)]
+            k = idx.shape[0]             # valid rows; the rest are padding
+            if self._uniform_steps and k < B:
+                idx = torch.cat([idx, idx.new_zeros(B - k)])
             sync = ((si + 1) % self.update_window == 0) or (si == n_steps - 1)
             if (self.use_graphs and sync and idx.shape[0] == self.batch_size
                     and self._ensure_graph()):
-                losses.append(self._graphed_step(idx))
+                losses.append(self._graphed_step(idx, k))
             else:
                 batch = self.train_data.slice(idx)
+                if k < idx.shape[0]:
+                    batch.weight = batch.weight.clone()
+                    batch.weight[k:] = 0.0   # padding rows are exact no-ops
                 # window mode: only every update_window-th (or last) step
                 # syncs+updates
                 losses.append(self.train_step(batch, sync=sync))
