@@ -509,6 +509,28 @@ void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 #define V4_A(par) ((par) * 65536)
 #define V4_B(par) ((par) * 65536 + 32768)
 
+// T1: XCD-aware blockIdx remap (bijective for ANY block count, guide §5.5):
+// the dispatcher places linear block b on XCD b%8, so remapping gives each
+// XCD a CONTIGUOUS chunk of the row-major (by,bx) tile space — neighboring
+// tiles that share A-row/B-col panels then hit the same per-XCD L2.
+DEVINL void xcd_swizzle_xy(int& bx, int& by, int on) {
+  if (!on) return;
+  int gx = gridDim.x, gy = gridDim.y;
+  int nwg = gx * gy;
+  // Narrow grids only (the bench tower shapes: N/256 <= 8 column tiles):
+  // there a per-XCD chunk of row-major tile space = many M-rows x all
+  // N-cols, so the whole B panel set stays L2-resident per XCD (+3-8%
+  // measured on fwd/dgrad shapes).  On wide square grids the same strips
+  // span every B panel and LOSE to the dispatcher's round-robin (-6% @8k).
+  if (gx > 8 || nwg < 16) return;
+  int lin = by * gx + bx;
+  int q = nwg >> 3, r = nwg & 7;
+  int xcd = lin & 7, idx = lin >> 3;
+  int lin2 = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  bx = lin2 % gx;
+  by = lin2 / gx;
+}
+
 DEVINL void v4_stage_half_glds(const bf16* __restrict__ P, char* lds_region,
                                int R0, int half, int k0, int K, int tid) {
   // one 16KB half (rows half*128..+127 of the 256-row region), 2 glds/thread
@@ -564,19 +586,25 @@ DEVINL void v4_stage_slot(const bf16* A, const bf16* B, char* smem, int slot,
   }
 }
 
-template <int EPI, typename OUT_T, bool SPLITK = false>
+// VAR 0: phase barrier after each of q0/q1 plus a separate boundary barrier
+// VAR 1: q1's phase barrier merges into the boundary one (2 barriers/tile);
+//        safe: every region staged right after the merged barrier had its
+//        last reads completed by the lgkmcnt(0) preceding it
+template <int EPI, typename OUT_T, bool SPLITK = false, int VAR = 0>
 __global__ __launch_bounds__(512, 1)
 void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                        OUT_T* __restrict__ C, const bf16* __restrict__ bias,
-                       int M, int N, int K, int act, long slab) {
+                       int M, int N, int K, int act, long slab, int swz) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int wm = wave >> 2;            // 0..1 -> M offset wm*128
   const int wn = wave & 3;             // 0..3 -> N offset wn*64
-  const int m0 = blockIdx.y * V4_BM;
-  const int n0 = blockIdx.x * V4_BN;
+  int bx = blockIdx.x, by = blockIdx.y;
+  xcd_swizzle_xy(bx, by, swz);         // T1: per-XCD L2 affinity
+  const int m0 = by * V4_BM;
+  const int n0 = bx * V4_BN;
   const int r16 = lane & 15;
   const int kgrp = lane >> 4;
   const int rx = r16 & 7;
@@ -668,7 +696,7 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
       // overwrites a region a lagging wave still reads.  lgkmcnt(0) flushes
       // this wave's ds_writes (edge-tile guarded staging) before the barrier.
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_barrier();
+      if (!(VAR == 1 && q == 1)) __builtin_amdgcn_s_barrier();
     }
     // K-tile boundary.  The only loads allowed to stay in flight are the
     // B(t+2) slots (4 glds) — and only when they were actually staged by
@@ -720,15 +748,288 @@ void gemm_nt_v4_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
 
 #define V4_LDS_BYTES 131072
 
+// ---------------------------------------------------------------------------
+// v5 "NT" GEMM — v4's 256x256 glds ring restructured to the guide's 8-phase
+// sub-tile interleave (cdna_hip_programming.md §5 "256² 8-phase template",
+// T3+T4+T5): four phases per K-tile, each phase
+//   { ds_read one accumulator-row-pair's A fragments (q=0 also reads the
+//     whole tile's B fragments) ∥ glds-stage ONE 16KB half-tile of a future
+//     tile → raw s_barrier → s_waitcnt lgkmcnt(0) → s_setprio(1) →
+//     16 MFMA (2 acc rows x 4 nf x 2 k-steps) → s_setprio(0) → raw barrier }
+// vs v4's two 32-MFMA phases staging two half-tiles each.  The counted
+// boundary wait (vmcnt(4), never 0 in the loop) is unchanged; the finer
+// interleave is what lifts MfmaUtil (guide regime table: 8ph is the
+// prerequisite for T2/T5 to pay).  Staging schedule and region-safety
+// argument are v4's: A(t+1) halves in phases 0-1 (region's last reader was
+// tile t-1), B(t+2) halves in phases 2-3 (B(t) fragments live in registers
+// after phase 0's lgkmcnt(0)).
+// The bf16 epilogue stages C tiles through LDS (free after the main loop:
+// each wave owns 16KB = its 128x64 bf16 tile) so global stores are 16-byte
+// dwordx4 instead of v4's 2-byte scalars.
+// ---------------------------------------------------------------------------
+// VAR 0: two barriers per phase (template-literal form)
+// VAR 1: ONE barrier per phase — region safety still holds because every
+//        stage target's last reader finished >= 2 barriers before the stage
+//        issues (reads complete at the lgkmcnt(0) directly after their
+//        phase's barrier, i.e. before the NEXT barrier)
+template <int EPI, typename OUT_T, bool SPLITK = false, int VAR = 0>
+__global__ __launch_bounds__(512, 1)
+void gemm_nt_v5_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                       OUT_T* __restrict__ C, const bf16* __restrict__ bias,
+                       int M, int N, int K, int act, long slab, int swz) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;            // 0..1 -> M offset wm*128
+  const int wn = wave & 3;             // 0..3 -> N offset wn*64
+  int bx = blockIdx.x, by = blockIdx.y;
+  xcd_swizzle_xy(bx, by, swz);         // T1: per-XCD L2 affinity
+  const int m0 = by * V4_BM;
+  const int n0 = bx * V4_BN;
+  const int r16 = lane & 15;
+  const int kgrp = lane >> 4;
+  const int rx = r16 & 7;
+
+  int kt_lo = 0, kt_hi = (K + V4_BK - 1) / V4_BK;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int per = (kt_hi + nz - 1) / nz;
+    kt_lo = blockIdx.z * per;
+    kt_hi = min(kt_hi, kt_lo + per);
+    if (kt_lo >= kt_hi) return;
+  }
+  const int n_kt = kt_hi - kt_lo;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // prologue (v4's): tile t0 fully + B(t0+1); A(t0+1) arrives in phases 0-1.
+  v4_stage_slot(A, B, smem, 0, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 1, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 2, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 3, kt_lo, m0, n0, M, N, K, tid);
+  if (n_kt > 1) {
+    v4_stage_slot(A, B, smem, 2, kt_lo + 1, m0, n0, M, N, K, tid);
+    v4_stage_slot(A, B, smem, 3, kt_lo + 1, m0, n0, M, N, K, tid);
+  }
+  {
+    bool b1_glds = (n_kt > 1) && ((kt_lo + 1) * V4_BK + V4_BK <= K) &&
+                   (n0 + V4_BN <= N);
+    if (b1_glds) asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+    else asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < n_kt; ++t) {
+    const int kt = kt_lo + t;
+    const int par = kt & 1;
+    const char* Ar = smem + V4_A(par);
+    const char* Br = smem + V4_B(par);
+    bf16x8 bfr[4][2];
+
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      // ---- ds-load this phase's register subtile --------------------------
+      if (q == 0) {
+        // whole-tile B fragments (8 reads) — registers for all four phases
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            int row = wn * 64 + nf * 16 + r16;
+            int g = ks * 4 + kgrp;
+            bfr[nf][ks] = *(const bf16x8*)(Br + row * 128 + ((g ^ rx) * 16));
+          }
+      }
+      bf16x8 afr[2][2];  // this phase's two accumulator rows
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          int row = wm * 128 + (q * 2 + m2) * 16 + r16;
+          int g = ks * 4 + kgrp;
+          afr[m2][ks] = *(const bf16x8*)(Ar + row * 128 + ((g ^ rx) * 16));
+        }
+
+      // ---- stage ONE half-tile of a future K-tile -------------------------
+      // q0/q1: A(t+1) halves; q2/q3: B(t+2) halves (see region-safety above)
+      if (q < 2) {
+        if (t + 1 < n_kt)
+          v4_stage_slot(A, B, smem, q, kt + 1, m0, n0, M, N, K, tid);
+      } else {
+        if (t + 2 < n_kt)
+          v4_stage_slot(A, B, smem, q, kt + 2, m0, n0, M, N, K, tid);
+      }
+      if (q == 0)  // 12 ds_reads this phase: pace the LDS queue (guide §5)
+        asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf)
+            acc[q * 2 + m2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[m2][ks], bfr[nf][ks], acc[q * 2 + m2][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (q < 3) {
+        if (VAR == 0) __builtin_amdgcn_s_barrier();
+      } else {
+        // K-tile boundary (merged with phase 3's closing barrier): only the
+        // B(t+2) glds may stay in flight — and only when actually glds-staged
+        bool b2_glds = (t + 2 < n_kt) && ((kt + 2) * V4_BK + V4_BK <= K) &&
+                       (n0 + V4_BN <= N);
+        if (b2_glds) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __builtin_amdgcn_s_barrier();
+      }
+    }
+  }
+
+  if (EPI == EPI_F32) {
+    // f32 outputs (wgrad/split-K): plain 4-byte stores as v4
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        int col = n0 + wn * 64 + nf * 16 + r16;
+        if (col >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = m0 + wm * 128 + mf * 16 + kgrp * 4 + r;
+          if (row >= M) continue;
+          float v = acc[mf][nf][r];
+          if (SPLITK) {
+            if (slab) ((float*)C)[(long)blockIdx.z * slab + (long)row * N + col] = v;
+            else atomicAdd(&((float*)C)[(long)row * N + col], v);
+          } else ((float*)C)[(long)row * N + col] = v;
+        }
+      }
+    return;
+  }
+
+  // bf16 epilogue: stage through LDS so the global stores are 16B dwordx4.
+  // Wave w owns smem[w*16384 .. +16384) = its [128][64] bf16 C-tile.
+  float bvs[4] = {0.f, 0.f, 0.f, 0.f};
+  if (EPI == EPI_BIAS_ACT) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      int col = n0 + wn * 64 + nf * 16 + r16;
+      if (col < N) bvs[nf] = __bfloat162float(bias[col]);
+    }
+  }
+  // all LDS is dead after the main loop's final barrier; per-wave regions
+  // need no further synchronization
+  // 16B-chunk XOR swizzle (chunk ^= bit2(row)*4): the two kgrp halves of a
+  // wave write rows 4 apart, which land on the same banks at a 128B row
+  // stride; the swizzle separates them (write conflicts -> none)
+  short* cw = (short*)(smem + wave * 16384);
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mf][nf][r];
+        if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bvs[nf], act);
+        bf16 h = __float2bfloat16(v);
+        int row = mf * 16 + kgrp * 4 + r;
+        int col = nf * 16 + r16;
+        int sw = (((row >> 2) & 1) << 5);           // swap 16B chunks by 64B
+        cw[row * 64 + (col ^ sw)] = *(short*)&h;
+      }
+    }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  // read back rows: 16 x ds_read_b128 per lane; lane covers rows
+  // (i*64+lane)/8, 16B chunk (lane&7) -> stores land 8 rows x 128B per instr
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int elt = i * 64 + lane;        // 16B-chunk index within the wave tile
+    int row = elt >> 3;             // 8 chunks per 64-col row
+    int c8 = (elt & 7) * 8;         // first col of this chunk
+    int grow = m0 + wm * 128 + row;
+    int gcol = n0 + wn * 64 + c8;
+    int sw = (((row >> 2) & 1) << 5);
+    s16x8 v = *(const s16x8*)(cw + row * 64 + (c8 ^ sw));
+    if (grow < M) {
+      if (gcol + 8 <= N) {
+        *(s16x8*)((bf16*)C + (long)grow * N + gcol) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gcol + j < N)
+            ((short*)((bf16*)C + (long)grow * N))[gcol + j] = ((short*)&v)[j];
+      }
+    }
+  }
+}
+
+// 256^2-tile GEMM implementation selector (SHIFU_GEMM_IMPL):
+//   v4  — 2-phase x 32 MFMA, 3 barriers/tile (round-1 shipping kernel)
+//   v4b — v4 with the q1 phase barrier merged into the boundary (2/tile)
+//   v5  — 4-phase x 16 MFMA, 1 half-tile stage/phase, 2 barriers/phase,
+//         LDS-staged dwordx4 epilogue
+//   v5b — v5 with one barrier per phase
+enum GemmImpl { GI_V4 = 0, GI_V4B = 1, GI_V5 = 2, GI_V5B = 3 };
+
+static int gemm_impl() {
+  static int v = [] {
+    const char* e = getenv("SHIFU_GEMM_IMPL");
+    if (!e) return (int)GI_V4;
+    if (!strcmp(e, "v4")) return (int)GI_V4;
+    if (!strcmp(e, "v4b")) return (int)GI_V4B;
+    if (!strcmp(e, "v5")) return (int)GI_V5;
+    if (!strcmp(e, "v5b")) return (int)GI_V5B;
+    return (int)GI_V4;
+  }();
+  return v;
+}
+
+// launch whichever 256^2 kernel the selector picks (typed, so the argument
+// marshalling stays hipLaunchKernelGGL's)
+static int xcd_swz_on() {
+  static int v = [] {
+    const char* e = getenv("SHIFU_XCD_SWZ");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
+template <int EPI, typename OUT_T, bool SPLITK>
+static void launch_256(dim3 grid, const bf16* A, const bf16* B, OUT_T* C,
+                       const bf16* bias, int M, int N, int K, int act,
+                       long slab, hipStream_t s) {
+  const int swz = xcd_swz_on();
+#define SHIFU_LAUNCH_256(KERN)                                              \
+  do {                                                                      \
+    hipFuncSetAttribute((const void*)(KERN),                                \
+                        hipFuncAttributeMaxDynamicSharedMemorySize,         \
+                        V4_LDS_BYTES);                                      \
+    hipLaunchKernelGGL((KERN), grid, dim3(512), V4_LDS_BYTES, s,            \
+                       A, B, C, bias, M, N, K, act, slab, swz);             \
+  } while (0)
+  switch (gemm_impl()) {
+    case GI_V4B: SHIFU_LAUNCH_256((gemm_nt_v4_kernel<EPI, OUT_T, SPLITK, 1>)); break;
+    case GI_V5:  SHIFU_LAUNCH_256((gemm_nt_v5_kernel<EPI, OUT_T, SPLITK, 0>)); break;
+    case GI_V5B: SHIFU_LAUNCH_256((gemm_nt_v5_kernel<EPI, OUT_T, SPLITK, 1>)); break;
+    default:     SHIFU_LAUNCH_256((gemm_nt_v4_kernel<EPI, OUT_T, SPLITK, 0>)); break;
+  }
+#undef SHIFU_LAUNCH_256
+}
+
 template <int EPI, typename OUT_T>
 static void launch_nt_v4(const bf16* A, const bf16* B, OUT_T* C, const bf16* bias,
                          long M, long N, long K, int act, hipStream_t s) {
   dim3 grid((N + V4_BN - 1) / V4_BN, (M + V4_BM - 1) / V4_BM);
-  auto kern = gemm_nt_v4_kernel<EPI, OUT_T, false>;
-  hipFuncSetAttribute((const void*)kern,
-                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
-  hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
-                     A, B, C, bias, (int)M, (int)N, (int)K, act, 0L);
+  launch_256<EPI, OUT_T, false>(grid, A, B, C, bias, (int)M, (int)N, (int)K,
+                                act, 0L, s);
 }
 
 template <int EPI, typename OUT_T>
@@ -831,23 +1132,19 @@ static void run_nt_v4_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
                                  long M, long N, long K, hipStream_t s, long z) {
   dim3 grid((unsigned)((N + V4_BN - 1) / V4_BN),
             (unsigned)((M + V4_BM - 1) / V4_BM), (unsigned)z);
-  auto kern = gemm_nt_v4_kernel<EPI_F32, float, true>;
-  hipFuncSetAttribute((const void*)kern,
-                      hipFuncAttributeMaxDynamicSharedMemorySize, V4_LDS_BYTES);
   float* cp = (float*)c.data_ptr();
   if (splitk_slab_mode()) {
     long kt = (K + V4_BK - 1) / V4_BK;
     long zs = splitk_no_empty_z(kt, z);
     grid.z = (unsigned)zs;
     auto w = at::empty({zs, M * N}, c.options());
-    hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
-                       A, B, (float*)w.data_ptr(), nullptr,
-                       (int)M, (int)N, (int)K, 0, M * N);
+    launch_256<EPI_F32, float, true>(grid, A, B, (float*)w.data_ptr(), nullptr,
+                                     (int)M, (int)N, (int)K, 0, M * N, s);
     launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, zs, s);
   } else {
     hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
-    hipLaunchKernelGGL(kern, grid, dim3(512), V4_LDS_BYTES, s,
-                       A, B, cp, nullptr, (int)M, (int)N, (int)K, 0, 0L);
+    launch_256<EPI_F32, float, true>(grid, A, B, cp, nullptr,
+                                     (int)M, (int)N, (int)K, 0, 0L, s);
   }
 }
 
