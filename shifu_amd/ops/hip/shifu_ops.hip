@@ -971,13 +971,208 @@ void gemm_nt_v5_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 }
 
+// ---------------------------------------------------------------------------
+// v6 "NT" GEMM — v4's loop skeleton on mfma_f32_32x32x16_bf16: half the MFMA
+// instructions at a higher per-shape µbench ceiling (2382 vs 2075 TF,
+// cdna_hip_programming.md §3), same LDS image/staging/barrier structure.
+// Fragment maps (verified by mfma_probe32):
+//   A: lane holds A[frag_row = lane&31][k = (lane>>5)*8 + j]
+//   B: lane holds B[frag_col = lane&31][same k]  (NT operand, row-major [N,K])
+//   C: col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+// Per wave: 4(M) x 2(N) fragments of 32x32 -> acc[4][2] f32x16 (128 VGPR);
+// per K-tile 32 MFMAs in two 16-MFMA phases.  bf16 epilogue LDS-staged as v5.
+// ---------------------------------------------------------------------------
+typedef __attribute__((ext_vector_type(16))) float f32x16v;
+
+template <int EPI, typename OUT_T, bool SPLITK = false>
+__global__ __launch_bounds__(512, 1)
+void gemm_nt_v6_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
+                       OUT_T* __restrict__ C, const bf16* __restrict__ bias,
+                       int M, int N, int K, int act, long slab, int swz) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  int bx = blockIdx.x, by = blockIdx.y;
+  xcd_swizzle_xy(bx, by, swz);
+  const int m0 = by * V4_BM;
+  const int n0 = bx * V4_BN;
+  const int r32 = lane & 31;
+  const int kg = lane >> 5;
+  const int rx = r32 & 7;
+
+  int kt_lo = 0, kt_hi = (K + V4_BK - 1) / V4_BK;
+  if (SPLITK) {
+    int nz = gridDim.z;
+    int per = (kt_hi + nz - 1) / nz;
+    kt_lo = blockIdx.z * per;
+    kt_hi = min(kt_hi, kt_lo + per);
+    if (kt_lo >= kt_hi) return;
+  }
+  const int n_kt = kt_hi - kt_lo;
+
+  f32x16v acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x16v{};
+
+  v4_stage_slot(A, B, smem, 0, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 1, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 2, kt_lo, m0, n0, M, N, K, tid);
+  v4_stage_slot(A, B, smem, 3, kt_lo, m0, n0, M, N, K, tid);
+  if (n_kt > 1) {
+    v4_stage_slot(A, B, smem, 2, kt_lo + 1, m0, n0, M, N, K, tid);
+    v4_stage_slot(A, B, smem, 3, kt_lo + 1, m0, n0, M, N, K, tid);
+  }
+  {
+    bool b1_glds = (n_kt > 1) && ((kt_lo + 1) * V4_BK + V4_BK <= K) &&
+                   (n0 + V4_BN <= N);
+    if (b1_glds) asm volatile("s_waitcnt vmcnt(4) lgkmcnt(0)" ::: "memory");
+    else asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+  }
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < n_kt; ++t) {
+    const int kt = kt_lo + t;
+    const int par = kt & 1;
+    const char* Ar = smem + V4_A(par);
+    const char* Br = smem + V4_B(par);
+
+    // B fragments for the whole tile (2 nf x 4 ks)
+    bf16x8 bfr[2][4];
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        int row = wn * 64 + nf * 32 + r32;
+        int g = ks * 2 + kg;
+        bfr[nf][ks] = *(const bf16x8*)(Br + row * 128 + ((g ^ rx) * 16));
+      }
+
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      bf16x8 afr[2][4];
+#pragma unroll
+      for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          int row = wm * 128 + (q * 2 + m2) * 32 + r32;
+          int g = ks * 2 + kg;
+          afr[m2][ks] = *(const bf16x8*)(Ar + row * 128 + ((g ^ rx) * 16));
+        }
+      if (q == 0 && t + 1 < n_kt) {
+        v4_stage_slot(A, B, smem, 0, kt + 1, m0, n0, M, N, K, tid);
+        v4_stage_slot(A, B, smem, 1, kt + 1, m0, n0, M, N, K, tid);
+      }
+      if (q == 1 && t + 2 < n_kt) {
+        v4_stage_slot(A, B, smem, 2, kt + 2, m0, n0, M, N, K, tid);
+        v4_stage_slot(A, B, smem, 3, kt + 2, m0, n0, M, N, K, tid);
+      }
+
+      __builtin_amdgcn_s_setprio(1);
+      // ks OUTER: all 4 independent accumulators issue between two touches
+      // of the same one (32x32 dependent-accumulator latency)
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks)
+#pragma unroll
+        for (int m2 = 0; m2 < 2; ++m2)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+            acc[q * 2 + m2][nf] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                afr[m2][ks], bfr[nf][ks], acc[q * 2 + m2][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+    {
+      bool b2_glds = (t + 2 < n_kt) && ((kt + 2) * V4_BK + V4_BK <= K) &&
+                     (n0 + V4_BN <= N);
+      if (b2_glds) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  if (EPI == EPI_F32) {
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        int col = n0 + wn * 64 + nf * 32 + r32;
+        if (col >= N) continue;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int row = m0 + wm * 128 + mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * kg;
+          if (row >= M) continue;
+          float v = acc[mf][nf][r];
+          if (SPLITK) {
+            if (slab) ((float*)C)[(long)blockIdx.z * slab + (long)row * N + col] = v;
+            else atomicAdd(&((float*)C)[(long)row * N + col], v);
+          } else ((float*)C)[(long)row * N + col] = v;
+        }
+      }
+    return;
+  }
+
+  // bf16 epilogue, LDS-staged (v5's scheme; same bank swizzle: the two kg
+  // halves write rows 4 apart)
+  float bvs[2] = {0.f, 0.f};
+  if (EPI == EPI_BIAS_ACT) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      int col = n0 + wn * 64 + nf * 32 + r32;
+      if (col < N) bvs[nf] = __bfloat162float(bias[col]);
+    }
+  }
+  short* cw = (short*)(smem + wave * 16384);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float v = acc[mf][nf][r];
+        if (EPI == EPI_BIAS_ACT) v = act_fwd(v + bvs[nf], act);
+        bf16 h = __float2bfloat16(v);
+        int row = mf * 32 + (r & 3) + 8 * (r >> 2) + 4 * kg;
+        int col = nf * 32 + r32;
+        int sw = (((row >> 2) & 1) << 5);
+        cw[row * 64 + (col ^ sw)] = *(short*)&h;
+      }
+    }
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+  for (int i = 0; i < 16; ++i) {
+    int elt = i * 64 + lane;
+    int row = elt >> 3;
+    int c8 = (elt & 7) * 8;
+    int grow = m0 + wm * 128 + row;
+    int gcol = n0 + wn * 64 + c8;
+    int sw = (((row >> 2) & 1) << 5);
+    s16x8 v = *(const s16x8*)(cw + row * 64 + (c8 ^ sw));
+    if (grow < M) {
+      if (gcol + 8 <= N) {
+        *(s16x8*)((bf16*)C + (long)grow * N + gcol) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gcol + j < N)
+            ((short*)((bf16*)C + (long)grow * N))[gcol + j] = ((short*)&v)[j];
+      }
+    }
+  }
+}
+
 // 256^2-tile GEMM implementation selector (SHIFU_GEMM_IMPL):
 //   v4  — 2-phase x 32 MFMA, 3 barriers/tile (round-1 shipping kernel)
 //   v4b — v4 with the q1 phase barrier merged into the boundary (2/tile)
 //   v5  — 4-phase x 16 MFMA, 1 half-tile stage/phase, 2 barriers/phase,
 //         LDS-staged dwordx4 epilogue
 //   v5b — v5 with one barrier per phase
-enum GemmImpl { GI_V4 = 0, GI_V4B = 1, GI_V5 = 2, GI_V5B = 3 };
+enum GemmImpl { GI_V4 = 0, GI_V4B = 1, GI_V5 = 2, GI_V5B = 3, GI_V6 = 4 };
 
 static int gemm_impl() {
   static int v = [] {
@@ -987,6 +1182,7 @@ static int gemm_impl() {
     if (!strcmp(e, "v4b")) return (int)GI_V4B;
     if (!strcmp(e, "v5")) return (int)GI_V5;
     if (!strcmp(e, "v5b")) return (int)GI_V5B;
+    if (!strcmp(e, "v6")) return (int)GI_V6;
     return (int)GI_V4;
   }();
   return v;
@@ -1019,6 +1215,7 @@ static void launch_256(dim3 grid, const bf16* A, const bf16* B, OUT_T* C,
     case GI_V4B: SHIFU_LAUNCH_256((gemm_nt_v4_kernel<EPI, OUT_T, SPLITK, 1>)); break;
     case GI_V5:  SHIFU_LAUNCH_256((gemm_nt_v5_kernel<EPI, OUT_T, SPLITK, 0>)); break;
     case GI_V5B: SHIFU_LAUNCH_256((gemm_nt_v5_kernel<EPI, OUT_T, SPLITK, 1>)); break;
+    case GI_V6:  SHIFU_LAUNCH_256((gemm_nt_v6_kernel<EPI, OUT_T, SPLITK>)); break;
     default:     SHIFU_LAUNCH_256((gemm_nt_v4_kernel<EPI, OUT_T, SPLITK, 0>)); break;
   }
 #undef SHIFU_LAUNCH_256
@@ -1549,6 +1746,38 @@ at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   CHECK_GPU(a); CHECK_BF16(a); CHECK_GPU(b); CHECK_BF16(b);
   auto d = at::zeros({16, 16}, a.options().dtype(at::kFloat));
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     (const bf16*)a.contiguous().data_ptr(),
+                     (const bf16*)b.contiguous().data_ptr(), (float*)d.data_ptr());
+  return d;
+}
+
+// 32x32x16 layout probe: a [32,16] row-major, b [32,16] row-major (NT — row
+// n holds that column's k values); d[32,32] = a @ b^T.  Assumed maps:
+//   A frag: lane holds A[lane&31][(lane>>5)*8 + j]
+//   B frag: lane holds B^T[(lane>>5)*8 + j][lane&31] = bsrc[lane&31][...]
+//   C/D   : col = lane&31, row = (reg&3) + 8*(reg>>2) + 4*(lane>>5)
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__global__ void mfma_probe32_kernel(const bf16* a, const bf16* b, float* d) {
+  int lane = threadIdx.x & 63;
+  int r32 = lane & 31, kg = lane >> 5;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ((__bf16*)&af)[j] = *(const __bf16*)&a[r32 * 16 + kg * 8 + j];
+    ((__bf16*)&bf)[j] = *(const __bf16*)&b[r32 * 16 + kg * 8 + j];
+  }
+  f32x16 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    d[((r & 3) + 8 * (r >> 2) + 4 * kg) * 32 + r32] = acc[r];
+}
+
+at::Tensor mfma_probe32(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_BF16(a); CHECK_GPU(b); CHECK_BF16(b);
+  auto d = at::zeros({32, 32}, a.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, cur_stream(),
                      (const bf16*)a.contiguous().data_ptr(),
                      (const bf16*)b.contiguous().data_ptr(), (float*)d.data_ptr());
   return d;
@@ -2336,6 +2565,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fm2_fwd", &fm2_fwd);
   m.def("fm2_bwd", &fm2_bwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma_probe32", &mfma_probe32);
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);
   m.def("act_grad_colsum", &act_grad_colsum);
