@@ -68,6 +68,9 @@ def main():
     ap.add_argument("--n-cat", type=int, default=N_CAT)
     ap.add_argument("--graphs", choices=["auto", "on", "off"], default="auto",
                     help="hipGraph-capture the whole training step")
+    ap.add_argument("--arena", choices=["unified", "split"], default="unified",
+                    help="wide+deep weights in ONE [R,D+2] arena (one gather/"
+                         "scatter/EP-exchange) or two separate arenas")
     ap.add_argument("--emb-mode", choices=["auto", "dp", "ep", "ep_row"],
                     default="auto",
                     help="embedding parallelism: replicated+sparse-allgather "
@@ -97,12 +100,18 @@ def main():
     use_ep = bool(sharded)
     model = WideDeep(N_DENSE, [args.vocab] * args.n_cat, args.embed_dim, TOWER, ACTS,
                      seed=777, sharded_embeddings=sharded, world=world,
-                     rank=rank, emb_fast_init=True).to(device)
+                     rank=rank, emb_fast_init=True,
+                     unified=(args.arena == "unified")).to(device)
     if on_gpu:
         # keep embedding arenas bf16 (HBM-resident, gathered by the HIP kernel)
         for p in model.parameters():
             if getattr(p, "_is_embedding_arena", False):
                 p.data = p.data.to(torch.bfloat16)
+    if world == 1:
+        from shifu_amd.ops.embedding import UnifiedMultiEmbedding
+        for m in model.modules():
+            if isinstance(m, UnifiedMultiEmbedding):
+                m.defer_grads = True   # unpacked-grad fast path
 
     dense_params, emb_params = split_params(model)
     from shifu_amd.ops.flat import bind_mirrors
@@ -209,6 +218,7 @@ def main():
                 "optimizer": "adam+rowwise_adagrad(emb)",
                 "loss": "sigmoid_ce",
                 "hipgraphs": use_graphs,
+                "arena": args.arena,
             },
         }
         print(json.dumps(out), flush=True)

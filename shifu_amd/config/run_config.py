@@ -37,6 +37,10 @@ class RunConfig:
     # -- model family --
     model_type: str = "auto"   # "mlp" | "wide_deep" | "deepfm" | "auto"
     embed_dim: int = 16
+    unified_arena: bool = True # wide scalar weights live as column D of the
+                               # deep [R, D+2] arena: one gather/scatter/EP
+                               # exchange for both (models/wide_deep.py);
+                               # False keeps two separate arenas
     emb_mode: str = "auto"     # "ep"/"ep_table": feature-sharded arenas with
                                # STATIC all-to-all splits (parallel/ep.py
                                # TableShardedEmbedding — no per-step host

@@ -55,14 +55,17 @@ class ShifuMLP(torch.nn.Module):
 
 def build_model(mc: ModelConfig, num_dense: int, vocab_sizes=None,
                 model_type: str = "mlp", embed_dim: int = 16, seed: int = 1234,
-                sharded_embeddings: bool = False, world: int = 1, rank: int = 0):
-    """Model factory over the supported families.  sharded_embeddings=True
-    shards the arenas row%world with all-to-all routing (parallel/ep.py)."""
+                sharded_embeddings: bool = False, world: int = 1, rank: int = 0,
+                unified: bool = False):
+    """Model factory over the supported families.  sharded_embeddings:
+    False | "table" (feature-sharded EP) | "row" (row%world EP); unified=True
+    uses the single [R, D+2] wide+deep arena (models/wide_deep.py)."""
     model_type = model_type.lower()
     vocab_sizes = list(vocab_sizes or [])
     if model_type == "mlp" or not vocab_sizes:
         return ShifuMLP.from_model_config(mc, num_dense, seed=seed)
-    kw = dict(sharded_embeddings=sharded_embeddings, world=world, rank=rank)
+    kw = dict(sharded_embeddings=sharded_embeddings, world=world, rank=rank,
+              unified=unified)
     if model_type in ("wide_deep", "widedeep", "wnd"):
         from shifu_amd.models.wide_deep import WideDeep
         return WideDeep(num_dense, vocab_sizes, embed_dim,

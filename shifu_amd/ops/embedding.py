@@ -110,6 +110,137 @@ def gather_concat(emb: "MultiEmbedding", ids: torch.Tensor,
     return torch.cat([dense, out.to(dense.dtype)], dim=1)
 
 
+class _UnifiedGatherFn(torch.autograd.Function):
+    """ONE [R, D+2] arena (deep cols 0..D-1, wide col D, zero pad D+1) ->
+    (tower_in [B, nd+F*D], wide [B, F]) in a single gather pass
+    (emb_gather_split kernel).  Backward packs the deep and wide output
+    grads into one sparse [n, D+2] value tensor, so the whole wide+deep
+    update runs ONE scatter/adagrad chain instead of round 1's two."""
+
+    @staticmethod
+    def forward(ctx, arena: torch.Tensor, flat_ids: torch.Tensor,
+                dense: torch.Tensor, F: int, D: int, defer: bool = False):
+        B, nd = dense.shape
+        hip = (arena.dtype == torch.bfloat16 and use_hip(arena)
+               and D % 8 == 0 and dense.dtype == torch.bfloat16
+               and not dense.requires_grad)
+        if hip:
+            ext = hip_ops()
+            out = torch.empty(B, nd + F * D, device=dense.device,
+                              dtype=dense.dtype)
+            out[:, :nd].copy_(dense)
+            wide = torch.empty(B, F, device=dense.device, dtype=arena.dtype)
+            ext.emb_gather_split(arena, flat_ids.contiguous(), out, wide,
+                                 nd, D)
+        else:
+            g = arena.index_select(0, flat_ids.reshape(-1))      # [n, DP]
+            deep = g[:, :D].reshape(B, F * D)
+            wide = g[:, D].reshape(B, F)
+            out = torch.cat([dense, deep.to(dense.dtype)], dim=1)
+        ctx.save_for_backward(flat_ids)
+        ctx.meta = (arena.shape, arena.dtype, F, D, nd)
+        ctx.defer = bool(defer) and hip
+        ctx.arena_ref = arena if ctx.defer else None
+        return out, wide
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor, dwide: torch.Tensor):
+        (flat_ids,) = ctx.saved_tensors
+        shape, dtype, F, D, nd = ctx.meta
+        B = dout.shape[0]
+        n = B * F
+        DP = shape[1]
+        if ctx.defer:
+            # fast path (single-rank, no window accumulation): hand the
+            # UNPACKED grad buffers straight to the optimizer's unified
+            # update kernels — no [n, D+2] value materialization, no 132B
+            # row stride in the scatter (ops/hip emb_update_unified)
+            lst = getattr(ctx.arena_ref, "_unified_grads", None)
+            if lst is None:
+                lst = []
+                ctx.arena_ref._unified_grads = lst
+            lst.append((flat_ids.reshape(-1), dout, nd, dwide.contiguous(),
+                        F, D))
+            return None, None, None, None, None, None
+        vals = torch.empty(n, DP, dtype=dtype, device=dout.device)
+        v3 = vals.view(B, F, DP)
+        v3[:, :, :D] = dout[:, nd:].reshape(B, F, D)
+        v3[:, :, D] = dwide.to(dtype)
+        v3[:, :, D + 1:] = 0
+        grad = torch.sparse_coo_tensor(flat_ids.reshape(1, n), vals, shape)
+        return grad, None, None, None, None, None
+
+
+class UnifiedMultiEmbedding(torch.nn.Module):
+    """Wide&Deep unified arena (ROADMAP item 3): per category ONE row holds
+    the deep D-vector, the wide scalar weight and a zero pad column
+    (4B-even rows for the packed-bf16 atomics).  One gather / one
+    scatter+adagrad chain serve both parts; at world>1 this also halves the
+    EP collective count vs the round-1 two-arena pair gather.
+
+    OPTIMIZER SEMANTICS: rowwise adagrad now normalizes the wide weight by
+    the row's COMBINED mean-square gradient instead of its own — not
+    numerically identical to the split arenas (gate: tools/auc_parity.py).
+
+    Exports keep the LOGICAL split (train/export.py splits the arena into
+    embeddings[R,D] + wide_cat[R,1]) so serving and the Java eval bundle
+    layout are unchanged."""
+
+    def __init__(self, vocab_sizes: Sequence[int], dim: int, seed: int = 0,
+                 dtype: torch.dtype = torch.float32, empty_init: bool = False):
+        super().__init__()
+        self.vocab_sizes = [int(v) for v in vocab_sizes]
+        self.dim = int(dim)             # DEEP dim; arena has dim+2 columns
+        self.cols = self.dim + 2
+        self.total_rows = int(sum(self.vocab_sizes))
+        offsets = torch.tensor(
+            [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
+            dtype=torch.int64)
+        self.register_buffer("offsets", offsets, persistent=False)
+        self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64),
+                             persistent=False)
+        if empty_init:
+            arena = torch.empty(self.total_rows, self.cols)
+        else:
+            gen = torch.Generator().manual_seed(seed)
+            arena = (torch.rand(self.total_rows, self.cols, generator=gen) * 2 - 1)
+            arena *= unified_col_scale(self.dim)
+        self.arena = torch.nn.Parameter(arena.to(dtype))
+        self.arena._is_embedding_arena = True
+        self.arena._unified_split = self.dim  # export: cols [:D] deep, [D] wide
+        # deferred-grad fast path: trainer/bench enable it when world==1 and
+        # update_window==1 (the optimizer then consumes the unpacked grad
+        # buffers directly — ops/hip emb_update_unified)
+        self.defer_grads = False
+
+    @property
+    def num_features(self) -> int:
+        return len(self.vocab_sizes)
+
+    def flat_ids(self, ids: torch.Tensor) -> torch.Tensor:
+        local = ids.clamp(min=0) % self.sizes
+        return local + self.offsets
+
+    def gather_split(self, ids: torch.Tensor, dense: torch.Tensor):
+        """-> (tower_in [B, nd+F*D], wide [B, F])."""
+        if ids.shape[1] != self.num_features:
+            raise ValueError(f"ids has {ids.shape[1]} features, expected "
+                             f"{self.num_features}")
+        return _UnifiedGatherFn.apply(self.arena, self.flat_ids(ids), dense,
+                                      self.num_features, self.dim,
+                                      self.defer_grads)
+
+
+def unified_col_scale(dim: int) -> torch.Tensor:
+    """Per-column init scale of a unified arena: deep cols like a D-dim
+    MultiEmbedding (1/sqrt(D)), wide col like the old D=1 arena (scale 1),
+    pad col zero."""
+    s = torch.ones(dim + 2)
+    s[:dim] = 1.0 / math.sqrt(max(dim, 1))
+    s[dim + 1] = 0.0
+    return s
+
+
 class MultiEmbedding(torch.nn.Module):
     """F categorical features -> concatenated [B, F*D] embeddings from one arena."""
 

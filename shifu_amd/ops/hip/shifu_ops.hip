@@ -2328,6 +2328,167 @@ void embedding_gather_into(at::Tensor arena, at::Tensor ids, at::Tensor out,
                      (bf16*)out.data_ptr(), rows, D, F, out_stride, col0);
 }
 
+// Unified Wide&Deep arena gather (ROADMAP item 3): ONE [R, D+2] arena holds
+// each category's deep D-vector (cols 0..D-1), its wide scalar weight
+// (col D) and a zero pad col (D+1, keeps rows 4B-even for the pk-bf16
+// atomics).  One pass writes the deep columns straight into the tower-input
+// concat buffer and the wide column to a separate [B, F] buffer — replacing
+// the round-1 pair of gathers over two arenas.
+__global__ void emb_gather_split_kernel(const bf16* __restrict__ arena,
+                                        const long* __restrict__ ids,
+                                        bf16* __restrict__ out,
+                                        bf16* __restrict__ wide,
+                                        long rows_bf, long D, long DP, long F,
+                                        long out_stride, long col0) {
+  long chunks_per_row = D / 8;
+  long deep_total = rows_bf * chunks_per_row;
+  long total = deep_total + rows_bf;           // + one wide element per row
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    if (t < deep_total) {
+      long rf = t / chunks_per_row;
+      long c = t % chunks_per_row;
+      long row = ids[rf];
+      long b = rf / F, f = rf % F;
+      *(s16x8*)(out + b * out_stride + col0 + f * D + c * 8) =
+          *(const s16x8*)(arena + row * DP + c * 8);
+    } else {
+      long rf = t - deep_total;
+      wide[rf] = arena[ids[rf] * DP + D];
+    }
+  }
+}
+
+// tower_in[:, col0:col0+F*D] and wide[B,F] from one [R, D+2] arena
+void emb_gather_split(at::Tensor arena, at::Tensor ids, at::Tensor out,
+                      at::Tensor wide, long col0, long D) {
+  CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
+  CHECK_GPU(out); CHECK_CONTIG(out); CHECK_BF16(out);
+  CHECK_GPU(wide); CHECK_CONTIG(wide); CHECK_BF16(wide);
+  TORCH_CHECK(ids.scalar_type() == at::kLong, "ids must be int64");
+  long B = ids.size(0), F = ids.size(1), DP = arena.size(1);
+  TORCH_CHECK(D % 8 == 0 && D < DP, "gather_split needs D%8==0 and D < arena cols");
+  long out_stride = out.size(1);
+  TORCH_CHECK(col0 + F * D <= out_stride, "slice out of range");
+  TORCH_CHECK(wide.numel() == B * F, "wide buffer must be [B,F]");
+  long rows = B * F;
+  long total = rows * (D / 8) + rows;
+  int blocks = (int)std::min((total + 255) / 256 + 1, (long)4096);
+  hipLaunchKernelGGL(emb_gather_split_kernel, dim3(blocks), dim3(256), 0, cur_stream(),
+                     (const bf16*)arena.data_ptr(), (const long*)ids.contiguous().data_ptr(),
+                     (bf16*)out.data_ptr(), (bf16*)wide.data_ptr(),
+                     rows, D, DP, F, out_stride, col0);
+}
+
+// ---------------------------------------------------------------------------
+// Unified-arena rowwise updates WITHOUT materializing a packed [n, D+2]
+// gradient: the deep gradient is read straight from the tower-input grad
+// buffer (entry e=(b,f): dgrad[b*dstride + dcol0 + f*D + d]) and the wide
+// gradient from its own buffer with stride wstride.  The wide column updates
+// through the SAME packed-bf16 atomic as the deep pairs — its pair partner
+// is the zero pad column, fed a 0 addend.  Passing dgrad=packed_vals,
+// dstride=F*(D+2), dcol0=0, wide=packed_vals+D, wstride=D+2 makes these the
+// fast path for packed unified values too (the generic D=66 kernels lose
+// >2x to the 132-byte row stride).
+// ---------------------------------------------------------------------------
+__global__ void emb_accsq_uni_kernel(float* __restrict__ acc, const long* __restrict__ rows,
+                                     const bf16* __restrict__ dgrad,
+                                     const bf16* __restrict__ wide,
+                                     long n, long F, long D, long DP,
+                                     long dstride, long dcol0, long wstride) {
+  long e = (long)blockIdx.x * (blockDim.x >> 5) + (threadIdx.x >> 5);
+  if (e >= n) return;
+  int lane = threadIdx.x & 31;
+  long pairs = D >> 1;
+  const unsigned* v2 = (const unsigned*)(dgrad + (e / F) * dstride + dcol0 + (e % F) * D);
+  float sq = 0.0f;
+  for (long p = lane; p < pairs; p += 32) {
+    unsigned u = v2[p];
+    unsigned short lo = (unsigned short)(u & 0xffff), hi = (unsigned short)(u >> 16);
+    float a = __bfloat162float(*(const bf16*)&lo);
+    float b = __bfloat162float(*(const bf16*)&hi);
+    sq += a * a + b * b;
+  }
+  if (lane == 0) {
+    float w = __bfloat162float(wide[e * wstride]);
+    sq += w * w;
+  }
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) sq += __shfl_down(sq, off, 32);
+  if (lane == 0) atomicAdd(&acc[rows[e]], sq / (float)DP);
+}
+
+__global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
+                                       const bf16* __restrict__ dgrad,
+                                       const bf16* __restrict__ wide,
+                                       const float* __restrict__ rowscale,
+                                       long n, long F, long D, long DP,
+                                       long dstride, long dcol0, long wstride,
+                                       float scale) {
+  long pairs = (D >> 1) + 1;               // deep pairs + the (wide, pad) pair
+  long total = n * pairs;
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long t = i; t < total; t += stride) {
+    long e = t / pairs, dp = t % pairs;
+    float sc = scale * (rowscale ? rowscale[e] : 1.0f);
+    long row = rows[e];
+    __hip_bfloat162 add;
+    if (dp < (D >> 1)) {
+      const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D + dp * 2;
+      add.x = __float2bfloat16(sc * __bfloat162float(src[0]));
+      add.y = __float2bfloat16(sc * __bfloat162float(src[1]));
+    } else {
+      add.x = __float2bfloat16(sc * __bfloat162float(wide[e * wstride]));
+      add.y = __float2bfloat16(0.0f);    // pad column stays zero
+    }
+    unsafeAtomicAdd((__hip_bfloat162*)(arena + row * DP + dp * 2), add);
+  }
+}
+
+// adagrad (rowwise, fp32 accumulator) over a unified arena from unpacked
+// grads; kind 0 = sgd (no accumulator)
+void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
+                        at::Tensor dgrad, long dcol0, at::Tensor wide,
+                        long wstride, long F, double lr, double eps,
+                        bool adagrad) {
+  CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
+  CHECK_GPU(dgrad); CHECK_BF16(dgrad);
+  CHECK_GPU(wide); CHECK_BF16(wide);
+  long n = rows.numel();
+  if (!n) return;
+  long DP = arena.size(1), D = DP - 2;
+  long dstride = dgrad.size(1);
+  TORCH_CHECK(D > 0 && D % 2 == 0, "unified update needs even deep D");
+  TORCH_CHECK(dgrad.stride(1) == 1 && dgrad.stride(0) == dstride,
+              "dgrad must be row-contiguous");
+  TORCH_CHECK(dcol0 % 2 == 0 && dstride % 2 == 0,
+              "unified update needs 4B-aligned deep columns");
+  auto s = cur_stream();
+  const float* rsp = nullptr;
+  at::Tensor rowscale;
+  if (adagrad) {
+    CHECK_F32(acc);
+    int epb = 8;
+    hipLaunchKernelGGL(emb_accsq_uni_kernel, dim3((unsigned)((n + epb - 1) / epb)),
+                       dim3(256), 0, s,
+                       (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                       (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
+                       n, F, D, DP, dstride, dcol0, wstride);
+    rowscale = at::empty({n}, acc.options());
+    hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
+                       (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                       (float*)rowscale.data_ptr(), n, (float)eps);
+    rsp = (const float*)rowscale.data_ptr();
+  }
+  long total = n * ((D >> 1) + 1);
+  hipLaunchKernelGGL(emb_scatter_uni_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
+                     (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
+                     (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
+                     rsp, n, F, D, DP, dstride, dcol0, wstride, (float)-lr);
+}
+
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
   CHECK_GPU(arena); CHECK_BF16(arena); CHECK_BF16(vals);
   long n = rows.numel(), D = arena.size(1);
@@ -2578,6 +2739,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adagrad_step", &adagrad_step);
   m.def("embedding_gather", &embedding_gather);
   m.def("embedding_gather_into", &embedding_gather_into);
+  m.def("emb_gather_split", &emb_gather_split);
+  m.def("emb_update_unified", &emb_update_unified);
   m.def("emb_sgd_step", &emb_sgd_step);
   m.def("emb_adagrad_step", &emb_adagrad_step);
 }

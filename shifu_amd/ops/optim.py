@@ -104,6 +104,20 @@ class FusedOptimizer:
 
     # ----------------------------------------------------------------- sparse
     def _emb_step(self, p: torch.nn.Parameter, idx: int) -> None:
+        stash = getattr(p, "_unified_grads", None)
+        if stash:
+            # deferred unified-arena grads (ops/embedding._UnifiedGatherFn
+            # fast path, GPU only): consume the unpacked buffers directly
+            ext = hip_ops()
+            adagrad = self.emb_kind == OPT_ADAGRAD
+            acc = (self.emb_state[idx] if adagrad
+                   else torch.empty(0, device=p.device, dtype=torch.float32))
+            for (rows, dout, nd, dwide, F, _D) in stash:
+                ext.emb_update_unified(p.data, acc, rows, dout, nd,
+                                       dwide, 1, F, self.emb_lr, self.eps,
+                                       adagrad)
+            p._unified_grads = []
+            return
         if p.grad is None:
             return
         from shifu_amd.ops.embedding import sparse_rows_values

@@ -213,7 +213,8 @@ class ShardedEmbedding(torch.nn.Module):
 
     def __init__(self, vocab_sizes: Sequence[int], dim: int, seed: int = 0,
                  world: int = 1, rank: int = 0,
-                 dtype: torch.dtype = torch.float32, fast_init: bool = False):
+                 dtype: torch.dtype = torch.float32, fast_init: bool = False,
+                 col_scale=None):
         super().__init__()
         self.vocab_sizes = [int(v) for v in vocab_sizes]
         self.dim = int(dim)
@@ -226,7 +227,9 @@ class ShardedEmbedding(torch.nn.Module):
         self.register_buffer("sizes", torch.tensor(self.vocab_sizes, dtype=torch.int64),
                              persistent=False)
 
-        scale = 1.0 / math.sqrt(max(self.dim, 1))
+        # per-column init scale (unified arenas) or uniform 1/sqrt(D)
+        scale = (col_scale if col_scale is not None
+                 else 1.0 / math.sqrt(max(self.dim, 1)))
         shard_rows = len(range(self.rank, self.total_rows, self.world))
         if fast_init:
             # per-shard seeded draw (NOT bit-identical to MultiEmbedding's
@@ -416,7 +419,8 @@ class TableShardedEmbedding(torch.nn.Module):
 
     def __init__(self, vocab_sizes: Sequence[int], dim: int, seed: int = 0,
                  world: int = 1, rank: int = 0,
-                 dtype: torch.dtype = torch.float32, fast_init: bool = False):
+                 dtype: torch.dtype = torch.float32, fast_init: bool = False,
+                 col_scale=None):
         super().__init__()
         self.vocab_sizes = [int(v) for v in vocab_sizes]
         self.dim = int(dim)
@@ -467,7 +471,8 @@ class TableShardedEmbedding(torch.nn.Module):
         self.val_in_splits = self.val_out_splits = [0] * self.world
         self._loff = loff
 
-        scale = 1.0 / math.sqrt(max(self.dim, 1))
+        scale = (col_scale if col_scale is not None
+                 else 1.0 / math.sqrt(max(self.dim, 1)))
         if fast_init:
             gen = torch.Generator().manual_seed(seed * 1000003 + rank)
             shard = (torch.rand(self.local_rows, self.dim, generator=gen) * 2 - 1) * scale

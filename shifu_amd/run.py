@@ -69,7 +69,8 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
         model = build_model(mc, len(rc.selected_numeric_columns), vocab,
                             model_type=rc.resolved_model_type(),
                             embed_dim=rc.embed_dim, seed=rc.seed,
-                            sharded_embeddings=sharded, world=world, rank=rank)
+                            sharded_embeddings=sharded, world=world, rank=rank,
+                            unified=getattr(rc, "unified_arena", True))
         trainer = Trainer(model, mc, rc, train, valid, rank=rank,
                           world_size=world, device=device, metric_sink=metric_sink,
                           heartbeat=heartbeat)

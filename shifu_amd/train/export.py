@@ -57,12 +57,31 @@ def _graph_spec(model: torch.nn.Module) -> Dict:
     raise ValueError(f"unsupported model family for export: {type(model).__name__}")
 
 
+def _split_unified_arenas(model: torch.nn.Module, state: Dict) -> Dict:
+    """Unified [R, D+2] arenas export as the LOGICAL split layout (deep
+    [R, D] under the original name + wide [R, 1] under the split model's
+    wide-arena name), so serving, score.py and old bundles are identical for
+    both training layouts (ROADMAP item 3)."""
+    wide_name = {"WideDeep": "wide_cat.arena", "DeepFM": "fm_first.arena"} \
+        .get(type(model).__name__)
+    for name, p in list(model.named_parameters()):
+        D = getattr(p, "_unified_split", None)
+        if D is None or name not in state:
+            continue
+        full = state.pop(name)
+        state[name] = full[:, :D].contiguous()
+        if wide_name:
+            state[wide_name] = full[:, D:D + 1].contiguous()
+    return state
+
+
 def export_model(model: torch.nn.Module, final_model_path: str,
                  model_name: str = "model", algorithm: str = "NN",
                  selected_columns: Optional[Sequence[int]] = None) -> str:
     """Write the export directory; returns its path."""
     os.makedirs(final_model_path, exist_ok=True)
     state = {k: v.detach().float().cpu().contiguous() for k, v in model.state_dict().items()}
+    state = _split_unified_arenas(model, state)
     try:
         from safetensors.torch import save_file
         weights_file = "model.safetensors"

@@ -200,6 +200,13 @@ class Trainer:
         self.loss_kind = p.loss
         self.batch_size = int(rc.batch_size or p.batch_size)
         self.update_window = max(int(p.update_window), 1)
+        if world_size == 1 and self.update_window == 1:
+            # unified arenas: hand unpacked grads straight to the optimizer
+            # (no [n, D+2] packing); needs single-rank, no window accumulation
+            from shifu_amd.ops.embedding import UnifiedMultiEmbedding
+            for m in self.model.modules():
+                if isinstance(m, UnifiedMultiEmbedding):
+                    m.defer_grads = True
         self.epochs = int(rc.epochs or mc.num_train_epochs)
         self.global_step = 0
         self.start_epoch = 0
@@ -562,6 +569,9 @@ class Trainer:
             rep = MultiEmbedding(mod.vocab_sizes, mod.dim, empty_init=True,
                                  dtype=shard.dtype).to(shard.device)
             rep.arena.data = full
+            if hasattr(mod.arena, "_unified_split"):
+                # unified EP arena: the export splits it wide/deep
+                rep.arena._unified_split = mod.arena._unified_split
             parent = self.model
             parts = name.split(".")
             for part in parts[:-1]:

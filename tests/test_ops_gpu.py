@@ -460,3 +460,105 @@ def test_splitk_atomic_fallback_mode():
                        capture_output=True, text=True, timeout=600)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "ATOMIC_OK" in r.stdout
+
+
+def test_unified_gather_split_gpu():
+    """emb_gather_split kernel: deep columns into the concat buffer + wide
+    column to its own [B,F], against a plain index_select reference."""
+    from shifu_amd.ops.embedding import UnifiedMultiEmbedding
+    torch.manual_seed(4)
+    emb = UnifiedMultiEmbedding([500, 300, 200], dim=16, seed=9,
+                                dtype=torch.bfloat16).cuda()
+    B, nd = 257, 12
+    g = torch.Generator().manual_seed(1)
+    ids = torch.stack([torch.randint(0, v, (B,), generator=g)
+                       for v in [500, 300, 200]], dim=1).cuda()
+    dense = torch.randn(B, nd, generator=g).to(torch.bfloat16).cuda()
+    out, wide = emb.gather_split(ids, dense)
+
+    flat = emb.flat_ids(ids).reshape(-1)
+    ref = emb.arena.detach().index_select(0, flat).reshape(B, 3, 18)
+    assert torch.equal(out[:, nd:].reshape(B, 3, 16), ref[:, :, :16])
+    assert torch.equal(wide, ref[:, :, 16])
+    assert torch.equal(out[:, :nd], dense)
+
+
+def test_unified_wide_deep_step_gpu():
+    """Full unified Wide&Deep training step on the HIP path: loss finite,
+    arena rows move, pad column stays zero."""
+    from shifu_amd.models.wide_deep import WideDeep
+    from shifu_amd.ops.flat import FlatParams, split_params, bind_mirrors
+    from shifu_amd.ops.loss import weighted_loss
+    from shifu_amd.ops.optim import FusedOptimizer
+    torch.manual_seed(0)
+    model = WideDeep(32, [4000] * 4, 16, [64], ["relu"], seed=1,
+                     unified=True).cuda()
+    for p in model.parameters():
+        if getattr(p, "_is_embedding_arena", False):
+            p.data = p.data.to(torch.bfloat16)
+    dense_params, emb_params = split_params(model)
+    flat = FlatParams(dense_params, mirror_bf16=True)
+    bind_mirrors(model, flat)
+    opt = FusedOptimizer(flat, emb_params, optimizer="adam", lr=1e-3,
+                         emb_optimizer="adagrad", emb_lr=0.05)
+    g = torch.Generator().manual_seed(2)
+    dense = torch.randn(512, 32, generator=g).to(torch.bfloat16).cuda()
+    cats = torch.randint(0, 4000, (512, 4), generator=g).cuda()
+    y = (torch.rand(512, generator=g) > 0.5).float().cuda()
+    w = torch.ones(512, device="cuda")
+    arena0 = model.embeddings.arena.data.clone()
+    for _ in range(3):
+        loss = weighted_loss(model(dense, cats), y, w, "sigmoid_ce")
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+    assert torch.isfinite(loss).all()
+    D = model.embed_dim
+    a = model.embeddings.arena.data
+    assert not torch.equal(a, arena0), "arena never updated"
+    assert torch.all(a[:, D + 1].float() == 0), "pad column corrupted"
+
+
+def test_unified_deferred_matches_packed():
+    """The deferred unpacked-grad update path (emb_update_unified) must match
+    the packed [n, D+2] sparse path over several adagrad steps."""
+    from shifu_amd.models.wide_deep import WideDeep
+    from shifu_amd.ops.embedding import UnifiedMultiEmbedding
+    from shifu_amd.ops.flat import FlatParams, split_params, bind_mirrors
+    from shifu_amd.ops.loss import weighted_loss
+    from shifu_amd.ops.optim import FusedOptimizer
+
+    def run(defer):
+        torch.manual_seed(0)
+        model = WideDeep(16, [2000, 3000], 16, [32], ["relu"], seed=3,
+                         unified=True).cuda()
+        for p in model.parameters():
+            if getattr(p, "_is_embedding_arena", False):
+                p.data = p.data.to(torch.bfloat16)
+        for m in model.modules():
+            if isinstance(m, UnifiedMultiEmbedding):
+                m.defer_grads = defer
+        dps, eps_ = split_params(model)
+        flat = FlatParams(dps, mirror_bf16=True)
+        bind_mirrors(model, flat)
+        opt = FusedOptimizer(flat, eps_, optimizer="adam", lr=1e-3,
+                             emb_optimizer="adagrad", emb_lr=0.05)
+        g = torch.Generator().manual_seed(2)
+        for _ in range(4):
+            dense = torch.randn(256, 16, generator=g).to(torch.bfloat16).cuda()
+            cats = torch.stack([torch.randint(0, 2000, (256,), generator=g),
+                                torch.randint(0, 3000, (256,), generator=g)],
+                               dim=1).cuda()
+            y = (torch.rand(256, generator=g) > 0.5).float().cuda()
+            loss = weighted_loss(model(dense, cats), y,
+                                 torch.ones(256, device="cuda"), "sigmoid_ce")
+            loss.backward()
+            opt.step()
+            opt.zero_grad()
+        return (model.embeddings.arena.data.float().cpu(),
+                opt.emb_state[0].cpu())
+
+    a1, s1 = run(True)
+    a0, s0 = run(False)
+    assert torch.allclose(s1, s0, atol=1e-4), "adagrad accumulators diverged"
+    assert torch.allclose(a1, a0, atol=2e-2), "arena values diverged"
