@@ -1900,6 +1900,115 @@ std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
   return {dz, db};
 }
 
+// act-grad + colsum + TRANSPOSE in one pass: dz = dy*act'(y) written both
+// row-major (dgrad operand) and transposed (wgrad operand), plus db =
+// colsum(dz).  Replaces act_grad_colsum + a separate transpose_bf16 of dz
+// (one full read of dz saved per layer per step, and one launch).
+// 64x64 bf16 tiles staged in LDS at a 66-short (odd-dword) row stride:
+// column reads in the transpose phase then hit 32 distinct banks.
+__global__ __launch_bounds__(256)
+void act_grad_colsum_T_kernel(const bf16* __restrict__ dy, const bf16* __restrict__ y,
+                              bf16* __restrict__ dz, bf16* __restrict__ dzT,
+                              float* __restrict__ db, long B, long N, int act) {
+  __shared__ short tile[64 * 66];
+  __shared__ float csum[64];
+  long n0 = (long)blockIdx.x * 64, b0 = (long)blockIdx.y * 64;
+  int t = threadIdx.x;
+  if (t < 64) csum[t] = 0.0f;
+  __syncthreads();
+
+  const bool vec = (N % 8 == 0);
+  int c8 = (t & 7) * 8;
+  int r_ = t >> 3;
+#pragma unroll
+  for (int rr = 0; rr < 2; ++rr) {
+    int r = r_ + rr * 32;
+    long gb = b0 + r;
+    s16x8 out;
+    if (gb < B) {
+      long base = gb * N + n0 + c8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ((short*)&out)[j] = 0;
+      if (vec && n0 + c8 + 8 <= N) {
+        s16x8 vdy = *(const s16x8*)(dy + base);
+        s16x8 vy = *(const s16x8*)(y + base);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = __bfloat162float(*(const bf16*)&((const short*)&vdy)[j]) *
+                    act_grad_from_y(__bfloat162float(*(const bf16*)&((const short*)&vy)[j]), act);
+          bf16 h = __float2bfloat16(g);
+          ((short*)&out)[j] = *(short*)&h;
+        }
+        *(s16x8*)(dz + base) = out;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          long gn = n0 + c8 + j;
+          if (gn < N) {
+            float g = __bfloat162float(dy[gb * N + gn]) *
+                      act_grad_from_y(__bfloat162float(y[gb * N + gn]), act);
+            bf16 h = __float2bfloat16(g);
+            ((short*)&out)[j] = *(short*)&h;
+            dz[gb * N + gn] = h;
+          }
+        }
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ((short*)&out)[j] = 0;
+    }
+#pragma unroll
+    for (int j = 0; j < 8; j += 2)
+      *(unsigned*)&tile[r * 66 + c8 + j] = *(unsigned*)&((short*)&out)[j];
+  }
+  __syncthreads();
+
+  // transpose out + column sums: task = (column c, 8-row chunk)
+#pragma unroll
+  for (int it = 0; it < 2; ++it) {
+    int task = t + it * 256;
+    int c = task >> 3;
+    int r8 = (task & 7) * 8;
+    s16x8 v;
+    float part = 0.0f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      short s = tile[(r8 + j) * 66 + c];
+      ((short*)&v)[j] = s;
+      part += __bfloat162float(*(bf16*)&s);
+    }
+    atomicAdd(&csum[c], part);
+    long gn = n0 + c;
+    long gb8 = b0 + r8;
+    if (gn < N) {
+      if (gb8 + 8 <= B) {
+        *(s16x8*)(dzT + gn * B + gb8) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gb8 + j < B) dzT[gn * B + gb8 + j] = *(bf16*)&((short*)&v)[j];
+      }
+    }
+  }
+  __syncthreads();
+  if (t < 64 && n0 + t < N) atomicAdd(&db[n0 + t], csum[t]);
+}
+
+std::vector<at::Tensor> act_grad_colsum_T(at::Tensor dy, at::Tensor y, long act) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+  long B = dy.size(0), N = dy.size(1);
+  auto dz = at::empty_like(dy);
+  auto dzT = at::empty({N, B}, dy.options());
+  auto db = at::zeros({N}, dy.options().dtype(at::kFloat));
+  dim3 grid((unsigned)((N + 63) / 64), (unsigned)((B + 63) / 64));
+  hipLaunchKernelGGL(act_grad_colsum_T_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                     (bf16*)dz.data_ptr(), (bf16*)dzT.data_ptr(),
+                     (float*)db.data_ptr(), B, N, (int)act);
+  return {dz, dzT, db};
+}
+
 // ---------------------------------------------------------------------------
 // fused sigmoid + weighted loss (K3)
 // fwd: p = sigmoid(z); per = w*(p-y)^2 (wmse) or w*bce (ce);
@@ -2730,6 +2839,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);
   m.def("act_grad_colsum", &act_grad_colsum);
+  m.def("act_grad_colsum_T", &act_grad_colsum_T);
   m.def("weighted_loss_fwd", &weighted_loss_fwd);
   m.def("weighted_loss_bwd", &weighted_loss_bwd);
   m.def("sgd_step", &sgd_step);

@@ -142,8 +142,10 @@ class _FusedLinearFn(torch.autograd.Function):
                 if not ctx.x_needs_grad:
                     dx = None
                 return dx, dw, db, None, None, None, None, None
-            # one fused pass: dz = dy*act'(y) and db = colsum(dz)
-            dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
+            # one fused pass: dz = dy*act'(y) (row-major AND transposed)
+            # + db = colsum(dz) — the dzT side feeds wgrad directly, so no
+            # separate transpose of dz runs (act_grad_colsum_T kernel)
+            dz, dzT, db = ext.act_grad_colsum_T(dy.contiguous(), y, act)
             if _ASYNC_WGRAD and gv_w is not None and gv_b is not None:
                 # wgrad on a side stream, accumulated straight into the flat
                 # arena; autograd gets None (no AccumulateGrad for w/b)
@@ -154,14 +156,13 @@ class _FusedLinearFn(torch.autograd.Function):
                 ws.wait_event(ev)
                 with torch.cuda.stream(ws):
                     xT = ext.transpose_bf16(x)
-                    dzT = ext.transpose_bf16(dz)
                     dwv = ext.gemm_ntv3_f32(dzT, xT)
                     gv_w.add_(dwv)
                     gv_b.add_(db)
                     done = torch.cuda.Event()
                     done.record(ws)
                 # caching-allocator safety: main-stream tensors used on ws
-                for t in (x, dz, db):
+                for t in (x, dzT, db):
                     t.record_stream(ws)
                 _WGRAD_EVENTS.append(done)
                 dx = None
@@ -170,7 +171,6 @@ class _FusedLinearFn(torch.autograd.Function):
                     dx = ext.gemm_ntv3_bf16(dz, wT)
                 return dx, None, None, None, None, None, None, None
             xT = ext.transpose_bf16(x)                # [K,B]
-            dzT = ext.transpose_bf16(dz)              # [N,B]
             dw = ext.gemm_ntv3_f32(dzT, xT)           # fp32 [N,K], split-K
             dx = None
             if ctx.x_needs_grad:

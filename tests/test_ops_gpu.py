@@ -562,3 +562,21 @@ def test_unified_deferred_matches_packed():
     a0, s0 = run(False)
     assert torch.allclose(s1, s0, atol=1e-4), "adagrad accumulators diverged"
     assert torch.allclose(a1, a0, atol=2e-2), "arena values diverged"
+
+
+@pytest.mark.parametrize("B,N,act", [(128, 64, 3), (513, 130, 2), (64, 33, 1),
+                                     (1000, 256, 3), (70, 7, 0)])
+def test_act_grad_colsum_T(B, N, act):
+    """Fused dz + dz^T + colsum against the fp32 reference (edge shapes
+    exercise the guarded tile paths)."""
+    dy = _rand_bf16(B, N, seed=B + 40)
+    y = torch.sigmoid(_rand_bf16(B, N, seed=N + 41).float()).to(torch.bfloat16).cuda()
+    dz, dzT, db = hip_ops().act_grad_colsum_T(dy, y, act)
+    grads = {0: lambda yy: torch.ones_like(yy), 1: lambda yy: yy * (1 - yy),
+             2: lambda yy: 1 - yy * yy, 3: lambda yy: (yy > 0).float()}
+    ref = dy.float() * grads[act](y.float())
+    ref_b = ref.to(torch.bfloat16).float()
+    assert torch.equal(dz.float(), ref_b), "dz mismatch"
+    assert torch.equal(dzT.float(), ref_b.t().contiguous()), "dzT mismatch"
+    ok, err = _rel_close(db, ref_b.sum(0), 1e-2)
+    assert ok, f"db maxdiff={err}"
