@@ -2109,7 +2109,7 @@ at::Tensor weighted_loss_bwd(at::Tensor p, at::Tensor y, at::Tensor w, long kind
 // K4 — fused optimizers over flat fp32 arena (g includes coupled L2 in-kernel)
 // ---------------------------------------------------------------------------
 __global__ void sgd_kernel(float* __restrict__ w, const float* __restrict__ g,
-                           long n, float lr, float l2) {
+                           bf16* __restrict__ mir, long n, float lr, float l2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   long n4 = n / 4;
@@ -2118,12 +2118,25 @@ __global__ void sgd_kernel(float* __restrict__ w, const float* __restrict__ g,
 #pragma unroll
     for (int j = 0; j < 4; ++j) wv[j] -= lr * (gv[j] + l2 * wv[j]);
     ((f32x4*)w)[v] = wv;
+    if (mir) {
+      s16x4t pack;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        bf16 h = __float2bfloat16(wv[j]);
+        ((short*)&pack)[j] = *(short*)&h;
+      }
+      *(s16x4t*)(mir + v * 4) = pack;
+    }
   }
-  for (long j = n4 * 4 + i; j < n; j += stride) w[j] -= lr * (g[j] + l2 * w[j]);
+  for (long j = n4 * 4 + i; j < n; j += stride) {
+    w[j] -= lr * (g[j] + l2 * w[j]);
+    if (mir) mir[j] = __float2bfloat16(w[j]);
+  }
 }
 
 __global__ void adam_kernel(float* __restrict__ w, const float* __restrict__ g,
                             float* __restrict__ m, float* __restrict__ v,
+                            bf16* __restrict__ mir,
                             long n, float lr, float b1, float b2, float eps,
                             float l2, float bc1, float sbc2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -2134,12 +2147,15 @@ __global__ void adam_kernel(float* __restrict__ w, const float* __restrict__ g,
     float mj = b1 * m[j] + (1.0f - b1) * gj;
     float vj = b2 * v[j] + (1.0f - b2) * gj * gj;
     m[j] = mj; v[j] = vj;
-    w[j] -= step * mj / (sqrtf(vj) + eps * sbc2);
+    float wn = w[j] - step * mj / (sqrtf(vj) + eps * sbc2);
+    w[j] = wn;
+    if (mir) mir[j] = __float2bfloat16(wn);
   }
 }
 
 __global__ void adadelta_kernel(float* __restrict__ w, const float* __restrict__ g,
                                 float* __restrict__ acc, float* __restrict__ dacc,
+                                bf16* __restrict__ mir,
                                 long n, float lr, float rho, float eps, float l2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
@@ -2149,39 +2165,50 @@ __global__ void adadelta_kernel(float* __restrict__ w, const float* __restrict__
     acc[j] = a;
     float upd = gj * sqrtf(dacc[j] + eps) / sqrtf(a + eps);
     dacc[j] = rho * dacc[j] + (1.0f - rho) * upd * upd;
-    w[j] -= lr * upd;
+    float wn = w[j] - lr * upd;
+    w[j] = wn;
+    if (mir) mir[j] = __float2bfloat16(wn);
   }
 }
 
 __global__ void adagrad_kernel(float* __restrict__ w, const float* __restrict__ g,
-                               float* __restrict__ m, long n, float lr, float eps, float l2) {
+                               float* __restrict__ m, bf16* __restrict__ mir,
+                               long n, float lr, float eps, float l2) {
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (long j = i; j < n; j += stride) {
     float gj = g[j] + l2 * w[j];
     float mj = m[j] + gj * gj;
     m[j] = mj;
-    w[j] -= lr * gj / (sqrtf(mj) + eps);
+    float wn = w[j] - lr * gj / (sqrtf(mj) + eps);
+    w[j] = wn;
+    if (mir) mir[j] = __float2bfloat16(wn);
   }
 }
 
 static int opt_blocks(long n) { return (int)std::min((n + 1023) / 1024 + 1, (long)2048); }
 
-void sgd_step(at::Tensor w, at::Tensor g, double lr, double l2) {
+static inline bf16* mir_ptr(at::Tensor& mir) {
+  return mir.numel() ? (bf16*)mir.data_ptr() : nullptr;
+}
+
+void sgd_step(at::Tensor w, at::Tensor g, at::Tensor mir, double lr, double l2) {
   CHECK_GPU(w); CHECK_F32(w); CHECK_F32(g);
   hipLaunchKernelGGL(sgd_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
-                     (float*)w.data_ptr(), (const float*)g.data_ptr(), w.numel(),
-                     (float)lr, (float)l2);
+                     (float*)w.data_ptr(), (const float*)g.data_ptr(),
+                     mir_ptr(mir), w.numel(), (float)lr, (float)l2);
 }
 
 void adam_step(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor v,
+               at::Tensor mir,
                double lr, double b1, double b2, double eps, double l2, long t) {
   CHECK_GPU(w); CHECK_F32(w);
   float bc1 = 1.0f - powf((float)b1, (float)t);
   float sbc2 = sqrtf(1.0f - powf((float)b2, (float)t));
   hipLaunchKernelGGL(adam_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
                      (float*)w.data_ptr(), (const float*)g.data_ptr(),
-                     (float*)m.data_ptr(), (float*)v.data_ptr(), w.numel(),
+                     (float*)m.data_ptr(), (float*)v.data_ptr(), mir_ptr(mir),
+                     w.numel(),
                      (float)lr, (float)b1, (float)b2, (float)eps, (float)l2, bc1, sbc2);
 }
 
@@ -2194,6 +2221,7 @@ __global__ void step_incr_kernel(float* step_buf) {
 
 __global__ void adam_dev_kernel(float* __restrict__ w, const float* __restrict__ g,
                                 float* __restrict__ m, float* __restrict__ v,
+                                bf16* __restrict__ mir,
                                 const float* __restrict__ step_buf, long n,
                                 float lr, float b1, float b2, float eps, float l2) {
   float t = step_buf[0];
@@ -2207,39 +2235,42 @@ __global__ void adam_dev_kernel(float* __restrict__ w, const float* __restrict__
     float mj = b1 * m[j] + (1.0f - b1) * gj;
     float vj = b2 * v[j] + (1.0f - b2) * gj * gj;
     m[j] = mj; v[j] = vj;
-    w[j] -= step * mj / (sqrtf(vj) + eps * sbc2);
+    float wn = w[j] - step * mj / (sqrtf(vj) + eps * sbc2);
+    w[j] = wn;
+    if (mir) mir[j] = __float2bfloat16(wn);
   }
 }
 
 void adam_step_dev(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor v,
-                   at::Tensor step_buf, double lr, double b1, double b2,
-                   double eps, double l2) {
+                   at::Tensor mir, at::Tensor step_buf, double lr, double b1,
+                   double b2, double eps, double l2) {
   CHECK_GPU(w); CHECK_F32(w); CHECK_F32(step_buf);
   auto s = cur_stream();
   hipLaunchKernelGGL(step_incr_kernel, dim3(1), dim3(64), 0, s,
                      (float*)step_buf.data_ptr());
   hipLaunchKernelGGL(adam_dev_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, s,
                      (float*)w.data_ptr(), (const float*)g.data_ptr(),
-                     (float*)m.data_ptr(), (float*)v.data_ptr(),
+                     (float*)m.data_ptr(), (float*)v.data_ptr(), mir_ptr(mir),
                      (const float*)step_buf.data_ptr(), w.numel(),
                      (float)lr, (float)b1, (float)b2, (float)eps, (float)l2);
 }
 
 void adadelta_step(at::Tensor w, at::Tensor g, at::Tensor acc, at::Tensor dacc,
-                   double lr, double rho, double eps, double l2) {
+                   at::Tensor mir, double lr, double rho, double eps, double l2) {
   CHECK_GPU(w); CHECK_F32(w);
   hipLaunchKernelGGL(adadelta_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
                      (float*)w.data_ptr(), (const float*)g.data_ptr(),
-                     (float*)acc.data_ptr(), (float*)dacc.data_ptr(), w.numel(),
-                     (float)lr, (float)rho, (float)eps, (float)l2);
+                     (float*)acc.data_ptr(), (float*)dacc.data_ptr(), mir_ptr(mir),
+                     w.numel(), (float)lr, (float)rho, (float)eps, (float)l2);
 }
 
-void adagrad_step(at::Tensor w, at::Tensor g, at::Tensor m,
+void adagrad_step(at::Tensor w, at::Tensor g, at::Tensor m, at::Tensor mir,
                   double lr, double eps, double l2) {
   CHECK_GPU(w); CHECK_F32(w);
   hipLaunchKernelGGL(adagrad_kernel, dim3(opt_blocks(w.numel())), dim3(256), 0, cur_stream(),
                      (float*)w.data_ptr(), (const float*)g.data_ptr(),
-                     (float*)m.data_ptr(), w.numel(), (float)lr, (float)eps, (float)l2);
+                     (float*)m.data_ptr(), mir_ptr(mir), w.numel(),
+                     (float)lr, (float)eps, (float)l2);
 }
 
 // ---------------------------------------------------------------------------
@@ -2535,24 +2566,30 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
                                        long n, long F, long D, long DP,
                                        long dstride, long dcol0, long wstride,
                                        float scale) {
-  long pairs = (D >> 1) + 1;               // deep pairs + the (wide, pad) pair
-  long total = n * pairs;
-  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long stride = (long)gridDim.x * blockDim.x;
-  for (long t = i; t < total; t += stride) {
-    long e = t / pairs, dp = t % pairs;
+  // 32 lanes per entry: lane k applies deep pairs k, k+32, ...; lane 0 also
+  // the (wide, pad) pair.  Shift-only index math — a 64-bit `t / 33` per
+  // element (the naive pairs-flattened mapping) measured ~2x slower.
+  long e = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 5;
+  long estride = ((long)gridDim.x * blockDim.x) >> 5;
+  int lane = threadIdx.x & 31;
+  long hp = D >> 1;
+  for (; e < n; e += estride) {
     float sc = scale * (rowscale ? rowscale[e] : 1.0f);
     long row = rows[e];
-    __hip_bfloat162 add;
-    if (dp < (D >> 1)) {
-      const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D + dp * 2;
-      add.x = __float2bfloat16(sc * __bfloat162float(src[0]));
-      add.y = __float2bfloat16(sc * __bfloat162float(src[1]));
-    } else {
+    const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D;
+    bf16* dst = arena + row * DP;
+    for (long dp = lane; dp < hp; dp += 32) {
+      __hip_bfloat162 add;
+      add.x = __float2bfloat16(sc * __bfloat162float(src[dp * 2]));
+      add.y = __float2bfloat16(sc * __bfloat162float(src[dp * 2 + 1]));
+      unsafeAtomicAdd((__hip_bfloat162*)(dst + dp * 2), add);
+    }
+    if (lane == 0) {
+      __hip_bfloat162 add;
       add.x = __float2bfloat16(sc * __bfloat162float(wide[e * wstride]));
       add.y = __float2bfloat16(0.0f);    // pad column stays zero
+      unsafeAtomicAdd((__hip_bfloat162*)(dst + D), add);
     }
-    unsafeAtomicAdd((__hip_bfloat162*)(arena + row * DP + dp * 2), add);
   }
 }
 
@@ -2591,7 +2628,7 @@ void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
                        (float*)rowscale.data_ptr(), n, (float)eps);
     rsp = (const float*)rowscale.data_ptr();
   }
-  long total = n * ((D >> 1) + 1);
+  long total = n * 32;   // 32 lanes per entry (shift-mapped)
   hipLaunchKernelGGL(emb_scatter_uni_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
                      (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),

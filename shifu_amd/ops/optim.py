@@ -84,23 +84,30 @@ class FusedOptimizer:
             self.m.addcmul_(g, g, value=1.0)
             w.addcdiv_(g, self.m.sqrt().add(self.eps), value=-self.lr)
 
-    def _dense_step_hip(self) -> None:
+    def _dense_step_hip(self) -> bool:
+        """Returns True when the kernel also refreshed the bf16 mirror
+        (the cast is fused into the update pass — one less arena-wide
+        copy per step)."""
         ext = hip_ops()
         w, g = self.flat.flat, self.flat.flat_grad
+        mir = self.flat.mirror
+        if mir is None:
+            mir = torch.empty(0, device=w.device, dtype=torch.bfloat16)
         if self.kind == OPT_SGD:
-            ext.sgd_step(w, g, self.lr, self.l2)
+            ext.sgd_step(w, g, mir, self.lr, self.l2)
         elif self.kind == OPT_ADAM:
             # device-side step counter -> hipGraph-replayable bias correction
             if not hasattr(self, "_step_buf"):
                 self._step_buf = torch.zeros(1, device=w.device)
                 self._step_buf.fill_(float(self.step_count - 1))
-            ext.adam_step_dev(w, g, self.m, self.v, self._step_buf,
+            ext.adam_step_dev(w, g, self.m, self.v, mir, self._step_buf,
                               self.lr, self.b1, self.b2, self.eps, self.l2)
         elif self.kind == OPT_ADADELTA:
-            ext.adadelta_step(w, g, self.m, self.v, self.lr, self.rho,
+            ext.adadelta_step(w, g, self.m, self.v, mir, self.lr, self.rho,
                               self.eps, self.l2)
         elif self.kind == OPT_ADAGRAD:
-            ext.adagrad_step(w, g, self.m, self.lr, self.eps, self.l2)
+            ext.adagrad_step(w, g, self.m, mir, self.lr, self.eps, self.l2)
+        return True
 
     # ----------------------------------------------------------------- sparse
     def _emb_step(self, p: torch.nn.Parameter, idx: int) -> None:
@@ -153,10 +160,10 @@ class FusedOptimizer:
                 drain_wgrad_events()
             self.flat.sync_grads()
             if use_hip(self.flat.flat):
-                self._dense_step_hip()
+                self._dense_step_hip()   # bf16 mirror refreshed in-kernel
             else:
                 self._dense_step_ref()
-            self.flat.refresh_mirror()   # one arena-wide bf16 cast per step
+                self.flat.refresh_mirror()
         for i, p in enumerate(self.emb_params):
             self._emb_step(p, i)
 
