@@ -93,6 +93,16 @@ def export_model(model: torch.nn.Module, final_model_path: str,
     with open(os.path.join(final_model_path, "graph.json"), "w") as f:
         json.dump(_graph_spec(model), f, indent=2)
 
+    # TF-1.x SavedModel for the Java eval drop-in (SavedModelBundle.load,
+    # TensorflowModel.java:169): frozen-Const graph, dense-input families
+    # only (the reference's eval path feeds a single float vector)
+    from shifu_amd.models.mlp import ShifuMLP
+    if isinstance(model, ShifuMLP):
+        from shifu_amd.train.tf_saved_model import (emit_saved_model,
+                                                    layers_from_mlp)
+        emit_saved_model(final_model_path, layers_from_mlp(model),
+                         model.num_features)
+
     # GenericModelConfig.json — field-compatible with the reference's
     # export_generic_config (ssgd_monitor.py:476-490)
     gmc = {

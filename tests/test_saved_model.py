@@ -1,0 +1,168 @@
+"""TF SavedModel emitter verification (train/tf_saved_model.py).
+
+An INDEPENDENT pure-python protobuf wire parser decodes the emitted
+`saved_model.pb`, checks the structural contract the TF-1.4 Java loader
+relies on (schema version, tags, frozen graph / no variables), re-executes
+the graph with numpy, and compares scores against the training model."""
+import math
+import os
+import struct
+
+import numpy as np
+import torch
+
+from shifu_amd.models.mlp import ShifuMLP
+from shifu_amd.train.tf_saved_model import emit_saved_model, layers_from_mlp
+
+
+# ------------------------------------------------------- generic wire parser
+def parse_message(buf):
+    """-> {field: [(wire, value), ...]} with raw bytes for len-delim."""
+    out = {}
+    i = 0
+    while i < len(buf):
+        tag, i = read_varint(buf, i)
+        field, wire = tag >> 3, tag & 7
+        if wire == 0:
+            v, i = read_varint(buf, i)
+        elif wire == 2:
+            ln, i = read_varint(buf, i)
+            v = buf[i:i + ln]
+            i += ln
+        elif wire == 5:
+            v = struct.unpack("<I", buf[i:i + 4])[0]
+            i += 4
+        elif wire == 1:
+            v = struct.unpack("<Q", buf[i:i + 8])[0]
+            i += 8
+        else:
+            raise ValueError(f"wire type {wire}")
+        out.setdefault(field, []).append((wire, v))
+    return out
+
+
+def read_varint(buf, i):
+    shift, val = 0, 0
+    while True:
+        b = buf[i]
+        i += 1
+        val |= (b & 0x7F) << shift
+        if not (b & 0x80):
+            return val, i
+        shift += 7
+
+
+def _shape(buf):
+    dims = []
+    for _, d in parse_message(buf).get(2, []):
+        size = parse_message(d)[1][0][1]
+        if size >= 1 << 63:
+            size -= 1 << 64
+        dims.append(size)
+    return dims
+
+
+def _tensor(buf):
+    m = parse_message(buf)
+    assert m[1][0][1] == 1, "dtype must be DT_FLOAT"
+    dims = _shape(m[2][0][1]) if 2 in m else []
+    content = m[4][0][1]
+    return np.frombuffer(content, dtype=np.float32).reshape(dims)
+
+
+def load_graph(path):
+    """saved_model.pb -> (nodes dict, tags, has_signature)."""
+    blob = open(path, "rb").read()
+    sm = parse_message(blob)
+    assert sm[1][0][1] == 1, "saved_model_schema_version must be 1"
+    mg = parse_message(sm[2][0][1])
+    meta_info = parse_message(mg[1][0][1])
+    tags = [v.decode() for _, v in meta_info.get(4, [])]
+    graph = parse_message(mg[2][0][1])
+    nodes = {}
+    for _, nb in graph[1]:
+        n = parse_message(nb)
+        name = n[1][0][1].decode()
+        op = n[2][0][1].decode()
+        inputs = [v.decode() for _, v in n.get(3, [])]
+        attrs = {}
+        for _, ab in n.get(5, []):
+            e = parse_message(ab)
+            attrs[e[1][0][1].decode()] = parse_message(e[2][0][1])
+        nodes[name] = (op, inputs, attrs)
+    return nodes, tags, (5 in mg)
+
+
+def run_graph(nodes, feeds):
+    vals = dict(feeds)
+
+    def ev(name):
+        if name in vals:
+            return vals[name]
+        op, inputs, attrs = nodes[name]
+        if op == "Const":
+            v = _tensor(attrs["value"][8][0][1])
+        elif op == "MatMul":
+            v = ev(inputs[0]) @ ev(inputs[1])
+        elif op == "Add":
+            v = ev(inputs[0]) + ev(inputs[1])
+        elif op == "Mul":
+            v = ev(inputs[0]) * ev(inputs[1])
+        elif op == "Maximum":
+            v = np.maximum(ev(inputs[0]), ev(inputs[1]))
+        elif op == "Sigmoid":
+            v = 1.0 / (1.0 + np.exp(-ev(inputs[0])))
+        elif op == "Tanh":
+            v = np.tanh(ev(inputs[0]))
+        elif op == "Relu":
+            v = np.maximum(ev(inputs[0]), 0.0)
+        else:
+            raise ValueError(f"unsupported op {op}")
+        vals[name] = v
+        return v
+
+    return ev
+
+
+def test_saved_model_contract(tmp_path):
+    model = ShifuMLP(12, [16, 8], ["relu", "tanh"], seed=4)
+    emit_saved_model(str(tmp_path), layers_from_mlp(model), 12)
+
+    path = tmp_path / "saved_model.pb"
+    assert path.exists()
+    nodes, tags, has_sig = load_graph(str(path))
+    assert tags == ["serve"], f"tags {tags} (TensorflowModel.java default)"
+    assert has_sig
+    assert "shifu_input_0" in nodes and nodes["shifu_input_0"][0] == "Placeholder"
+    assert nodes["shifu_output_0"][0] == "Sigmoid"
+    # frozen graph: no Variable/restore machinery, no variables dir ->
+    # loader.cc skips RunRestore
+    assert not any(op in ("VariableV2", "Assign", "RestoreV2")
+                   for op, _, _ in nodes.values())
+    assert not os.path.exists(tmp_path / "variables" / "variables.index")
+
+
+@torch.no_grad()
+def test_saved_model_scores_match_model(tmp_path):
+    for acts in (["relu", "relu"], ["sigmoid", "tanh"], ["leakyrelu", "relu"]):
+        model = ShifuMLP(20, [32, 16], acts, seed=7)
+        emit_saved_model(str(tmp_path), layers_from_mlp(model), 20)
+        nodes, _, _ = load_graph(str(tmp_path / "saved_model.pb"))
+
+        g = torch.Generator().manual_seed(3)
+        x = torch.randn(64, 20, generator=g)
+        want = torch.sigmoid(model(x)).reshape(-1).numpy()
+        ev = run_graph(nodes, {"shifu_input_0": x.numpy()})
+        got = ev("shifu_output_0").reshape(-1)
+        assert np.allclose(got, want, atol=1e-5), f"acts={acts}"
+
+
+def test_saved_model_single_row_java_shape(tmp_path):
+    """The Java compute() path: feed [1][N] float, read [0][0]."""
+    model = ShifuMLP(1522, [30], ["tanh"], seed=1)   # the reference test's 1522
+    emit_saved_model(str(tmp_path), layers_from_mlp(model), 1522)
+    nodes, _, _ = load_graph(str(tmp_path / "saved_model.pb"))
+    x = np.random.default_rng(0).standard_normal((1, 1522)).astype(np.float32)
+    p = run_graph(nodes, {"shifu_input_0": x})("shifu_output_0")
+    assert p.shape == (1, 1)
+    assert 0.0 <= float(p[0, 0]) <= 1.0   # the reference test's assertion
