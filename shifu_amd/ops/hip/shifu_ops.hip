@@ -2395,6 +2395,43 @@ __global__ void emb_scatter_kernel(bf16* __restrict__ arena, const long* __restr
   }
 }
 
+// generic scatter with the adagrad denominator inline: sc =
+// -lr/(sqrt(acc[row])+eps) per entry (replaces emb_denom + rowscale buffer)
+__global__ void emb_scatter_acc_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
+                                       const bf16* __restrict__ vals,
+                                       const float* __restrict__ acc,
+                                       long n, long D, float scale, float eps) {
+  long pairs = D / 2;
+  long total = n * (pairs ? pairs : 1);
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  if (pairs) {
+    for (long t = i; t < total; t += stride) {
+      long e = t / pairs, dp = t % pairs;
+      long row = rows[e];
+      float sc = scale / (sqrtf(acc[row]) + eps);
+      __hip_bfloat162 add;
+      add.x = __float2bfloat16(sc * __bfloat162float(vals[e * D + dp * 2]));
+      add.y = __float2bfloat16(sc * __bfloat162float(vals[e * D + dp * 2 + 1]));
+      unsafeAtomicAdd((__hip_bfloat162*)(arena + row * D + dp * 2), add);
+    }
+    if (D & 1) {
+      for (long e = i; e < n; e += stride) {
+        long row = rows[e];
+        float sc = scale / (sqrtf(acc[row]) + eps);
+        atomic_add_bf16_scalar(arena + row * D + D - 1,
+                               sc * __bfloat162float(vals[e * D + D - 1]));
+      }
+    }
+  } else {
+    for (long e = i; e < n; e += stride) {
+      long row = rows[e];
+      float sc = scale / (sqrtf(acc[row]) + eps);
+      atomic_add_bf16_scalar(arena + row, sc * __bfloat162float(vals[e]));
+    }
+  }
+}
+
 // small-D phase 1: one THREAD per entry (a wave per entry would idle 63/64
 // lanes at D=1 — the wide-column arena case)
 __global__ void emb_accsq_small_kernel(float* __restrict__ acc, const long* __restrict__ rows,
@@ -2562,10 +2599,10 @@ __global__ void emb_accsq_uni_kernel(float* __restrict__ acc, const long* __rest
 __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
                                        const bf16* __restrict__ dgrad,
                                        const bf16* __restrict__ wide,
-                                       const float* __restrict__ rowscale,
+                                       const float* __restrict__ acc,
                                        long n, long F, long D, long DP,
                                        long dstride, long dcol0, long wstride,
-                                       float scale) {
+                                       float scale, float eps) {
   // 32 lanes per entry: lane k applies deep pairs k, k+32, ...; lane 0 also
   // the (wide, pad) pair.  Shift-only index math — a 64-bit `t / 33` per
   // element (the naive pairs-flattened mapping) measured ~2x slower.
@@ -2574,8 +2611,10 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
   int lane = threadIdx.x & 31;
   long hp = D >> 1;
   for (; e < n; e += estride) {
-    float sc = scale * (rowscale ? rowscale[e] : 1.0f);
     long row = rows[e];
+    // adagrad denominator inline (accsq pass completed): one broadcast
+    // read replaces the separate emb_denom kernel + rowscale buffer
+    float sc = acc ? scale / (sqrtf(acc[row]) + eps) : scale;
     const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D;
     bf16* dst = arena + row * DP;
     for (long dp = lane; dp < hp; dp += 32) {
@@ -2612,8 +2651,7 @@ void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
   TORCH_CHECK(dcol0 % 2 == 0 && dstride % 2 == 0,
               "unified update needs 4B-aligned deep columns");
   auto s = cur_stream();
-  const float* rsp = nullptr;
-  at::Tensor rowscale;
+  const float* accp = nullptr;
   if (adagrad) {
     CHECK_F32(acc);
     int epb = 8;
@@ -2622,17 +2660,14 @@ void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                        (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
                        n, F, D, DP, dstride, dcol0, wstride);
-    rowscale = at::empty({n}, acc.options());
-    hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
-                       (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
-                       (float*)rowscale.data_ptr(), n, (float)eps);
-    rsp = (const float*)rowscale.data_ptr();
+    accp = (const float*)acc.data_ptr();
   }
   long total = n * 32;   // 32 lanes per entry (shift-mapped)
   hipLaunchKernelGGL(emb_scatter_uni_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
                      (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
-                     rsp, n, F, D, DP, dstride, dcol0, wstride, (float)-lr);
+                     accp, n, F, D, DP, dstride, dcol0, wstride, (float)-lr,
+                     (float)eps);
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
@@ -2662,15 +2697,11 @@ void emb_adagrad_step(at::Tensor arena, at::Tensor acc, at::Tensor rows, at::Ten
                        (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
                        (const bf16*)vals.data_ptr(), n, D);
   }
-  auto rowscale = at::empty({n}, vals.options().dtype(at::kFloat));
-  hipLaunchKernelGGL(emb_denom_kernel, dim3(scat_blocks(n)), dim3(256), 0, s,
-                     (const float*)acc.data_ptr(), (const long*)rows.data_ptr(),
-                     (float*)rowscale.data_ptr(), n, (float)eps);
   long total = n * std::max<long>(D / 2, 1);
-  hipLaunchKernelGGL(emb_scatter_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
+  hipLaunchKernelGGL(emb_scatter_acc_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
-                     (const bf16*)vals.data_ptr(), (const float*)rowscale.data_ptr(),
-                     n, D, (float)-lr);
+                     (const bf16*)vals.data_ptr(), (const float*)acc.data_ptr(),
+                     n, D, (float)-lr, (float)eps);
 }
 
 // ---------------------------------------------------------------------------
