@@ -86,7 +86,12 @@ class RunConfig:
     checkpoint_every_epochs: int = 1
     checkpoint_every_secs: float = 0.0   # >0: also time-based mid-epoch saves
                                          # (reference: Supervisor save_model_secs,
-                                         #  ssgd.py:124-128)
+                                         #  ssgd.py:124-128); rank-local clocks,
+                                         # so disabled for EP-sharded arenas
+    checkpoint_every_steps: int = 0      # >0: mid-epoch saves every N sync
+                                         # steps — rank-synchronized (uniform
+                                         # stepping), so valid for EP shards
+                                         # too
 
     # -- misc --
     log_dir: str = "./logs"

@@ -16,6 +16,20 @@ import numpy as np
 import torch
 
 
+class _SavedModelGraph:
+    """Minimal predict() wrapper over a frozen saved_model.pb (numpy
+    execution — train/tf_saved_model.py run_saved_model)."""
+
+    def __init__(self, nodes):
+        self.nodes = nodes
+
+    def predict(self, dense: "torch.Tensor"):
+        from shifu_amd.train.tf_saved_model import run_saved_model
+        x = dense.detach().cpu().float().numpy()
+        p = run_saved_model(self.nodes, {"shifu_input_0": x})
+        return torch.from_numpy(np.ascontiguousarray(p))
+
+
 class ShifuScorer:
     def __init__(self):
         self.model: Optional[torch.nn.Module] = None
@@ -32,13 +46,33 @@ class ShifuScorer:
             os.path.abspath(generic_model_config_path))
         if not os.path.isdir(model_path):
             model_path = os.path.dirname(os.path.abspath(generic_model_config_path))
-        from shifu_amd.train.export import load_exported
-        self.model = load_exported(model_path, device=device)
-        self.device = device
-        with open(os.path.join(model_path, "graph.json")) as f:
-            spec = json.load(f)
-        self.num_dense = int(spec["num_dense"])
-        self.num_cat = len(spec.get("vocab_sizes", []))
+        gj = os.path.join(model_path, "graph.json")
+        if os.path.exists(gj):
+            from shifu_amd.train.export import load_exported
+            self.model = load_exported(model_path, device=device)
+            self.device = device
+            with open(gj) as f:
+                spec = json.load(f)
+            self.num_dense = int(spec["num_dense"])
+            self.num_cat = len(spec.get("vocab_sizes", []))
+            return
+        # no portable layout: score straight from the TF SavedModel artifact
+        # (the same file the Java SavedModelBundle path loads)
+        pb = os.path.join(model_path, "saved_model.pb")
+        if not os.path.exists(pb):
+            raise FileNotFoundError(
+                f"{model_path}: neither graph.json nor saved_model.pb")
+        from shifu_amd.train.tf_saved_model import load_saved_model
+        self._pb_nodes, _ = load_saved_model(pb)
+        op, _, attrs = self._pb_nodes["shifu_input_0"]
+        assert op == "Placeholder"
+        # input width from the placeholder's declared shape
+        from shifu_amd.train.tf_saved_model import _parse_shape
+        dims = _parse_shape(attrs["shape"][7][0][1])
+        self.num_dense = int(dims[-1])
+        self.num_cat = 0
+        self.model = _SavedModelGraph(self._pb_nodes)
+        self.device = "cpu"
 
     def compute(self, row: Sequence[float]) -> float:
         """Score one row: first num_dense values are normalized floats, the

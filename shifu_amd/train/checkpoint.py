@@ -86,7 +86,7 @@ def _atomic_save(blob: dict, path: str) -> None:
 def save_checkpoint(ckpt_dir: str, epoch: int, global_step: int,
                     model: torch.nn.Module, optimizer, keep_last: int = 3,
                     extra: Optional[dict] = None, rank: int = 0,
-                    world: int = 1) -> str:
+                    world: int = 1, stamp: Optional[int] = None) -> str:
     """Write this rank's part of checkpoint `epoch`.  Rank 0 writes the main
     file; every rank owning EP shards writes its shard file.  Call from ALL
     ranks when the model holds ShardedEmbeddings."""
@@ -109,7 +109,7 @@ def save_checkpoint(ckpt_dir: str, epoch: int, global_step: int,
                 if acc is not None:
                     arena_state[n] = acc.cpu()
         _atomic_save({
-            "epoch": epoch, "rank": rank, "world": world,
+            "epoch": epoch, "rank": rank, "world": world, "stamp": stamp,
             "arenas": {n: params[n].data.cpu() for n in sharded},
             "arena_state": arena_state,
         }, shard_checkpoint_path(ckpt_dir, epoch, rank, world))
@@ -118,6 +118,7 @@ def save_checkpoint(ckpt_dir: str, epoch: int, global_step: int,
         skip = set(sharded)
         _atomic_save({
             "epoch": epoch,
+            "stamp": stamp,
             "global_step": global_step,
             "model": {k: v.cpu() for k, v in model.state_dict().items()
                       if k not in skip},
@@ -186,6 +187,12 @@ def load_checkpoint(path: str, model: torch.nn.Module, optimizer=None,
             spath = shard_checkpoint_path(os.path.dirname(path),
                                           int(blob["epoch"]), rank, world)
             sblob = torch.load(spath, map_location="cpu", weights_only=False)
+            if sblob.get("stamp") != blob.get("stamp"):
+                raise RuntimeError(
+                    f"checkpoint {path} and shard {spath} come from different "
+                    "mid-epoch save points (the job died during a cadence "
+                    "save); delete this epoch's ckpt files to resume from "
+                    "the previous complete checkpoint")
             for n in sharded_here:
                 src = sblob["arenas"][n]
                 if src.shape != params[n].shape:

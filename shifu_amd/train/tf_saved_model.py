@@ -219,6 +219,128 @@ def emit_saved_model(final_model_path: str, layers, num_dense: int,
     return out
 
 
+# --------------------------------------------------------------- reader side
+# Independent decode + numpy re-execution of an emitted saved_model.pb.
+# Used by tests (emitter verification) and by serve.ShifuScorer, which can
+# score straight from the TF artifact — the same file the Java
+# SavedModelBundle path loads.
+
+def _read_varint(buf: bytes, i: int):
+    shift, val = 0, 0
+    while True:
+        b = buf[i]
+        i += 1
+        val |= (b & 0x7F) << shift
+        if not (b & 0x80):
+            return val, i
+        shift += 7
+
+
+def parse_message(buf: bytes) -> Dict:
+    """Generic wire decode -> {field: [(wire_type, value), ...]}."""
+    out: Dict = {}
+    i = 0
+    while i < len(buf):
+        tag, i = _read_varint(buf, i)
+        field, wire = tag >> 3, tag & 7
+        if wire == 0:
+            v, i = _read_varint(buf, i)
+        elif wire == 2:
+            ln, i = _read_varint(buf, i)
+            v = buf[i:i + ln]
+            i += ln
+        elif wire == 5:
+            v = struct.unpack("<I", buf[i:i + 4])[0]
+            i += 4
+        elif wire == 1:
+            v = struct.unpack("<Q", buf[i:i + 8])[0]
+            i += 8
+        else:
+            raise ValueError(f"wire type {wire}")
+        out.setdefault(field, []).append((wire, v))
+    return out
+
+
+def _parse_shape(buf: bytes) -> List[int]:
+    dims = []
+    for _, d in parse_message(buf).get(2, []):
+        size = parse_message(d)[1][0][1]
+        if size >= 1 << 63:
+            size -= 1 << 64
+        dims.append(size)
+    return dims
+
+
+def _parse_tensor(buf: bytes) -> np.ndarray:
+    m = parse_message(buf)
+    if m[1][0][1] != DT_FLOAT:
+        raise ValueError("only DT_FLOAT tensors supported")
+    dims = _parse_shape(m[2][0][1]) if 2 in m else []
+    return np.frombuffer(m[4][0][1], dtype=np.float32).reshape(dims)
+
+
+def load_saved_model(path: str):
+    """saved_model.pb (file or directory) -> (nodes, tags).
+
+    nodes: {name: (op, [inputs], {attr: parsed AttrValue})}."""
+    if os.path.isdir(path):
+        path = os.path.join(path, "saved_model.pb")
+    with open(path, "rb") as f:
+        blob = f.read()
+    sm = parse_message(blob)
+    if sm[1][0][1] != 1:
+        raise ValueError("unsupported saved_model_schema_version")
+    mg = parse_message(sm[2][0][1])
+    meta_info = parse_message(mg[1][0][1])
+    tags = [v.decode() for _, v in meta_info.get(4, [])]
+    graph = parse_message(mg[2][0][1])
+    nodes = {}
+    for _, nb in graph[1]:
+        n = parse_message(nb)
+        name = n[1][0][1].decode()
+        op = n[2][0][1].decode()
+        inputs = [v.decode() for _, v in n.get(3, [])]
+        attrs = {}
+        for _, ab in n.get(5, []):
+            e = parse_message(ab)
+            attrs[e[1][0][1].decode()] = parse_message(e[2][0][1])
+        nodes[name] = (op, inputs, attrs)
+    return nodes, tags
+
+
+def run_saved_model(nodes: Dict, feeds: Dict[str, np.ndarray],
+                    fetch: str = "shifu_output_0") -> np.ndarray:
+    """Execute the frozen graph with numpy (the op set the emitter uses)."""
+    vals = dict(feeds)
+
+    def ev(name):
+        if name in vals:
+            return vals[name]
+        op, inputs, attrs = nodes[name]
+        if op == "Const":
+            v = _parse_tensor(attrs["value"][8][0][1])
+        elif op == "MatMul":
+            v = ev(inputs[0]) @ ev(inputs[1])
+        elif op == "Add":
+            v = ev(inputs[0]) + ev(inputs[1])
+        elif op == "Mul":
+            v = ev(inputs[0]) * ev(inputs[1])
+        elif op == "Maximum":
+            v = np.maximum(ev(inputs[0]), ev(inputs[1]))
+        elif op == "Sigmoid":
+            v = 1.0 / (1.0 + np.exp(-ev(inputs[0])))
+        elif op == "Tanh":
+            v = np.tanh(ev(inputs[0]))
+        elif op == "Relu":
+            v = np.maximum(ev(inputs[0]), 0.0)
+        else:
+            raise ValueError(f"unsupported op {op}")
+        vals[name] = v
+        return v
+
+    return ev(fetch)
+
+
 def layers_from_mlp(model) -> List:
     """(W [in,out], b, act) list from a ShifuMLP (weights stored [out,in])."""
     layers = []

@@ -522,6 +522,15 @@ class Trainer:
                 self._save_checkpoint(epoch - 1, global_step=step_pre,
                                       np_rng=rng_pre)
                 self._ckpt_last = time.time()
+            ces = int(getattr(self.rc, "checkpoint_every_steps", 0))
+            if ces > 0 and sync and (si + 1) % ces == 0 and si + 1 < n_steps:
+                # step-cadence save: si is identical on every rank (uniform
+                # stepping), so EP shards stay mutually consistent; the stamp
+                # lets resume DETECT a crash that interleaved two save points
+                self._save_checkpoint(epoch - 1, global_step=step_pre,
+                                      np_rng=rng_pre, stamp=self.global_step)
+                if is_distributed():
+                    torch.distributed.barrier()
         mean_loss = float(torch.stack(losses).float().mean()) if losses else 0.0
         if self.device.type == "cuda":
             torch.cuda.synchronize()
@@ -542,15 +551,16 @@ class Trainer:
         )
 
     def _save_checkpoint(self, epoch: int, global_step: Optional[int] = None,
-                         np_rng=None) -> None:
+                         np_rng=None, stamp: Optional[int] = None) -> None:
         """EP-aware save: every rank writes its shard, rank 0 the main file
-        (train/checkpoint.py layout)."""
+        (train/checkpoint.py layout).  `stamp` disambiguates repeated saves
+        under the same epoch tag (step-cadence mid-epoch saves)."""
         gs = self.global_step if global_step is None else global_step
         rng = self._rng.bit_generator.state if np_rng is None else np_rng
         ckpt.save_checkpoint(self.rc.tmp_model_path, epoch, gs,
                              self.model, self.optimizer,
                              extra={"np_rng": rng},
-                             rank=self.rank, world=self.world)
+                             rank=self.rank, world=self.world, stamp=stamp)
 
     def maybe_resume(self) -> None:
         path = ckpt.latest_checkpoint(self.rc.tmp_model_path, world=self.world)

