@@ -44,28 +44,33 @@ def _emb_factory(vocab_sizes, sharded, world: int, rank: int, fast_init: bool):
 
 def make_unified_embedding(vocab_sizes, embed_dim: int, seed: int,
                            sharded, world: int, rank: int, fast_init: bool):
-    """One [R, D+2] arena for wide+deep (ops/embedding.py
-    UnifiedMultiEmbedding docstring).  Sharded variants reuse the EP modules
-    with dim=D+2 and the unified per-column init scale."""
-    from shifu_amd.ops.embedding import (UnifiedMultiEmbedding,
-                                         unified_col_scale)
+    """One [R, D+4] arena for wide+deep+rowwise-adagrad-accumulator
+    (ops/embedding.py UnifiedMultiEmbedding docstring).  Sharded variants
+    reuse the EP modules with dim=D+4 and the same per-column init scale,
+    so a sharded model is numerically identical to the replicated one."""
+    import math as _math
+    import torch as _torch
+    from shifu_amd.ops.embedding import UnifiedMultiEmbedding
     if not sharded or world <= 1:
         return UnifiedMultiEmbedding(vocab_sizes, embed_dim, seed=seed)
     mode = "table" if sharded is True else str(sharded)
-    cs = unified_col_scale(embed_dim)
+    cs = _torch.ones(embed_dim + 4)
+    cs[:embed_dim] = 1.0 / _math.sqrt(max(embed_dim, 1))
+    cs[embed_dim + 1:] = 0.0      # pad + accumulator start at zero
     if mode == "table":
         from shifu_amd.parallel.ep import TableShardedEmbedding
-        m = TableShardedEmbedding(vocab_sizes, embed_dim + 2, seed=seed,
+        m = TableShardedEmbedding(vocab_sizes, embed_dim + 4, seed=seed,
                                   world=world, rank=rank, fast_init=fast_init,
                                   col_scale=cs)
     elif mode == "row":
         from shifu_amd.parallel.ep import ShardedEmbedding
-        m = ShardedEmbedding(vocab_sizes, embed_dim + 2, seed=seed,
+        m = ShardedEmbedding(vocab_sizes, embed_dim + 4, seed=seed,
                              world=world, rank=rank, fast_init=fast_init,
                              col_scale=cs)
     else:
         raise ValueError(f"unknown sharded_embeddings mode {sharded!r}")
     m.arena._unified_split = embed_dim
+    m.arena._acc_in_arena = True
     return m
 
 
@@ -113,10 +118,11 @@ class WideDeep(torch.nn.Module):
         D = self.embed_dim
         if isinstance(self.embeddings, UnifiedMultiEmbedding):
             return self.embeddings.gather_split(cats, dense)
-        # EP-sharded unified arena: one routing pass moves [.., D+2] rows
+        # EP-sharded unified arena: one routing pass moves [.., D+4] rows
         B = dense.shape[0]
-        out = self.embeddings(cats)                  # [B, F*(D+2)]
-        v = out.view(B, -1, D + 2)
+        DP = self.embeddings.dim                     # EP module arena width
+        out = self.embeddings(cats)                  # [B, F*DP]
+        v = out.view(B, -1, DP)
         deep = v[:, :, :D].reshape(B, -1)
         wide_e = v[:, :, D]
         return torch.cat([dense, deep.to(dense.dtype)], dim=1), wide_e

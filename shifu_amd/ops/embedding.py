@@ -190,8 +190,14 @@ class UnifiedMultiEmbedding(torch.nn.Module):
                  dtype: torch.dtype = torch.float32, empty_init: bool = False):
         super().__init__()
         self.vocab_sizes = [int(v) for v in vocab_sizes]
-        self.dim = int(dim)             # DEEP dim; arena has dim+2 columns
-        self.cols = self.dim + 2
+        self.dim = int(dim)             # DEEP dim; arena has dim+4 columns:
+        # [0..D) deep | D wide | D+1 zero pad | D+2..D+3 rowwise-adagrad
+        # accumulator (raw f32 bits in two bf16 slots on GPU, a plain f32 in
+        # col D+2 on fp32/CPU arenas).  Keeping the accumulator IN the row
+        # puts the whole update chain (accsq atomic, denominator read,
+        # scatter) on one cacheline neighborhood per row instead of a second
+        # random-access array.
+        self.cols = self.dim + 4
         self.total_rows = int(sum(self.vocab_sizes))
         offsets = torch.tensor(
             [0] + list(torch.cumsum(torch.tensor(self.vocab_sizes), 0)[:-1]),
@@ -204,10 +210,14 @@ class UnifiedMultiEmbedding(torch.nn.Module):
         else:
             gen = torch.Generator().manual_seed(seed)
             arena = (torch.rand(self.total_rows, self.cols, generator=gen) * 2 - 1)
-            arena *= unified_col_scale(self.dim)
+            s = torch.ones(self.cols)
+            s[:self.dim] = 1.0 / math.sqrt(max(self.dim, 1))
+            s[self.dim + 1:] = 0.0     # pad + accumulator start at zero
+            arena *= s
         self.arena = torch.nn.Parameter(arena.to(dtype))
         self.arena._is_embedding_arena = True
         self.arena._unified_split = self.dim  # export: cols [:D] deep, [D] wide
+        self.arena._acc_in_arena = True       # optimizer: no external emb_state
         # deferred-grad fast path: trainer/bench enable it when world==1 and
         # update_window==1 (the optimizer then consumes the unpacked grad
         # buffers directly — ops/hip emb_update_unified)
@@ -220,6 +230,14 @@ class UnifiedMultiEmbedding(torch.nn.Module):
     def flat_ids(self, ids: torch.Tensor) -> torch.Tensor:
         local = ids.clamp(min=0) % self.sizes
         return local + self.offsets
+
+    def adagrad_acc(self) -> torch.Tensor:
+        """Decode the in-row rowwise-adagrad accumulator -> [R] f32."""
+        D = self.dim
+        if self.arena.dtype == torch.bfloat16:
+            raw = self.arena.data[:, D + 2:D + 4].contiguous()
+            return raw.view(torch.float32).reshape(-1)
+        return self.arena.data[:, D + 2].clone()
 
     def gather_split(self, ids: torch.Tensor, dense: torch.Tensor):
         """-> (tower_in [B, nd+F*D], wide [B, F])."""

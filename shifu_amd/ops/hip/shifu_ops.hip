@@ -2569,11 +2569,16 @@ void emb_gather_split(at::Tensor arena, at::Tensor ids, at::Tensor out,
 // fast path for packed unified values too (the generic D=66 kernels lose
 // >2x to the 132-byte row stride).
 // ---------------------------------------------------------------------------
+// accumulator addressing: addr = acc + row*astride + aoff.  External [R]
+// array: astride=1, aoff=0.  IN-ARENA accumulator (f32 bits in bf16 cols
+// D+2..D+3): acc = (float*)arena, astride=DP/2, aoff=(D+2)/2 — the atomic
+// then lands in the same row the scatter updates.
 __global__ void emb_accsq_uni_kernel(float* __restrict__ acc, const long* __restrict__ rows,
                                      const bf16* __restrict__ dgrad,
                                      const bf16* __restrict__ wide,
                                      long n, long F, long D, long DP,
-                                     long dstride, long dcol0, long wstride) {
+                                     long dstride, long dcol0, long wstride,
+                                     long astride, long aoff) {
   long e = (long)blockIdx.x * (blockDim.x >> 5) + (threadIdx.x >> 5);
   if (e >= n) return;
   int lane = threadIdx.x & 31;
@@ -2593,7 +2598,7 @@ __global__ void emb_accsq_uni_kernel(float* __restrict__ acc, const long* __rest
   }
 #pragma unroll
   for (int off = 16; off > 0; off >>= 1) sq += __shfl_down(sq, off, 32);
-  if (lane == 0) atomicAdd(&acc[rows[e]], sq / (float)DP);
+  if (lane == 0) atomicAdd(&acc[rows[e] * astride + aoff], sq / (float)DP);
 }
 
 __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __restrict__ rows,
@@ -2602,7 +2607,8 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
                                        const float* __restrict__ acc,
                                        long n, long F, long D, long DP,
                                        long dstride, long dcol0, long wstride,
-                                       float scale, float eps) {
+                                       float scale, float eps,
+                                       long astride, long aoff) {
   // 32 lanes per entry: lane k applies deep pairs k, k+32, ...; lane 0 also
   // the (wide, pad) pair.  Shift-only index math — a 64-bit `t / 33` per
   // element (the naive pairs-flattened mapping) measured ~2x slower.
@@ -2614,7 +2620,7 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
     long row = rows[e];
     // adagrad denominator inline (accsq pass completed): one broadcast
     // read replaces the separate emb_denom kernel + rowscale buffer
-    float sc = acc ? scale / (sqrtf(acc[row]) + eps) : scale;
+    float sc = acc ? scale / (sqrtf(acc[row * astride + aoff]) + eps) : scale;
     const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D;
     bf16* dst = arena + row * DP;
     for (long dp = lane; dp < hp; dp += 32) {
@@ -2637,13 +2643,14 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
 void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
                         at::Tensor dgrad, long dcol0, at::Tensor wide,
                         long wstride, long F, double lr, double eps,
-                        bool adagrad) {
+                        bool adagrad, bool acc_in_arena) {
   CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
   CHECK_GPU(dgrad); CHECK_BF16(dgrad);
   CHECK_GPU(wide); CHECK_BF16(wide);
   long n = rows.numel();
   if (!n) return;
-  long DP = arena.size(1), D = DP - 2;
+  long DP = arena.size(1);
+  long D = DP - (acc_in_arena ? 4 : 2);
   long dstride = dgrad.size(1);
   TORCH_CHECK(D > 0 && D % 2 == 0, "unified update needs even deep D");
   TORCH_CHECK(dgrad.stride(1) == 1 && dgrad.stride(0) == dstride,
@@ -2652,22 +2659,34 @@ void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
               "unified update needs 4B-aligned deep columns");
   auto s = cur_stream();
   const float* accp = nullptr;
-  if (adagrad) {
+  float* accw = nullptr;
+  long astride = 1, aoff = 0;
+  if (acc_in_arena) {
+    // D+4 layout: raw f32 accumulator in bf16 cols D+2..D+3 (4B-aligned
+    // because D and DP are even)
+    TORCH_CHECK(D % 2 == 0 && DP % 2 == 0, "acc_in_arena needs even D/DP");
+    accw = (float*)arena.data_ptr();
+    astride = DP / 2;
+    aoff = (D + 2) / 2;
+  } else if (adagrad) {
     CHECK_F32(acc);
+    accw = (float*)acc.data_ptr();
+  }
+  if (adagrad) {
     int epb = 8;
     hipLaunchKernelGGL(emb_accsq_uni_kernel, dim3((unsigned)((n + epb - 1) / epb)),
                        dim3(256), 0, s,
-                       (float*)acc.data_ptr(), (const long*)rows.data_ptr(),
+                       accw, (const long*)rows.data_ptr(),
                        (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
-                       n, F, D, DP, dstride, dcol0, wstride);
-    accp = (const float*)acc.data_ptr();
+                       n, F, D, DP, dstride, dcol0, wstride, astride, aoff);
+    accp = accw;
   }
   long total = n * 32;   // 32 lanes per entry (shift-mapped)
   hipLaunchKernelGGL(emb_scatter_uni_kernel, dim3(scat_blocks(total)), dim3(256), 0, s,
                      (bf16*)arena.data_ptr(), (const long*)rows.data_ptr(),
                      (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
                      accp, n, F, D, DP, dstride, dcol0, wstride, (float)-lr,
-                     (float)eps);
+                     (float)eps, astride, aoff);
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {

@@ -51,11 +51,13 @@ class FusedOptimizer:
         # moment buffers (fp32): adam: m,v ; adadelta: accum, delta_accum ; adagrad: m
         self.m = torch.zeros(n, device=dev) if self.kind in (OPT_ADAM, OPT_ADADELTA, OPT_ADAGRAD) else None
         self.v = torch.zeros(n, device=dev) if self.kind in (OPT_ADAM, OPT_ADADELTA) else None
-        # rowwise fp32 accumulators for embedding adagrad
+        # rowwise fp32 accumulators for embedding adagrad (arenas with the
+        # D+4 unified layout carry their accumulator INSIDE the row instead)
         self.emb_state: Dict[int, torch.Tensor] = {}
         if self.emb_kind == OPT_ADAGRAD:
             for i, p in enumerate(self.emb_params):
-                self.emb_state[i] = torch.zeros(p.shape[0], device=p.device)
+                if not getattr(p, "_acc_in_arena", False):
+                    self.emb_state[i] = torch.zeros(p.shape[0], device=p.device)
 
     # ------------------------------------------------------------------ dense
     def _dense_step_ref(self) -> None:
@@ -111,18 +113,19 @@ class FusedOptimizer:
 
     # ----------------------------------------------------------------- sparse
     def _emb_step(self, p: torch.nn.Parameter, idx: int) -> None:
+        adagrad = self.emb_kind == OPT_ADAGRAD
+        in_arena = bool(getattr(p, "_acc_in_arena", False))
         stash = getattr(p, "_unified_grads", None)
         if stash:
             # deferred unified-arena grads (ops/embedding._UnifiedGatherFn
             # fast path, GPU only): consume the unpacked buffers directly
             ext = hip_ops()
-            adagrad = self.emb_kind == OPT_ADAGRAD
-            acc = (self.emb_state[idx] if adagrad
+            acc = (self.emb_state[idx] if (adagrad and not in_arena)
                    else torch.empty(0, device=p.device, dtype=torch.float32))
             for (rows, dout, nd, dwide, F, _D) in stash:
                 ext.emb_update_unified(p.data, acc, rows, dout, nd,
                                        dwide, 1, F, self.emb_lr, self.eps,
-                                       adagrad)
+                                       adagrad, in_arena)
             p._unified_grads = []
             return
         if p.grad is None:
@@ -131,7 +134,9 @@ class FusedOptimizer:
         rows, vals = sparse_rows_values(p.grad)
         if rows.numel() == 0:
             return
-        if self.emb_kind == OPT_ADAGRAD:
+        if in_arena:
+            self._emb_step_in_arena(p, rows, vals, adagrad)
+        elif adagrad:
             acc = self.emb_state[idx]
             if p.dtype == torch.bfloat16 and use_hip(p):
                 hip_ops().emb_adagrad_step(p.data, acc, rows.contiguous(),
@@ -148,6 +153,34 @@ class FusedOptimizer:
                                        self.emb_lr)
             else:
                 p.data.index_add_(0, rows, (-self.emb_lr * vals.float()).to(p.dtype))
+        p.grad = None
+
+    def _emb_step_in_arena(self, p, rows, vals, adagrad: bool) -> None:
+        """Packed [n, D+4] grads on a D+4 unified arena.  The generic kernels
+        must NOT touch the accumulator columns (a packed-bf16 add of 0 can
+        canonicalize a NaN bit pattern inside the raw f32), so the packed
+        values route through the unified kernels: dgrad=vals, dcol0=0,
+        wide=vals[:, D] (stride DP)."""
+        DP = p.shape[1]
+        D = DP - 4
+        vals = vals.contiguous()
+        if p.dtype == torch.bfloat16 and use_hip(p):
+            hip_ops().emb_update_unified(
+                p.data, torch.empty(0, device=p.device, dtype=torch.float32),
+                rows.contiguous(), vals, 0, vals[:, D], DP, 1,
+                self.emb_lr, self.eps, adagrad, True)
+        else:
+            vf = vals.float()
+            gsq = (vf[:, :D] * vf[:, :D]).sum(dim=1) + vf[:, D] * vf[:, D]
+            upd = torch.zeros_like(vf)
+            if adagrad:
+                acc = p.data[:, D + 2]          # fp32 arena: plain f32 col
+                acc.index_add_(0, rows, gsq / DP)
+                denom = acc[rows].add(self.eps).sqrt_().unsqueeze(1)
+                upd[:, :D + 1] = -self.emb_lr * vf[:, :D + 1] / denom
+            else:
+                upd[:, :D + 1] = -self.emb_lr * vf[:, :D + 1]
+            p.data.index_add_(0, rows, upd.to(p.dtype))
         p.grad = None
 
     # ------------------------------------------------------------------ api

@@ -477,7 +477,7 @@ def test_unified_gather_split_gpu():
     out, wide = emb.gather_split(ids, dense)
 
     flat = emb.flat_ids(ids).reshape(-1)
-    ref = emb.arena.detach().index_select(0, flat).reshape(B, 3, 18)
+    ref = emb.arena.detach().index_select(0, flat).reshape(B, 3, emb.cols)
     assert torch.equal(out[:, nd:].reshape(B, 3, 16), ref[:, :, :16])
     assert torch.equal(wide, ref[:, :, 16])
     assert torch.equal(out[:, :nd], dense)
@@ -555,8 +555,9 @@ def test_unified_deferred_matches_packed():
             loss.backward()
             opt.step()
             opt.zero_grad()
-        return (model.embeddings.arena.data.float().cpu(),
-                opt.emb_state[0].cpu())
+        D = model.embed_dim
+        return (model.embeddings.arena.data[:, :D + 1].float().cpu(),
+                model.embeddings.adagrad_acc().cpu())
 
     a1, s1 = run(True)
     a0, s0 = run(False)

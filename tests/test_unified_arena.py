@@ -195,9 +195,13 @@ def _unified_ep_worker(rank, port, q):
 
         m_ep = one_step(build("table"))
         m_dp = one_step(build(False))
+        # EP arenas are [.., D+2]; the replicated unified arena is [.., D+4]
+        # (in-row adagrad accumulator): compare the deep+wide columns
+        D = m_ep.embed_dim
         ok_step = torch.allclose(
-            m_ep.embeddings.arena.data,
-            m_ep.embeddings.shard_from_full(m_dp.embeddings.arena.data),
+            m_ep.embeddings.arena.data[:, :D + 1],
+            m_ep.embeddings.shard_from_full(
+                m_dp.embeddings.arena.data[:, :D + 2])[:, :D + 1],
             atol=1e-5)
         q.put((rank, bool(ok_fwd), bool(ok_step)))
         dist.barrier()
