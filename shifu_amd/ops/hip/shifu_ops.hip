@@ -1680,7 +1680,11 @@ at::Tensor gemm_tt_f32(at::Tensor dz, at::Tensor x) {
 __global__ __launch_bounds__(256)
 void transpose_bf16_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
                            int R, int C) {
-  __shared__ bf16 tile[64][72];
+  // 66-short rows = 33-dword (odd) stride: the transposed column reads then
+  // walk 33r mod 32 = distinct banks for all 64 rows (the old +8 pad made a
+  // 36-dword stride -> 8-way conflicts).  Writes go as 4-byte pairs (any
+  // wider LDS write would be misaligned at a 132-byte stride).
+  __shared__ short tile[64 * 66];
   int rt = blockIdx.y * 64, ct = blockIdx.x * 64;
   int tid = threadIdx.x;
   // load 64x64: 512 chunks of 8, 2 per thread
@@ -1696,7 +1700,9 @@ void transpose_bf16_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
       else for (int j = 0; j < 8; ++j)
         ((short*)&v)[j] = (gc + j < C) ? ((const short*)src)[j] : (short)0;
     }
-    *(s16x8*)&tile[r][g] = v;
+#pragma unroll
+    for (int j = 0; j < 8; j += 2)
+      *(unsigned*)&tile[r * 66 + g + j] = *(unsigned*)&((short*)&v)[j];
   }
   __syncthreads();
   // store 64x64 transposed: thread writes out[ct+c][rt+r..r+7]
@@ -1708,7 +1714,7 @@ void transpose_bf16_kernel(const bf16* __restrict__ in, bf16* __restrict__ out,
     if (oc >= C) continue;
     s16x8 v;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) ((short*)&v)[j] = ((short*)&tile[g + j][c])[0];
+    for (int j = 0; j < 8; ++j) ((short*)&v)[j] = tile[(g + j) * 66 + c];
     bf16* dst = out + (long)oc * R + orr;
     if (orr + 8 <= R) *(s16x8*)dst = v;
     else for (int j = 0; j < 8 && orr + j < R; ++j) ((short*)dst)[j] = ((short*)&v)[j];
@@ -2812,6 +2818,9 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
   auto db = at::empty({1}, x.options().dtype(at::kFloat));
   hipMemsetAsync(dw.data_ptr(), 0, (size_t)K * 4, cur_stream());
   hipMemsetAsync(db.data_ptr(), 0, 4, cur_stream());
+  // NOTE a finer row chop (4096 chunks) was tried and REGRESSED 8% e2e:
+  // every chunk atomically adds all K partials onto the same ~K*4 bytes, and
+  // the line contention dwarfs the latency win.  Keep the coarse grid.
   long gx, chunks, rpc;
   colsum_grid(B, K, gx, chunks, rpc);
   hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)gx, (unsigned)chunks),
