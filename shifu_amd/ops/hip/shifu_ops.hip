@@ -1257,7 +1257,10 @@ static long splitk_target_blocks_v4() {
   return t;
 }
 
-// C[i] = sum_z W[z*MN + i] — the slab-mode reduction (f32x4 grid-stride).
+// C[i] (+)= sum_z W[z*MN + i] — the slab-mode reduction (f32x4 grid-stride).
+// ACC=1 accumulates into C (wgrad written straight into the flat-grad view:
+// no separate AccumulateGrad elementwise add per layer per step).
+template <int ACC>
 __global__ __launch_bounds__(256)
 void splitk_reduce_kernel(const float* __restrict__ W, float* __restrict__ C,
                           long MN, int z) {
@@ -1268,23 +1271,29 @@ void splitk_reduce_kernel(const float* __restrict__ W, float* __restrict__ C,
     for (long i = t0; i < nv; i += nthr) {
       f32x4 s = ((const f32x4*)W)[i];
       for (int zz = 1; zz < z; ++zz) s += ((const f32x4*)(W + (long)zz * MN))[i];
+      if (ACC) s += ((const f32x4*)C)[i];
       ((f32x4*)C)[i] = s;
     }
   } else {
     for (long i = t0; i < MN; i += nthr) {
       float s = W[i];
       for (int zz = 1; zz < z; ++zz) s += W[(long)zz * MN + i];
+      if (ACC) s += C[i];
       C[i] = s;
     }
   }
 }
 
 static void launch_splitk_reduce(const float* W, float* C, long MN, long z,
-                                 hipStream_t s) {
+                                 hipStream_t s, bool accumulate = false) {
   long work = (MN & 3) == 0 ? MN >> 2 : MN;
   int blocks = (int)std::min<long>((work + 255) / 256, 2048);
-  hipLaunchKernelGGL(splitk_reduce_kernel, dim3(blocks), dim3(256), 0, s,
-                     W, C, MN, (int)z);
+  if (accumulate)
+    hipLaunchKernelGGL(splitk_reduce_kernel<1>, dim3(blocks), dim3(256), 0, s,
+                       W, C, MN, (int)z);
+  else
+    hipLaunchKernelGGL(splitk_reduce_kernel<0>, dim3(blocks), dim3(256), 0, s,
+                       W, C, MN, (int)z);
 }
 
 // Split-K accumulation mode (default slab): each z-part plain-stores its
@@ -1324,9 +1333,11 @@ static long splitk_min_kt() {
   return t;
 }
 
-// v4 split-K into c[M,N] f32 (handles its own zero/workspace; c may be uninit).
+// v4 split-K into c[M,N] f32 (handles its own zero/workspace; c may be
+// uninit unless accumulate=true, in which case c's content is ADDED to).
 static void run_nt_v4_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
-                                 long M, long N, long K, hipStream_t s, long z) {
+                                 long M, long N, long K, hipStream_t s, long z,
+                                 bool accumulate = false) {
   dim3 grid((unsigned)((N + V4_BN - 1) / V4_BN),
             (unsigned)((M + V4_BM - 1) / V4_BM), (unsigned)z);
   float* cp = (float*)c.data_ptr();
@@ -1337,9 +1348,9 @@ static void run_nt_v4_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
     auto w = at::empty({zs, M * N}, c.options());
     launch_256<EPI_F32, float, true>(grid, A, B, (float*)w.data_ptr(), nullptr,
                                      (int)M, (int)N, (int)K, 0, M * N, s);
-    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, zs, s);
+    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, zs, s, accumulate);
   } else {
-    hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
+    if (!accumulate) hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
     launch_256<EPI_F32, float, true>(grid, A, B, cp, nullptr,
                                      (int)M, (int)N, (int)K, 0, 0L, s);
   }
@@ -1347,7 +1358,8 @@ static void run_nt_v4_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
 
 // v3 split-K into c[M,N] f32 (handles its own zero/workspace; c may be uninit).
 static void run_nt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
-                              long M, long N, long K, hipStream_t s) {
+                              long M, long N, long K, hipStream_t s,
+                              bool accumulate = false) {
   long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
   long max_z = (K + NT_BK - 1) / NT_BK;
   long z = std::min<long>(
@@ -1366,7 +1378,14 @@ static void run_nt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
   if (max_z >= 256 && gx * gy * z16 >= 256) z = std::min(z, z16);
   float* cp = (float*)c.data_ptr();
   if (z <= 1) {
-    launch_nt<EPI_F32, float>(A, B, cp, nullptr, M, N, K, 0, s);
+    if (accumulate) {
+      auto tmp = at::empty({M, N}, c.options());
+      launch_nt<EPI_F32, float>(A, B, (float*)tmp.data_ptr(), nullptr,
+                                M, N, K, 0, s);
+      c.add_(tmp.view_as(c));
+    } else {
+      launch_nt<EPI_F32, float>(A, B, cp, nullptr, M, N, K, 0, s);
+    }
     return;
   }
   if (splitk_slab_mode()) {
@@ -1376,9 +1395,9 @@ static void run_nt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
                        dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
                        NT_LDS_BYTES, s, A, B, (float*)w.data_ptr(), nullptr,
                        (int)M, (int)N, (int)K, 0, M * N);
-    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, z, s);
+    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, z, s, accumulate);
   } else {
-    hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
+    if (!accumulate) hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
     hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true>),
                        dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
                        NT_LDS_BYTES, s, A, B, cp, nullptr,
@@ -1433,6 +1452,29 @@ at::Tensor gemm_ntv3_bf16(at::Tensor a, at::Tensor b) {
     launch_nt<EPI_PLAIN, bf16>((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
                                (bf16*)c.data_ptr(), nullptr, M, N, K, 0, cur_stream());
   return c;
+}
+
+// wgrad ACCUMULATED into `out` (the [M,N] f32 flat-grad view): out += a@b^T.
+// Saves the fresh-dw allocation and the per-layer AccumulateGrad add.
+void gemm_ntv3_f32_into(at::Tensor a, at::Tensor b, at::Tensor out) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  CHECK_GPU(out); CHECK_CONTIG(out); CHECK_F32(out);
+  long M = a.size(0), K = a.size(1), N = b.size(0);
+  TORCH_CHECK(b.size(1) == K && out.size(0) == M && out.size(1) == N,
+              "shape mismatch out += a@b^T");
+  long gx4 = (N + V4_BN - 1) / V4_BN, gy4 = (M + V4_BM - 1) / V4_BM;
+  long kt4 = (K + V4_BK - 1) / V4_BK;
+  long z24 = std::min<long>(
+      std::max<long>(splitk_target_blocks_v4() / std::max<long>(gx4 * gy4, 1), 1),
+      std::max<long>(kt4 / splitk_min_kt(), 1));
+  if (M >= 256 && N >= 256 && z24 > 1 && gx4 * gy4 * z24 >= 200) {
+    run_nt_v4_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                         out, M, N, K, cur_stream(), z24, /*accumulate=*/true);
+  } else {
+    run_nt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                      out, M, N, K, cur_stream(), /*accumulate=*/true);
+  }
 }
 
 // c[M,N] f32 = a[M,K] @ b[N,K]^T with split-K (wgrad)
@@ -2015,6 +2057,25 @@ std::vector<at::Tensor> act_grad_colsum_T(at::Tensor dy, at::Tensor y, long act)
   return {dz, dzT, db};
 }
 
+// same, but the colsum atomically ACCUMULATES into db (the bias flat-grad
+// view): no db zeros, no separate AccumulateGrad add
+std::vector<at::Tensor> act_grad_colsum_T_into(at::Tensor dy, at::Tensor y,
+                                               long act, at::Tensor db) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+  CHECK_GPU(db); CHECK_F32(db);
+  long B = dy.size(0), N = dy.size(1);
+  TORCH_CHECK(db.numel() == N && db.is_contiguous(), "db view mismatch");
+  auto dz = at::empty_like(dy);
+  auto dzT = at::empty({N, B}, dy.options());
+  dim3 grid((unsigned)((N + 63) / 64), (unsigned)((B + 63) / 64));
+  hipLaunchKernelGGL(act_grad_colsum_T_kernel, grid, dim3(256), 0, cur_stream(),
+                     (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                     (bf16*)dz.data_ptr(), (bf16*)dzT.data_ptr(),
+                     (float*)db.data_ptr(), B, N, (int)act);
+  return {dz, dzT};
+}
+
 // ---------------------------------------------------------------------------
 // fused sigmoid + weighted loss (K3)
 // fwd: p = sigmoid(z); per = w*(p-y)^2 (wmse) or w*bce (ce);
@@ -2073,8 +2134,9 @@ std::vector<at::Tensor> weighted_loss_fwd(at::Tensor z, at::Tensor y, at::Tensor
   CHECK_GPU(w); CHECK_CONTIG(w); CHECK_F32(w);
   long n = z.numel();
   auto p = at::empty({n}, z.options().dtype(at::kFloat));
-  auto loss_sum = at::zeros({}, z.options().dtype(at::kFloat));
-  auto wsum = at::zeros({}, z.options().dtype(at::kFloat));
+  auto sums = at::zeros({2}, z.options().dtype(at::kFloat));
+  auto loss_sum = sums.narrow(0, 0, 1).reshape({});
+  auto wsum = sums.narrow(0, 1, 1).reshape({});
   int blocks = (int)std::min((n + 511) / 512 + 1, (long)1024);
   hipLaunchKernelGGL(loss_fwd_kernel, dim3(blocks), dim3(512), 0, cur_stream(),
                      (const bf16*)z.data_ptr(), (const float*)y.data_ptr(),
@@ -2814,10 +2876,12 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
                                  bool need_dx) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
   long B = x.size(0), K = x.size(1);
-  auto dw = at::empty({1, K}, x.options().dtype(at::kFloat));
-  auto db = at::empty({1}, x.options().dtype(at::kFloat));
-  hipMemsetAsync(dw.data_ptr(), 0, (size_t)K * 4, cur_stream());
-  hipMemsetAsync(db.data_ptr(), 0, 4, cur_stream());
+  // one allocation + one memset for dw|db (small-kernel launch overhead
+  // was ~5us per fill inside the captured step)
+  auto wb = at::empty({K + 1}, x.options().dtype(at::kFloat));
+  auto dw = wb.narrow(0, 0, K).view({1, K});
+  auto db = wb.narrow(0, K, 1);
+  hipMemsetAsync(wb.data_ptr(), 0, (size_t)(K + 1) * 4, cur_stream());
   // NOTE a finer row chop (4096 chunks) was tried and REGRESSED 8% e2e:
   // every chunk atomically adds all K partials onto the same ~K*4 bytes, and
   // the line contention dwarfs the latency win.  Keep the coarse grid.
@@ -2924,6 +2988,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_nt_fwd", &linear_nt_fwd);
   m.def("gemm_ntv3_bf16", &gemm_ntv3_bf16);
   m.def("gemm_ntv3_f32", &gemm_ntv3_f32);
+  m.def("gemm_ntv3_f32_into", &gemm_ntv3_f32_into);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("gemm_tt_f32", &gemm_tt_f32);
   m.def("gemv_fwd", &gemv_fwd);
@@ -2936,6 +3001,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_f32", &colsum_f32);
   m.def("act_grad_colsum", &act_grad_colsum);
   m.def("act_grad_colsum_T", &act_grad_colsum_T);
+  m.def("act_grad_colsum_T_into", &act_grad_colsum_T_into);
   m.def("weighted_loss_fwd", &weighted_loss_fwd);
   m.def("weighted_loss_bwd", &weighted_loss_bwd);
   m.def("sgd_step", &sgd_step);

@@ -76,6 +76,13 @@ def act_id(name: str) -> int:
     return _ACT_IDS[name.lower()]
 
 
+def _dp_overlap_active() -> bool:
+    """True when multi-rank bucketed all-reduce overlap is live: wgrad must
+    then flow through AccumulateGrad so the per-bucket hooks fire."""
+    import torch.distributed as dist
+    return dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1
+
+
 def _act_fwd_ref(z: torch.Tensor, act: int) -> torch.Tensor:
     if act == ACT_SIGMOID:
         return torch.sigmoid(z)
@@ -145,6 +152,21 @@ class _FusedLinearFn(torch.autograd.Function):
             # one fused pass: dz = dy*act'(y) (row-major AND transposed)
             # + db = colsum(dz) — the dzT side feeds wgrad directly, so no
             # separate transpose of dz runs (act_grad_colsum_T kernel)
+            if (gv_w is not None and gv_b is not None and not _ASYNC_WGRAD
+                    and not _dp_overlap_active()):
+                # single-rank fast path: colsum accumulates into the bias
+                # flat-grad view and the split-K wgrad reduce accumulates
+                # into the weight view — autograd gets None for w/b (no
+                # fresh dw alloc, no AccumulateGrad adds, no db zeros)
+                dz, dzT = ext.act_grad_colsum_T_into(dy.contiguous(), y, act,
+                                                     gv_b)
+                xT = ext.transpose_bf16(x)
+                ext.gemm_ntv3_f32_into(dzT, xT, gv_w)
+                dx = None
+                if ctx.x_needs_grad:
+                    wT = ext.transpose_bf16(w)
+                    dx = ext.gemm_ntv3_bf16(dz, wT)
+                return dx, None, None, None, None, None, None, None
             dz, dzT, db = ext.act_grad_colsum_T(dy.contiguous(), y, act)
             if _ASYNC_WGRAD and gv_w is not None and gv_b is not None:
                 # wgrad on a side stream, accumulated straight into the flat

@@ -581,3 +581,34 @@ def test_act_grad_colsum_T(B, N, act):
     assert torch.equal(dzT.float(), ref_b.t().contiguous()), "dzT mismatch"
     ok, err = _rel_close(db, ref_b.sum(0), 1e-2)
     assert ok, f"db maxdiff={err}"
+
+
+def test_wgrad_into_gradview_matches_plain():
+    """The accumulate-into-flat-grad backward (act_grad_colsum_T_into +
+    gemm_ntv3_f32_into) must produce the same dense grads as the plain
+    AccumulateGrad route."""
+    from shifu_amd.models.mlp import ShifuMLP
+    from shifu_amd.ops.flat import FlatParams, split_params, bind_mirrors
+    from shifu_amd.ops.loss import weighted_loss
+
+    def grads(bind):
+        torch.manual_seed(0)
+        model = ShifuMLP(64, [128, 64], ["relu", "tanh"], seed=2).cuda()
+        dense_params, _ = split_params(model)
+        flat = FlatParams(dense_params, mirror_bf16=True)
+        if bind:
+            bind_mirrors(model, flat)   # gradviews present -> into-path
+        g = torch.Generator().manual_seed(1)
+        x = torch.randn(512, 64, generator=g).to(torch.bfloat16).cuda()
+        y = (torch.rand(512, generator=g) > 0.5).float().cuda()
+        for _ in range(2):   # accumulation across two backwards
+            loss = weighted_loss(model(x), y, torch.ones(512, device="cuda"),
+                                 "sigmoid_ce")
+            loss.backward()
+        flat.sync_grads()
+        return flat.flat_grad.cpu()
+
+    gi = grads(True)
+    gp = grads(False)
+    rel = float((gi - gp).abs().max() / gp.abs().max().clamp_min(1e-6))
+    assert rel < 5e-3, f"into-path grads diverge: rel {rel}"
