@@ -59,3 +59,21 @@ def test_bench_torchrun_2rank_dp_mode():
     d = json.loads(lines[0])
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+
+
+def test_bench_torchrun_4rank_table_ep():
+    """world=4 (gloo CPU): the table-EP feature assignment puts 1 feature on
+    each of 4 ranks for n-cat=4 — a different split shape than the 2-rank
+    tests; exactly what the driver's SCALE run exercises at N=4."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29655", "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "16", "--vocab", "500", "--n-cat", "6"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 4
+    assert "ep4" in d["config"]["parallelism"]
