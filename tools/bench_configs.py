@@ -175,13 +175,17 @@ def main():
 
     elif args.config == "4":
         from shifu_amd.models.deepfm import DeepFM
+        from shifu_amd.ops.embedding import UnifiedMultiEmbedding
         vocab = [1_000_000] * 26
         model = DeepFM(200, vocab, 64, [1024, 512, 256], ["relu"] * 3,
-                       seed=3).to(device)
+                       seed=3, unified=True).to(device)
         if on_gpu:
             for p in model.parameters():
                 if getattr(p, "_is_embedding_arena", False):
                     p.data = p.data.to(torch.bfloat16)
+        for m in model.modules():
+            if isinstance(m, UnifiedMultiEmbedding):
+                m.defer_grads = True   # single-rank unpacked-grad fast path
         b = _batches(args.batch, 200, vocab, device, dtype)
         el = _train_throughput(model, b, args.steps, args.warmup)
         _emit("deepfm", args.batch * args.steps, el, args.steps,
