@@ -53,6 +53,11 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
             s, e = shard_rows(len(train), rank, world)
             train = train.subset(np.arange(s, e))
 
+        # per-worker pos/neg counts like the reference's load_data logging
+        # (ssgd_monitor.py:448-452)
+        print(f"[rank {rank}] train rows={len(train)} pos={train.pos_count} "
+              f"neg={train.neg_count}; valid rows={len(valid)}", flush=True)
+
         device = torch.device(rc.resolved_device(),
                               rank % max(torch.cuda.device_count(), 1)
                               if rc.resolved_device() == "cuda" else 0) \
