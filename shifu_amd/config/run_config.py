@@ -94,6 +94,11 @@ class RunConfig:
                                          # too
 
     # -- misc --
+    data_cache: bool = True              # cache parsed shards under
+                                         # log_dir/shard_cache so a launcher
+                                         # restart skips the CSV re-parse
+                                         # (keyed on file sizes/mtimes +
+                                         # column spec)
     log_dir: str = "./logs"
     device: str = "auto"                 # "cuda" | "cpu" | "auto"
     enable_trace: bool = False           # per-phase step timing (utils/trace.py)

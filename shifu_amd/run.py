@@ -30,33 +30,51 @@ def default_rank_entry(rank: int, world: int, rc: RunConfig, mc: ModelConfig,
     try:
         files = list_training_files(rc.training_data_path)
         from shifu_amd.io import load_csv_native
+        from shifu_amd.data.shard_cache import load_split_cached
         import numpy as np
-        if world > 1 and len(files) >= world:
-            # reference behavior (TrainingDataSet.java:55-89): round-robin
-            # FILES across workers, each rank parses only its shard — the
-            # 100M-row config never parses the full dataset per rank.
-            # Valid split is then per-rank (seeded per rank, like the
-            # reference's per-worker random split).
-            my_files = shard_files(files, rank, world)
-            shard = load_csv_native(my_files, rc.selected_numeric_columns,
-                                    rc.selected_categorical_columns,
-                                    rc.target_column, rc.weight_column, rc.delimiter)
-            train, valid = shard.split(rc.valid_set_rate, seed=rc.seed + rank)
-        else:
+
+        spec = {"num": rc.selected_numeric_columns,
+                "cat": rc.selected_categorical_columns,
+                "target": rc.target_column, "weight": rc.weight_column,
+                "delim": rc.delimiter, "valid": rc.valid_set_rate,
+                "seed": rc.seed}
+        cache_dir = (os.path.join(rc.log_dir, "shard_cache")
+                     if getattr(rc, "data_cache", True) else None)
+
+        def build():
+            if world > 1 and len(files) >= world:
+                # reference behavior (TrainingDataSet.java:55-89): round-robin
+                # FILES across workers, each rank parses only its shard — the
+                # 100M-row config never parses the full dataset per rank.
+                # Valid split is then per-rank (seeded per rank, like the
+                # reference's per-worker random split).
+                my_files = shard_files(files, rank, world)
+                shard = load_csv_native(my_files, rc.selected_numeric_columns,
+                                        rc.selected_categorical_columns,
+                                        rc.target_column, rc.weight_column,
+                                        rc.delimiter)
+                return shard.split(rc.valid_set_rate, seed=rc.seed + rank)
             # few big files: every rank parses, then row-range shards; a
             # shared-seed global split keeps valid metrics identical across
             # ranks
             full = load_csv_native(files, rc.selected_numeric_columns,
                                    rc.selected_categorical_columns,
-                                   rc.target_column, rc.weight_column, rc.delimiter)
-            train, valid = full.split(rc.valid_set_rate, seed=rc.seed)
-            s, e = shard_rows(len(train), rank, world)
-            train = train.subset(np.arange(s, e))
+                                   rc.target_column, rc.weight_column,
+                                   rc.delimiter)
+            tr_, va_ = full.split(rc.valid_set_rate, seed=rc.seed)
+            s, e = shard_rows(len(tr_), rank, world)
+            return tr_.subset(np.arange(s, e)), va_
+
+        # restart-from-checkpoint skips the CSV re-parse: the parsed shard is
+        # cached keyed on file sizes/mtimes + column spec (data/shard_cache.py)
+        train, valid, cached = load_split_cached(cache_dir, files, spec,
+                                                 rank, world, build)
 
         # per-worker pos/neg counts like the reference's load_data logging
         # (ssgd_monitor.py:448-452)
         print(f"[rank {rank}] train rows={len(train)} pos={train.pos_count} "
-              f"neg={train.neg_count}; valid rows={len(valid)}", flush=True)
+              f"neg={train.neg_count}; valid rows={len(valid)}"
+              f"{' (shard cache)' if cached else ''}", flush=True)
 
         device = torch.device(rc.resolved_device(),
                               rank % max(torch.cuda.device_count(), 1)

@@ -90,3 +90,49 @@ def test_shard_rows_cover():
     assert covered == 103
     sizes = [e - s for s, e in spans]
     assert max(sizes) - min(sizes) <= 1
+
+
+def test_shard_cache_roundtrip_and_invalidation(tmp_path):
+    """data/shard_cache.py: second load comes from cache with identical
+    arrays; touching an input invalidates the entry."""
+    import os
+    import time
+    import numpy as np
+    from shifu_amd.data.synthetic import generate_synthetic_csv
+    from shifu_amd.data.csv_loader import load_csv_files, list_training_files
+    from shifu_amd.data.shard_cache import load_split_cached
+
+    data_dir = tmp_path / "data"
+    generate_synthetic_csv(str(data_dir), 300, 5, (11,), seed=3, n_files=1)
+    files = list_training_files([str(data_dir)])
+    spec = {"num": [2, 3, 4, 5, 6], "cat": [7], "target": 0, "weight": 1,
+            "valid": 0.2, "seed": 1}
+
+    calls = {"n": 0}
+
+    def build():
+        calls["n"] += 1
+        full = load_csv_files(files, [2, 3, 4, 5, 6], [7], 0, 1, "|")
+        return full.split(0.2, seed=1)
+
+    cache = str(tmp_path / "cache")
+    t1, v1, c1 = load_split_cached(cache, files, spec, 0, 1, build)
+    t2, v2, c2 = load_split_cached(cache, files, spec, 0, 1, build)
+    assert not c1 and c2 and calls["n"] == 1
+    assert np.array_equal(t1.dense, t2.dense)
+    assert np.array_equal(t1.target, t2.target)
+    assert np.array_equal(v1.dense, v2.dense)
+
+    # different rank -> different entry
+    _, _, c3 = load_split_cached(cache, files, spec, 1, 2, build)
+    assert not c3 and calls["n"] == 2
+
+    # touching the input invalidates
+    time.sleep(0.01)
+    os.utime(files[0])
+    _, _, c4 = load_split_cached(cache, files, spec, 0, 1, build)
+    assert not c4 and calls["n"] == 3
+
+    # cache disabled
+    _, _, c5 = load_split_cached(None, files, spec, 0, 1, build)
+    assert not c5 and calls["n"] == 4
