@@ -101,7 +101,8 @@ def load_csv_files(
                     continue
                 parts = line.split(delimiter)
                 try:
-                    t = float(parts[target_column])
+                    # target_column < 0: no-target (scoring-only) layout
+                    t = float(parts[target_column]) if target_column >= 0 else 0.0
                 except (ValueError, IndexError):
                     continue  # skip header-ish / malformed rows
                 if weight_column >= 0:

@@ -153,9 +153,12 @@ void parse_file(const std::string& path, const ColPlan& plan, FileOut* out) {
       }
     }
     int nf = (int)starts.size();
-    if (plan.target_col >= nf) continue;
-    float tgt;
-    if (!parse_float(starts[plan.target_col], ends[plan.target_col], tgt)) continue;
+    float tgt = 0.0f;
+    if (plan.target_col >= 0) {   // target_col < 0: no-target layout
+                                  // (scoring-only datasets) — tgt stays 0
+      if (plan.target_col >= nf) continue;
+      if (!parse_float(starts[plan.target_col], ends[plan.target_col], tgt)) continue;
+    }
     float wgt = 1.0f;
     if (plan.weight_col >= 0 && plan.weight_col < nf) {
       float w;

@@ -49,10 +49,12 @@ def score_files(model_dir: str, data_paths: List[str],
         ds = load_csv_native(data_paths, num_cols, cat_cols, tgt, wcol, delimiter)
         targets = ds.target
     else:
-        # raw layout: nd floats then nc ids, no target/weight
+        # raw layout: nd floats then nc ids, no target/weight column at all
+        # (target_column=-1: the loader fills targets with zeros, which we
+        # discard — column 0 stays a plain dense feature)
         ds = load_csv_native(data_paths, list(range(nd)),
                              list(range(nd, nd + nc)),
-                             target_column=0, weight_column=-1,
+                             target_column=-1, weight_column=-1,
                              delimiter=delimiter)
         targets = None
 

@@ -86,3 +86,20 @@ def test_python_fallback_when_ext_missing(tmp_path, monkeypatch):
     np.testing.assert_array_equal(fallback.cats, native.cats)
     np.testing.assert_allclose(fallback.target, native.target, atol=1e-6)
     np.testing.assert_allclose(fallback.weight, native.weight, atol=1e-6)
+
+
+def test_no_target_layout(tmp_path):
+    """target_column=-1 (scoring-only datasets): rows parse with zero
+    targets, column 0 stays a plain feature — native and python loaders
+    agree (the round-1 score.py raw layout reused column 0 as target)."""
+    import numpy as np
+    from shifu_amd.data.csv_loader import load_csv_files
+    from shifu_amd.io import load_csv_native
+    p = tmp_path / "raw.csv"
+    p.write_text("0.5|1.5|7\n-0.25|2.5|3\n")
+    for loader in (load_csv_files, load_csv_native):
+        ds = loader([str(p)], [0, 1], [2], -1, -1, "|")
+        assert len(ds) == 2
+        assert np.allclose(ds.dense[:, 0], [0.5, -0.25])
+        assert np.all(ds.target == 0.0)
+        assert np.all(ds.weight == 1.0)
