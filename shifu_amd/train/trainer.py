@@ -259,6 +259,12 @@ class Trainer:
         # nonzero-weight count and their gradients are identically zero.
         self._uniform_steps = None
         if world_size > 1 and is_distributed():
+            if len(self.train_data) == 0:
+                # an empty TRAIN shard cannot pad batches (no row to repeat);
+                # it means the sharding itself is wrong — fail loudly
+                raise ValueError(
+                    f"rank {rank}: empty train shard at world={world_size} — "
+                    "check data sharding (fewer rows than ranks?)")
             import math as _math
             t = torch.tensor([_math.ceil(len(self.train_data) / self.batch_size)],
                              device=self.device if self.device.type == "cuda"
