@@ -235,3 +235,16 @@ def test_rccl_op_smoke_world1():
         assert int(t) == 7
     finally:
         dist.destroy_process_group()
+
+
+def test_hipgraph_captures_rccl_collectives():
+    """hipGraph capture of a full table-EP training step INCLUDING the RCCL
+    collectives (all_to_all_single + bucketed all_reduce) at world=1 — the
+    mechanism proof for enabling graphs at world>1 (tools/graph_rccl_test)."""
+    import subprocess
+    import sys
+    out = subprocess.run([sys.executable, "tools/graph_rccl_test.py"],
+                         capture_output=True, text=True, timeout=300,
+                         cwd=os.path.dirname(os.path.dirname(
+                             os.path.abspath(__file__))))
+    assert "GRAPH_RCCL_OK" in out.stdout, out.stderr[-1500:]
