@@ -98,6 +98,10 @@ class Launcher:
     def run(self) -> List[EpochStats]:
         attempts = 0
         while True:
+            # children inherit this (spawn copies the env): checkpoint saves
+            # stamp it so resume can DETECT shard files mixed across a
+            # crash+restart generation instead of silently loading them
+            os.environ["SHIFU_RUN_ATTEMPT"] = str(attempts)
             ok, err = self._run_once()
             if ok:
                 return self.stats

@@ -563,6 +563,12 @@ class Trainer:
         under the same epoch tag (step-cadence mid-epoch saves)."""
         gs = self.global_step if global_step is None else global_step
         rng = self._rng.bit_generator.state if np_rng is None else np_rng
+        if stamp is None:
+            # (step, launcher attempt): equal across ranks of ONE save, and
+            # different across a crash+restart retraining the same epoch —
+            # load_checkpoint then detects mixed shard generations
+            stamp = (self.global_step << 8) | (
+                int(os.environ.get("SHIFU_RUN_ATTEMPT", "0")) & 0xFF)
         ckpt.save_checkpoint(self.rc.tmp_model_path, epoch, gs,
                              self.model, self.optimizer,
                              extra={"np_rng": rng},
