@@ -98,3 +98,62 @@ def test_heartbeat_timeout_after_start(tmp_path):
     with pytest.raises(RuntimeError, match="missed heartbeats"):
         Launcher(rc, ModelConfig(), _hb_then_hang_entry).run()
     assert time.time() - t0 < 25.0   # detected via budget, not the 60s sleep
+
+
+def _ep_crash_once_entry(rank, world, rc, mc, sink, heartbeat):
+    """Real training entry: first attempt trains 1 epoch (writing EP shard
+    checkpoints) then rank 1 dies; the restart resumes from the shards and
+    finishes the full run."""
+    import dataclasses
+    from shifu_amd.run import default_rank_entry
+    marker = os.path.join(rc.log_dir, "crashed_once")
+    first = not os.path.exists(marker)
+    rc_eff = dataclasses.replace(rc, epochs=1) if first else rc
+    default_rank_entry(rank, world, rc_eff, mc, sink, heartbeat)
+    if first and rank == 1:
+        with open(marker, "w") as f:
+            f.write("x")
+        raise RuntimeError("injected EP rank failure")
+
+
+def test_launcher_ep_restart_resumes_from_shards(tmp_path):
+    """Whole recovery path end-to-end at world=2 (gloo): EP table-sharded
+    unified arenas -> per-rank shard checkpoints -> injected rank death ->
+    launcher restart -> resume from the newest complete epoch (shard cache
+    skips the CSV re-parse) -> full run completes and exports."""
+    from shifu_amd.data.synthetic import generate_synthetic_csv
+    from shifu_amd.config.model_config import ModelConfig
+
+    data_dir = tmp_path / "data"
+    generate_synthetic_csv(str(data_dir), 600, 4, (23, 31), seed=9, n_files=4)
+    mc = ModelConfig.from_dict({
+        "train": {"numTrainEpochs": 3, "validSetRate": 0.2,
+                  "params": {"NumHiddenLayers": 1, "NumHiddenNodes": [8],
+                             "ActivationFunc": ["relu"], "LearningRate": 0.05,
+                             "Optimizer": "sgd", "Loss": "sigmoid_ce",
+                             "MiniBatchSize": 32, "L2Reg": 0.0}}})
+    rc = RunConfig(num_gpus=2, log_dir=str(tmp_path / "logs"),
+                   training_data_path=[str(data_dir)],
+                   selected_numeric_columns=[2, 3, 4, 5],
+                   selected_categorical_columns=[6, 7],
+                   vocab_sizes=[23, 31], target_column=0, weight_column=1,
+                   tmp_model_path=str(tmp_path / "ckpt"),
+                   final_model_path=str(tmp_path / "final"),
+                   model_type="wide_deep", embed_dim=4, device="cpu",
+                   batch_size=32, max_rank_restarts=2,
+                   master_port=29763)
+    la = Launcher(rc, mc, _ep_crash_once_entry)
+    stats = la.run()
+    assert len(stats) >= 2, f"expected resumed epochs, got {len(stats)}"
+    # per-rank EP shard checkpoints were written
+    names = os.listdir(tmp_path / "ckpt")
+    assert any("shard0of2" in n for n in names), names
+    assert any("shard1of2" in n for n in names), names
+    # export happened (chief, after the resumed run)
+    assert os.path.exists(tmp_path / "final" / "GenericModelConfig.json")
+    # the restart went through the recovery path
+    board = (tmp_path / "logs" / "progress.board").read_text()
+    assert "restarting from" in board
+    # shard cache was populated (the restart's loads skip the re-parse)
+    cache = tmp_path / "logs" / "shard_cache"
+    assert cache.is_dir() and len(os.listdir(cache)) >= 2
