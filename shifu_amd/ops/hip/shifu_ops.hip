@@ -386,11 +386,49 @@ DEVINL void nt_stage(const bf16* __restrict__ P, bf16* lds, int r0, int k0,
   }
 }
 
+// TRANSPOSED staging for the wgrad route: P is stored [K, R] (reduction-major
+// ROWS — dz [B,M] / x [B,N] exactly as forward/act-grad produce them) and we
+// build the SAME swizzled [128 x 64] image nt_stage does, so the MFMA loop's
+// b128 fragment reads are untouched.  Global side: s16x8 along the contiguous
+// R direction; a wave's 64 lanes read 64 consecutive K-rows at one 16 B
+// column window, and the 8 windows sharing each 128 B line are issued by the
+// same wave's other iterations (L1-resident after the first).  LDS side: b16
+// scatter writes; bank = 4*((k>>3)^(r&7)) + ((k&7)>>1) depends only on (k,
+// r&7), and each write instruction has r&7 uniform with 64 distinct k across
+// the wave -> exactly 2-way conflicts (the two k sharing a dword).
+// Alignment contract: R % 8 == 0 (callers guarantee; python falls back to
+// the transpose route otherwise).
+DEVINL void nt_stage_t(const bf16* __restrict__ P, bf16* lds, int r0, int k0,
+                       int R, int K, int tid) {
+#pragma unroll
+  for (int it = 0; it < 4; ++it) {
+    int i = tid + it * 256;
+    int k = i & 63;
+    int mc = (i >> 6) * 8;
+    int gk = k0 + k, gm = r0 + mc;
+    s16x8 v = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (gk < K) {
+      const bf16* src = P + (long)gk * R + gm;
+      if (gm + 8 <= R) v = *(const s16x8*)(src);
+      else for (int j = 0; j < 8; ++j)
+        ((short*)&v)[j] = (gm + j < R) ? ((const short*)src)[j] : (short)0;
+    }
+    const int kh = k >> 3, kl = k & 7;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int r = mc + j;
+      ((short*)(lds + r * NT_BK + ((kh ^ (r & 7)) << 3) + kl))[0] = ((short*)&v)[j];
+    }
+  }
+}
+
 // slab != 0 (split-K only): instead of atomicAdd into C[M,N], each z-part
 // plain-stores its partial into C + blockIdx.z*slab (slab = M*N) and a
 // separate splitk_reduce_kernel sums the parts — no f32 atomics, bitwise
 // deterministic.  The host guarantees every z-part is non-empty in slab mode.
-template <int EPI, typename OUT_T, bool SPLITK = false>
+// TSTAGE=1: operands stored [K, M] / [K, N] (batch-major activations), staged
+// by nt_stage_t — the transpose-free wgrad path.
+template <int EPI, typename OUT_T, bool SPLITK = false, int TSTAGE = 0>
 __global__ __launch_bounds__(256)
 void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
                     OUT_T* __restrict__ C, const bf16* __restrict__ bias,
@@ -427,8 +465,13 @@ void gemm_nt_kernel(const bf16* __restrict__ A, const bf16* __restrict__ B,
   }
 
   for (int k0 = k_lo; k0 < k_hi; k0 += NT_BK) {
-    nt_stage(A, As, m0, k0, M, K, tid);
-    nt_stage(B, Bs, n0, k0, N, K, tid);
+    if (TSTAGE) {
+      nt_stage_t(A, As, m0, k0, M, K, tid);
+      nt_stage_t(B, Bs, n0, k0, N, K, tid);
+    } else {
+      nt_stage(A, As, m0, k0, M, K, tid);
+      nt_stage(B, Bs, n0, k0, N, K, tid);
+    }
     __syncthreads();   // hipcc emits the vmcnt(0) drain for in-flight glds here
 
 #pragma unroll
@@ -1506,6 +1549,89 @@ at::Tensor gemm_ntv3_f32(at::Tensor a, at::Tensor b) {
 }
 
 // ---------------------------------------------------------------------------
+// TT wgrad on the v3 skeleton: C[M,N] f32 = sum_b A[b,m] * B[b,n] with NO
+// pre-transposes — nt_stage_t scatter-builds the identical swizzled image,
+// the MFMA loop and the deterministic slab split-K are shared with v3.
+// Replaces transpose_bf16(x) + gemm_ntv3_f32(dzT, xT): 1/3 the HBM traffic
+// on the x operand (no transpose read+write) and two fewer launches per
+// layer per step.
+// ---------------------------------------------------------------------------
+static void run_tt_splitk_f32(const bf16* A, const bf16* B, at::Tensor& c,
+                              long M, long N, long K, hipStream_t s,
+                              bool accumulate = false) {
+  long gx = (N + NT_BN - 1) / NT_BN, gy = (M + NT_BM - 1) / NT_BM;
+  long max_z = (K + NT_BK - 1) / NT_BK;
+  long z = std::min<long>(
+      std::max<long>(splitk_target_blocks() / std::max<long>(gx * gy, 1), 1), max_z);
+  static long v3minkt = [] {
+    const char* e = getenv("SHIFU_SPLITK_V3MINKT");
+    return e ? atol(e) : 16L;
+  }();
+  long z16 = std::max<long>(max_z / std::max(v3minkt, 1L), 1);
+  if (max_z >= 256 && gx * gy * z16 >= 256) z = std::min(z, z16);
+  float* cp = (float*)c.data_ptr();
+  if (z <= 1) {
+    dim3 grid((unsigned)gx, (unsigned)gy);
+    if (accumulate) {
+      auto tmp = at::empty({M, N}, c.options());
+      hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, false, 1>), grid,
+                         dim3(256), NT_LDS_BYTES, s, A, B,
+                         (float*)tmp.data_ptr(), nullptr,
+                         (int)M, (int)N, (int)K, 0, 0L);
+      c.add_(tmp.view_as(c));
+    } else {
+      hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, false, 1>), grid,
+                         dim3(256), NT_LDS_BYTES, s, A, B, cp, nullptr,
+                         (int)M, (int)N, (int)K, 0, 0L);
+    }
+    return;
+  }
+  if (splitk_slab_mode()) {
+    z = splitk_no_empty_z(max_z, z);
+    auto w = at::empty({z, M * N}, c.options());
+    hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true, 1>),
+                       dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
+                       NT_LDS_BYTES, s, A, B, (float*)w.data_ptr(), nullptr,
+                       (int)M, (int)N, (int)K, 0, M * N);
+    launch_splitk_reduce((const float*)w.data_ptr(), cp, M * N, z, s, accumulate);
+  } else {
+    if (!accumulate) hipMemsetAsync(cp, 0, (size_t)M * N * 4, s);
+    hipLaunchKernelGGL((gemm_nt_kernel<EPI_F32, float, true, 1>),
+                       dim3((unsigned)gx, (unsigned)gy, (unsigned)z), dim3(256),
+                       NT_LDS_BYTES, s, A, B, cp, nullptr,
+                       (int)M, (int)N, (int)K, 0, 0L);
+  }
+}
+
+// out[M,N] += a[R,M]^T @ b[R,N] — transpose-free wgrad straight into the
+// flat-grad view.  Requires M % 8 == 0 && N % 8 == 0 (16 B row alignment for
+// the staging loads; linear.py falls back to the transpose route otherwise).
+void gemm_ttv3_f32_into(at::Tensor a, at::Tensor b, at::Tensor out) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  CHECK_GPU(out); CHECK_CONTIG(out); CHECK_F32(out);
+  long R = a.size(0), M = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == R && out.size(0) == M && out.size(1) == N,
+              "shape mismatch out += a^T@b");
+  TORCH_CHECK((M & 7) == 0 && (N & 7) == 0, "ttv3 needs M,N %% 8 == 0");
+  run_tt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                    out, M, N, R, cur_stream(), /*accumulate=*/true);
+}
+
+// c[M,N] f32 = a[R,M]^T @ b[R,N] (transpose-free wgrad, fresh output)
+at::Tensor gemm_ttv3_f32(at::Tensor a, at::Tensor b) {
+  CHECK_GPU(a); CHECK_CONTIG(a); CHECK_BF16(a);
+  CHECK_GPU(b); CHECK_CONTIG(b); CHECK_BF16(b);
+  long R = a.size(0), M = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == R, "shape mismatch a^T@b");
+  TORCH_CHECK((M & 7) == 0 && (N & 7) == 0, "ttv3 needs M,N %% 8 == 0");
+  auto c = at::empty({M, N}, a.options().dtype(at::kFloat));
+  run_tt_splitk_f32((const bf16*)a.data_ptr(), (const bf16*)b.data_ptr(),
+                    c, M, N, R, cur_stream());
+  return c;
+}
+
+// ---------------------------------------------------------------------------
 // v5 "TT" wgrad GEMM — C[M,N] f32 = sum_b dz[b,m] * x[b,n], both operands
 // stored reduction-major-ROWS ([B, M] / [B, N]) exactly as the activations
 // come out of forward/act-grad: NO pre-transposes.  The column reads MFMA
@@ -1916,7 +2042,8 @@ at::Tensor colsum_f32(at::Tensor dz) {
 __global__ void act_grad_colsum_kernel(const bf16* __restrict__ dy,
                                        const bf16* __restrict__ y,
                                        bf16* __restrict__ dz, float* __restrict__ db,
-                                       long B, long N, long rows_per_chunk, int act) {
+                                       long B, long N, long rows_per_chunk, int act,
+                                       int accum) {
   long n = (long)blockIdx.x * blockDim.x + threadIdx.x;
   if (n >= N) return;
   long b0 = (long)blockIdx.y * rows_per_chunk;
@@ -1928,7 +2055,8 @@ __global__ void act_grad_colsum_kernel(const bf16* __restrict__ dy,
     dz[i] = __float2bfloat16(g);
     acc += g;
   }
-  if (gridDim.y == 1) db[n] = acc;
+  if (!accum && gridDim.y == 1) db[n] = acc;
+  else if (accum && gridDim.y == 1) db[n] += acc;
   else atomicAdd(&db[n], acc);
 }
 
@@ -1944,8 +2072,29 @@ std::vector<at::Tensor> act_grad_colsum(at::Tensor dy, at::Tensor y, long act) {
                      dim3(256), 0, cur_stream(),
                      (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
                      (bf16*)dz.data_ptr(), (float*)db.data_ptr(),
-                     B, N, rpc, (int)act);
+                     B, N, rpc, (int)act, 0);
   return {dz, db};
+}
+
+// act-grad with the colsum ACCUMULATED into db (the bias flat-grad view):
+// the dzT-free partner of the ttv3 wgrad route — no transposed copy of dz
+// is materialized at all (gemm_ttv3_f32_into reads dz batch-major).
+at::Tensor act_grad_colsum_into(at::Tensor dy, at::Tensor y, long act,
+                                at::Tensor db) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_BF16(dy);
+  CHECK_GPU(y); CHECK_CONTIG(y); CHECK_BF16(y);
+  CHECK_GPU(db); CHECK_F32(db);
+  long B = dy.size(0), N = dy.size(1);
+  TORCH_CHECK(db.numel() == N && db.is_contiguous(), "db view mismatch");
+  auto dz = at::empty_like(dy);
+  long gx, chunks, rpc;
+  colsum_grid(B, N, gx, chunks, rpc);
+  hipLaunchKernelGGL(act_grad_colsum_kernel, dim3((unsigned)gx, (unsigned)chunks),
+                     dim3(256), 0, cur_stream(),
+                     (const bf16*)dy.data_ptr(), (const bf16*)y.data_ptr(),
+                     (bf16*)dz.data_ptr(), (float*)db.data_ptr(),
+                     B, N, rpc, (int)act, 1);
+  return dz;
 }
 
 // act-grad + colsum + TRANSPOSE in one pass: dz = dy*act'(y) written both
@@ -2991,6 +3140,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_ntv3_f32_into", &gemm_ntv3_f32_into);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("gemm_tt_f32", &gemm_tt_f32);
+  m.def("gemm_ttv3_f32", &gemm_ttv3_f32);
+  m.def("gemm_ttv3_f32_into", &gemm_ttv3_f32_into);
   m.def("gemv_fwd", &gemv_fwd);
   m.def("gemv_bwd", &gemv_bwd);
   m.def("fm2_fwd", &fm2_fwd);
@@ -3000,6 +3151,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_grad", &act_grad);
   m.def("colsum_f32", &colsum_f32);
   m.def("act_grad_colsum", &act_grad_colsum);
+  m.def("act_grad_colsum_into", &act_grad_colsum_into);
   m.def("act_grad_colsum_T", &act_grad_colsum_T);
   m.def("act_grad_colsum_T_into", &act_grad_colsum_T_into);
   m.def("weighted_loss_fwd", &weighted_loss_fwd);

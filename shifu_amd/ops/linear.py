@@ -45,6 +45,14 @@ from shifu_amd.ops.dispatch import use_hip, hip_ops
 # with it off, per-bucket all-reduce/backward overlap via the accumulate-grad
 # hooks applies on multi-GPU.  Opt in with SHIFU_ASYNC_WGRAD=1.
 _ASYNC_WGRAD = os.environ.get("SHIFU_ASYNC_WGRAD", "0") == "1"
+
+# Transpose-free wgrad (default): dw += dz^T @ x via the ttv3 kernel, which
+# scatter-stages the batch-major operands into the same swizzled LDS image
+# the NT kernels use — no xT/dzT copies, no transpose launches, 1/3 the HBM
+# traffic on the activation operand.  Requires both weight dims % 8 (16 B row
+# alignment of the staging loads); other shapes fall back to the transpose
+# route.  SHIFU_WGRAD_TT=0 restores transpose+NT everywhere.
+_WGRAD_TT = os.environ.get("SHIFU_WGRAD_TT", "1") == "1"
 _WGRAD_STREAM = None
 _WGRAD_EVENTS: list = []
 
@@ -149,25 +157,36 @@ class _FusedLinearFn(torch.autograd.Function):
                 if not ctx.x_needs_grad:
                     dx = None
                 return dx, dw, db, None, None, None, None, None
-            # one fused pass: dz = dy*act'(y) (row-major AND transposed)
-            # + db = colsum(dz) — the dzT side feeds wgrad directly, so no
-            # separate transpose of dz runs (act_grad_colsum_T kernel)
+            # transpose-free wgrad when both weight dims are 16 B-aligned:
+            # dw comes from the ttv3 kernel on the batch-major dz/x directly
+            tt = _WGRAD_TT and w.shape[0] % 8 == 0 and w.shape[1] % 8 == 0
+            # one fused pass: dz = dy*act'(y) (row-major AND, on the
+            # transpose route, transposed) + db = colsum(dz)
             if (gv_w is not None and gv_b is not None and not _ASYNC_WGRAD
                     and not _dp_overlap_active()):
                 # single-rank fast path: colsum accumulates into the bias
                 # flat-grad view and the split-K wgrad reduce accumulates
                 # into the weight view — autograd gets None for w/b (no
                 # fresh dw alloc, no AccumulateGrad adds, no db zeros)
-                dz, dzT = ext.act_grad_colsum_T_into(dy.contiguous(), y, act,
-                                                     gv_b)
-                xT = ext.transpose_bf16(x)
-                ext.gemm_ntv3_f32_into(dzT, xT, gv_w)
+                if tt:
+                    dz = ext.act_grad_colsum_into(dy.contiguous(), y, act,
+                                                  gv_b)
+                    ext.gemm_ttv3_f32_into(dz, x, gv_w)
+                else:
+                    dz, dzT = ext.act_grad_colsum_T_into(dy.contiguous(), y,
+                                                         act, gv_b)
+                    xT = ext.transpose_bf16(x)
+                    ext.gemm_ntv3_f32_into(dzT, xT, gv_w)
                 dx = None
                 if ctx.x_needs_grad:
                     wT = ext.transpose_bf16(w)
                     dx = ext.gemm_ntv3_bf16(dz, wT)
                 return dx, None, None, None, None, None, None, None
-            dz, dzT, db = ext.act_grad_colsum_T(dy.contiguous(), y, act)
+            if tt:
+                dz, db = ext.act_grad_colsum(dy.contiguous(), y, act)
+                dzT = None
+            else:
+                dz, dzT, db = ext.act_grad_colsum_T(dy.contiguous(), y, act)
             if _ASYNC_WGRAD and gv_w is not None and gv_b is not None:
                 # wgrad on a side stream, accumulated straight into the flat
                 # arena; autograd gets None (no AccumulateGrad for w/b)
@@ -177,14 +196,17 @@ class _FusedLinearFn(torch.autograd.Function):
                 ws = _wgrad_stream()
                 ws.wait_event(ev)
                 with torch.cuda.stream(ws):
-                    xT = ext.transpose_bf16(x)
-                    dwv = ext.gemm_ntv3_f32(dzT, xT)
+                    if tt:
+                        dwv = ext.gemm_ttv3_f32(dz, x)
+                    else:
+                        xT = ext.transpose_bf16(x)
+                        dwv = ext.gemm_ntv3_f32(dzT, xT)
                     gv_w.add_(dwv)
                     gv_b.add_(db)
                     done = torch.cuda.Event()
                     done.record(ws)
                 # caching-allocator safety: main-stream tensors used on ws
-                for t in (x, dzT, db):
+                for t in (x, dz if tt else dzT, db):
                     t.record_stream(ws)
                 _WGRAD_EVENTS.append(done)
                 dx = None
@@ -192,8 +214,11 @@ class _FusedLinearFn(torch.autograd.Function):
                     wT = ext.transpose_bf16(w)
                     dx = ext.gemm_ntv3_bf16(dz, wT)
                 return dx, None, None, None, None, None, None, None
-            xT = ext.transpose_bf16(x)                # [K,B]
-            dw = ext.gemm_ntv3_f32(dzT, xT)           # fp32 [N,K], split-K
+            if tt:
+                dw = ext.gemm_ttv3_f32(dz, x)         # fp32 [N,K], split-K
+            else:
+                xT = ext.transpose_bf16(x)            # [K,B]
+                dw = ext.gemm_ntv3_f32(dzT, xT)       # fp32 [N,K], split-K
             dx = None
             if ctx.x_needs_grad:
                 wT = ext.transpose_bf16(w)            # [K,N]

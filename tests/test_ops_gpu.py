@@ -186,6 +186,44 @@ def test_gemm_tt_wgrad(R, M, N):
     assert ok, f"gemm_tt {R}x{M}x{N} maxdiff={err}"
 
 
+@pytest.mark.parametrize("R,M,N", [(64, 128, 128), (512, 256, 384),
+                                   (16384, 1024, 1864), (200, 104, 136),
+                                   (1000, 40, 64), (32768, 512, 464)])
+def test_gemm_ttv3_wgrad(R, M, N):
+    """ttv3 transpose-free wgrad (scatter-staged v3 image): dw = dz^T @ x,
+    fresh-output and accumulate-into variants (deterministic slab split-K)."""
+    dz = _rand_bf16(R, M, seed=R + 31, scale=0.5)
+    x = _rand_bf16(R, N, seed=N + 32, scale=0.5)
+    want = dz.float().t() @ x.float()
+    c = hip_ops().gemm_ttv3_f32(dz, x)
+    ok, err = _rel_close(c, want)
+    assert ok, f"gemm_ttv3 {R}x{M}x{N} maxdiff={err}"
+    # into-variant accumulates
+    base = torch.randn(M, N, device="cuda") * 0.1
+    out = base.clone()
+    hip_ops().gemm_ttv3_f32_into(dz, x, out)
+    ok, err = _rel_close(out, base + want)
+    assert ok, f"gemm_ttv3_into {R}x{M}x{N} maxdiff={err}"
+    # bitwise deterministic across runs (slab split-K, no f32 atomics)
+    c2 = hip_ops().gemm_ttv3_f32(dz, x)
+    assert torch.equal(c, c2)
+
+
+def test_act_grad_colsum_into():
+    """dzT-free partner of ttv3: dz = dy*act'(y), db ACCUMULATED into view."""
+    torch.manual_seed(5)
+    dy = _rand_bf16(700, 96, seed=51)
+    y = torch.rand(700, 96, device="cuda").to(torch.bfloat16)  # in (0,1)
+    base = torch.randn(96, device="cuda")
+    db = base.clone()
+    dz = hip_ops().act_grad_colsum_into(dy, y, 1, db)  # sigmoid
+    want_dz = dy.float() * y.float() * (1 - y.float())
+    ok, err = _rel_close(dz, want_dz)
+    assert ok, f"act_grad_colsum_into dz maxdiff={err}"
+    ok, err = _rel_close(db, base + want_dz.sum(0))
+    assert ok, f"act_grad_colsum_into db maxdiff={err}"
+
+
 def test_gemm_nt():
     dz = _rand_bf16(320, 96, seed=8)
     w = _rand_bf16(130, 96, seed=9)

@@ -22,8 +22,13 @@ SHAPES = [
     ("l1_wgrd3", 1024, 1864, 8192, "v3f"),
     ("l1_wgrdT", 1024, 1864, 8192, "tt"),
     ("l1wgT16k", 1024, 1864, 16384, "tt"),
+    ("l1_wgr3T", 1024, 1864, 8192, "tt3"),
     ("l1wg32k", 1024, 1864, 32768, "v3f"),
+    ("l1wg32kT", 1024, 1864, 32768, "tt3"),
     ("l2wg32k", 512, 1024, 32768, "v3f"),
+    ("l2wg32kT", 512, 1024, 32768, "tt3"),
+    ("l3wg32kT", 256, 512, 32768, "tt3"),
+    ("l3wg32k", 256, 512, 32768, "v3f"),
     ("l1f16k", 16384, 1024, 1864, "v3"),
     ("l1d16k", 16384, 1864, 1024, "v3"),
     ("sq4096", 4096, 4096, 4096, "nn"),
@@ -79,6 +84,12 @@ def main():
             ours = lambda: ext.gemm_tt_f32(a0, b0)
             ref = lambda: a0.t().float() @ b0.float()
             a, b = ext.transpose_bf16(a0), ext.transpose_bf16(b0)
+        elif kind == "tt3":  # ttv3 transpose-free wgrad (scatter-staged v3)
+            a0 = torch.randn(K, M, device="cuda").to(torch.bfloat16)
+            b0 = torch.randn(K, N, device="cuda").to(torch.bfloat16)
+            ours = lambda: ext.gemm_ttv3_f32(a0, b0)
+            ref = lambda: a0.t().float() @ b0.float()
+            a, b = ext.transpose_bf16(a0), ext.transpose_bf16(b0)
         else:  # v3f: wgrad incl. the two activation transposes
             a0 = torch.randn(K, M, device="cuda").to(torch.bfloat16)  # dz [B,N']
             b0 = torch.randn(K, N, device="cuda").to(torch.bfloat16)  # x [B,K']
@@ -88,7 +99,7 @@ def main():
 
         c1 = ours().float()
         c2 = (a.float() @ b.float()) if kind == "nn" else \
-             (a.float() @ b.float().t()) if kind in ("nt", "v3", "v3f", "tt") else \
+             (a.float() @ b.float().t()) if kind in ("nt", "v3", "v3f", "tt", "tt3") else \
              (a.float().t() @ b.float())
         rel = float((c1 - c2).abs().max() / c2.abs().max().clamp_min(1e-3))
 
