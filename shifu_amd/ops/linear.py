@@ -46,13 +46,15 @@ from shifu_amd.ops.dispatch import use_hip, hip_ops
 # hooks applies on multi-GPU.  Opt in with SHIFU_ASYNC_WGRAD=1.
 _ASYNC_WGRAD = os.environ.get("SHIFU_ASYNC_WGRAD", "0") == "1"
 
-# Transpose-free wgrad (default): dw += dz^T @ x via the ttv3 kernel, which
-# scatter-stages the batch-major operands into the same swizzled LDS image
-# the NT kernels use — no xT/dzT copies, no transpose launches, 1/3 the HBM
-# traffic on the activation operand.  Requires both weight dims % 8 (16 B row
-# alignment of the staging loads); other shapes fall back to the transpose
-# route.  SHIFU_WGRAD_TT=0 restores transpose+NT everywhere.
-_WGRAD_TT = os.environ.get("SHIFU_WGRAD_TT", "1") == "1"
+# Transpose-free wgrad (ttv3): dw += dz^T @ x with the batch-major operands
+# scatter-staged into the same swizzled LDS image the NT kernels use — no
+# xT/dzT copies or transpose launches.  MEASURED SLOWER than the transpose
+# route and kept OFF: 116 vs 108 us at [1024,1864,red 8192] and 431 vs 257 us
+# at red 32768 (chain numbers INCLUDE both transposes); end-to-end 16.8M vs
+# 20.0M samples/s.  The 256 MB L3 keeps the transposed copies LLC-resident,
+# so the chain's extra traffic is nearly free, while ttv3's b16 scatter
+# staging gives up the glds direct-to-LDS path.  SHIFU_WGRAD_TT=1 enables.
+_WGRAD_TT = os.environ.get("SHIFU_WGRAD_TT", "0") == "1"
 _WGRAD_STREAM = None
 _WGRAD_EVENTS: list = []
 
