@@ -2855,12 +2855,33 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
   }
 }
 
+void emb_update_unified_binned(at::Tensor arena, at::Tensor acc, at::Tensor rows,
+                               at::Tensor dgrad, long dcol0, at::Tensor wide,
+                               long wstride, long F, double lr, double eps,
+                               bool adagrad, bool acc_in_arena);
+
+// SHIFU_EMB_BINNED=1 routes eligible unified updates through the atomic-free
+// binned path (above); default off pending measurement.
+static bool emb_binned_mode() {
+  static int v = [] {
+    const char* e = getenv("SHIFU_EMB_BINNED");
+    return e ? atoi(e) : 0;
+  }();
+  return v != 0;
+}
+
 // adagrad (rowwise, fp32 accumulator) over a unified arena from unpacked
 // grads; kind 0 = sgd (no accumulator)
 void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
                         at::Tensor dgrad, long dcol0, at::Tensor wide,
                         long wstride, long F, double lr, double eps,
                         bool adagrad, bool acc_in_arena) {
+  if (emb_binned_mode() && arena.size(1) <= 132 && arena.size(0) < (1L << 31)
+      && rows.numel() < (1L << 31)) {
+    emb_update_unified_binned(arena, acc, rows, dgrad, dcol0, wide, wstride,
+                              F, lr, eps, adagrad, acc_in_arena);
+    return;
+  }
   CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
   CHECK_GPU(dgrad); CHECK_BF16(dgrad);
   CHECK_GPU(wide); CHECK_BF16(wide);
@@ -2904,6 +2925,258 @@ void emb_update_unified(at::Tensor arena, at::Tensor acc, at::Tensor rows,
                      (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
                      accp, n, F, D, DP, dstride, dcol0, wstride, (float)-lr,
                      (float)eps, astride, aoff);
+}
+
+// ---------------------------------------------------------------------------
+// BINNED unified update — atomic-free owner-partitioned rewrite of
+// accsq+scatter.  The atomic chain issues ~33 packed-bf16 atomics per entry
+// (~28M per bench step), which bounds it on atomic issue/L2-RMW, not HBM.
+// Here entries are bucketed by arena-row HIGH bits (kernels: count → scan →
+// bin), so ONE workgroup owns every duplicate of a row; inside the workgroup
+// same-row entries are chained through an LDS hash on the row LOW bits
+// (unique within a bucket by construction — a bucket spans exactly the rows
+// sharing its high bits).  One 32-lane group then applies a row's WHOLE
+// update: f32-accumulate the chain's gradients (dgrad read ONCE — the
+// separate accsq pass is gone), update the adagrad accumulator, and RMW the
+// arena row with plain vector loads/stores.  Zero global atomics, and the
+// denominator still sees the row's full-step accumulator because all of a
+// row's entries sit in one chain.  (Only a bucket overflowing one 1024-entry
+// chunk — pathological id skew — splits a chain; those rows then see
+// partial-step denominators for the early chunks, which is still a valid
+// adagrad-style step.)
+// ---------------------------------------------------------------------------
+#define EB_CH 1024
+#define EB_HEAD 4096   // max bucket span (shift cap 12)
+
+__global__ void emb_bin_count_kernel(const long* __restrict__ rows,
+                                     int* __restrict__ counts, long n, int shift) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long e = i; e < n; e += stride)
+    atomicAdd(&counts[rows[e] >> shift], 1);
+}
+
+// offsets[0..nb] = exclusive scan; cursor = copy of the starts (kernel C
+// consumes it with atomics).  Single block, chunked Hillis-Steele.
+__global__ __launch_bounds__(256)
+void emb_bin_scan_kernel(const int* __restrict__ counts, int* __restrict__ offsets,
+                         int* __restrict__ cursor, long nb) {
+  __shared__ int buf[256];
+  __shared__ int carry;
+  int tid = threadIdx.x;
+  if (tid == 0) { carry = 0; offsets[0] = 0; }
+  __syncthreads();
+  for (long c0 = 0; c0 < nb; c0 += 256) {
+    int v = (c0 + tid < nb) ? counts[c0 + tid] : 0;
+    buf[tid] = v;
+    __syncthreads();
+    for (int off = 1; off < 256; off <<= 1) {
+      int add = (tid >= off) ? buf[tid - off] : 0;
+      __syncthreads();
+      buf[tid] += add;
+      __syncthreads();
+    }
+    if (c0 + tid < nb) {
+      int incl = carry + buf[tid];
+      offsets[c0 + tid + 1] = incl;
+      cursor[c0 + tid] = incl - v;
+    }
+    __syncthreads();
+    if (tid == 0) carry += buf[255];
+    __syncthreads();
+  }
+}
+
+__global__ void emb_bin_scatter_kernel(const long* __restrict__ rows,
+                                       int* __restrict__ cursor,
+                                       unsigned* __restrict__ ebin,
+                                       unsigned* __restrict__ rbin,
+                                       long n, int shift) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (long e = i; e < n; e += stride) {
+    long r = rows[e];
+    int slot = atomicAdd(&cursor[r >> shift], 1);
+    ebin[slot] = (unsigned)e;
+    rbin[slot] = (unsigned)r;
+  }
+}
+
+// one block per bucket; 32-lane groups each own whole rows (D <= 128)
+__global__ __launch_bounds__(256)
+void emb_apply_binned_kernel(bf16* __restrict__ arena,
+                             const unsigned* __restrict__ ebin,
+                             const unsigned* __restrict__ rbin,
+                             const int* __restrict__ offsets,
+                             const bf16* __restrict__ dgrad,
+                             const bf16* __restrict__ wide,
+                             float* __restrict__ accw,
+                             long F, long D, long DP,
+                             long dstride, long dcol0, long wstride,
+                             float scale, float eps, int shift,
+                             long astride, long aoff) {
+  __shared__ unsigned s_e[EB_CH];
+  __shared__ unsigned s_row[EB_CH];
+  __shared__ int s_next[EB_CH];
+  __shared__ int s_head[EB_HEAD];
+  __shared__ int s_work[EB_CH];
+  __shared__ int s_nwork;
+  const int b = blockIdx.x;
+  const int start = offsets[b], end = offsets[b + 1];
+  if (start >= end) return;
+  const int tid = threadIdx.x;
+  const int mask = (1 << shift) - 1;
+  const int warp = tid >> 5, lane = tid & 31;
+  const long pairs = D >> 1;
+
+  for (int c0 = start; c0 < end; c0 += EB_CH) {
+    int cnt = min(EB_CH, end - c0);
+    for (int i = tid; i < EB_HEAD; i += 256) s_head[i] = -1;
+    if (tid == 0) s_nwork = 0;
+    __syncthreads();
+    for (int j = tid; j < cnt; j += 256) {
+      unsigned e = ebin[c0 + j];
+      unsigned r = rbin[c0 + j];
+      s_e[j] = e;
+      s_row[j] = r;
+      int old = atomicExch(&s_head[r & mask], j);
+      s_next[j] = old;
+      if (old < 0) {                       // first occurrence claims the row
+        int w = atomicAdd(&s_nwork, 1);
+        s_work[w] = (int)(r & mask);
+      }
+    }
+    __syncthreads();
+    const int nw = s_nwork;
+    for (int w = warp; w < nw; w += 8) {
+      const int hl = s_work[w];
+      float ga[2][2] = {{0.f, 0.f}, {0.f, 0.f}};
+      float sqsum = 0.f, gw = 0.f;
+      for (int j = s_head[hl]; j >= 0; j = s_next[j]) {
+        unsigned e = s_e[j];
+        const unsigned* v2 = (const unsigned*)(dgrad + (long)(e / F) * dstride +
+                                               dcol0 + (long)(e % F) * D);
+#pragma unroll
+        for (int pi = 0; pi < 2; ++pi) {
+          long p = lane + pi * 32;
+          if (p < pairs) {
+            unsigned u = v2[p];
+            unsigned short lo = (unsigned short)(u & 0xffff);
+            unsigned short hi = (unsigned short)(u >> 16);
+            float a = __bfloat162float(*(const bf16*)&lo);
+            float c = __bfloat162float(*(const bf16*)&hi);
+            ga[pi][0] += a;
+            ga[pi][1] += c;
+            sqsum += a * a + c * c;
+          }
+        }
+        if (lane == 0) {
+          float wv = __bfloat162float(wide[(long)e * wstride]);
+          gw += wv;
+          sqsum += wv * wv;
+        }
+      }
+      const long row = (long)s_row[s_head[hl]];
+      float sc = scale;
+      if (accw) {
+#pragma unroll
+        for (int off = 16; off > 0; off >>= 1) sqsum += __shfl_down(sqsum, off, 32);
+        if (lane == 0) {
+          float* ap = &accw[row * astride + aoff];
+          float anew = *ap + sqsum / (float)DP;
+          *ap = anew;                       // exclusive owner: plain RMW
+          sc = scale / (sqrtf(anew) + eps);
+        }
+        sc = __shfl(sc, 0, 32);
+      }
+      unsigned* d32 = (unsigned*)(arena + row * DP);
+#pragma unroll
+      for (int pi = 0; pi < 2; ++pi) {
+        long p = lane + pi * 32;
+        if (p < pairs) {
+          unsigned u = d32[p];
+          unsigned short lo = (unsigned short)(u & 0xffff);
+          unsigned short hi = (unsigned short)(u >> 16);
+          bf16 nl = __float2bfloat16(__bfloat162float(*(const bf16*)&lo) + sc * ga[pi][0]);
+          bf16 nh = __float2bfloat16(__bfloat162float(*(const bf16*)&hi) + sc * ga[pi][1]);
+          d32[p] = (unsigned)*(unsigned short*)&nl | ((unsigned)*(unsigned short*)&nh << 16);
+        }
+      }
+      if (lane == 0) {                      // (wide, pad) pair — pad untouched
+        unsigned u = d32[pairs];
+        unsigned short lo = (unsigned short)(u & 0xffff);
+        bf16 nl = __float2bfloat16(__bfloat162float(*(const bf16*)&lo) + sc * gw);
+        d32[pairs] = (u & 0xffff0000u) | (unsigned)*(unsigned short*)&nl;
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// same contract as emb_update_unified, atomic-free binned implementation
+void emb_update_unified_binned(at::Tensor arena, at::Tensor acc, at::Tensor rows,
+                               at::Tensor dgrad, long dcol0, at::Tensor wide,
+                               long wstride, long F, double lr, double eps,
+                               bool adagrad, bool acc_in_arena) {
+  CHECK_GPU(arena); CHECK_CONTIG(arena); CHECK_BF16(arena);
+  CHECK_GPU(dgrad); CHECK_BF16(dgrad);
+  CHECK_GPU(wide); CHECK_BF16(wide);
+  long n = rows.numel();
+  if (!n) return;
+  long R = arena.size(0), DP = arena.size(1);
+  long D = DP - (acc_in_arena ? 4 : 2);
+  long dstride = dgrad.size(1);
+  TORCH_CHECK(D > 0 && D % 2 == 0 && D <= 128, "binned update needs even D <= 128");
+  TORCH_CHECK(R < (1L << 31) && n < (1L << 31), "binned update: 32-bit ids");
+  TORCH_CHECK(dgrad.stride(1) == 1 && dgrad.stride(0) == dstride,
+              "dgrad must be row-contiguous");
+  TORCH_CHECK(dcol0 % 2 == 0 && dstride % 2 == 0,
+              "unified update needs 4B-aligned deep columns");
+  auto s = cur_stream();
+  float* accw = nullptr;
+  long astride = 1, aoff = 0;
+  if (acc_in_arena) {
+    accw = (float*)arena.data_ptr();
+    astride = DP / 2;
+    aoff = (D + 2) / 2;
+  } else if (adagrad) {
+    CHECK_F32(acc);
+    accw = (float*)acc.data_ptr();
+  }
+  if (!adagrad) accw = nullptr;
+
+  // bucket span: largest power of two <= R/(n/256) so buckets average ~256
+  // entries (one LDS chunk), capped at the head-table size
+  long nb_t = std::max(n / 256, 1L);
+  long span = std::max(R / nb_t, 1L);
+  int shift = 0;
+  while ((2L << shift) <= span && shift < 12) shift++;
+  long nb = (R >> shift) + 1;
+
+  auto iopt = arena.options().dtype(at::kInt);
+  auto counts = at::zeros({nb}, iopt);
+  auto offsets = at::empty({nb + 1}, iopt);
+  auto cursor = at::empty({nb}, iopt);
+  auto ebin = at::empty({n}, iopt);
+  auto rbin = at::empty({n}, iopt);
+  int blocks = (int)std::min((n + 255) / 256, 4096L);
+  const long* rp = (const long*)rows.data_ptr();
+  hipLaunchKernelGGL(emb_bin_count_kernel, dim3(blocks), dim3(256), 0, s,
+                     rp, (int*)counts.data_ptr(), n, shift);
+  hipLaunchKernelGGL(emb_bin_scan_kernel, dim3(1), dim3(256), 0, s,
+                     (const int*)counts.data_ptr(), (int*)offsets.data_ptr(),
+                     (int*)cursor.data_ptr(), nb);
+  hipLaunchKernelGGL(emb_bin_scatter_kernel, dim3(blocks), dim3(256), 0, s,
+                     rp, (int*)cursor.data_ptr(),
+                     (unsigned*)ebin.data_ptr(), (unsigned*)rbin.data_ptr(),
+                     n, shift);
+  hipLaunchKernelGGL(emb_apply_binned_kernel, dim3((unsigned)nb), dim3(256), 0, s,
+                     (bf16*)arena.data_ptr(),
+                     (const unsigned*)ebin.data_ptr(), (const unsigned*)rbin.data_ptr(),
+                     (const int*)offsets.data_ptr(),
+                     (const bf16*)dgrad.data_ptr(), (const bf16*)wide.data_ptr(),
+                     accw, F, D, DP, dstride, dcol0, wstride,
+                     (float)-lr, (float)eps, shift, astride, aoff);
 }
 
 void emb_sgd_step(at::Tensor arena, at::Tensor rows, at::Tensor vals, double lr) {
@@ -3165,6 +3438,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_gather_into", &embedding_gather_into);
   m.def("emb_gather_split", &emb_gather_split);
   m.def("emb_update_unified", &emb_update_unified);
+  m.def("emb_update_unified_binned", &emb_update_unified_binned);
   m.def("emb_sgd_step", &emb_sgd_step);
   m.def("emb_adagrad_step", &emb_adagrad_step);
 }

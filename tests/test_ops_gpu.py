@@ -603,6 +603,52 @@ def test_unified_deferred_matches_packed():
     assert torch.allclose(a1, a0, atol=2e-2), "arena values diverged"
 
 
+@pytest.mark.parametrize("R,n,D,adagrad,in_arena", [
+    (5000, 8192, 64, True, True),     # bench-like, light dups
+    (64, 4096, 64, True, True),       # heavy dups: long chains, multi-chunk
+    (300, 2048, 16, True, False),     # external accumulator array
+    (300, 2048, 16, False, True),     # sgd mode (no accumulator)
+])
+def test_emb_update_binned_matches_atomic(R, n, D, adagrad, in_arena):
+    """The atomic-free binned unified update must match the atomic
+    accsq+scatter chain (same full-step adagrad denominator, f32-accumulated
+    values vs per-entry bf16 atomic adds -> small tolerance)."""
+    ext = hip_ops()
+    DP = D + (4 if in_arena else 2)
+    F = 4
+    torch.manual_seed(R + D)
+    arena = (torch.randn(R, DP, device="cuda") * 0.1).to(torch.bfloat16)
+    arena[:, D + 1] = 0  # pad col
+    if in_arena:
+        arena[:, D + 2:].view(torch.float32).fill_(0.01)
+    acc = torch.full((R,), 0.01, device="cuda")
+    ids = torch.randint(0, R, (n // F, F), device="cuda")
+    rows = ids.reshape(-1)
+    dgrad = (torch.randn(n // F, F * D, device="cuda") * 0.1).to(torch.bfloat16)
+    wide = (torch.randn(n // F, F, device="cuda") * 0.1).to(torch.bfloat16)
+
+    def run(fn):
+        a = arena.clone()
+        s = acc.clone()
+        fn(a, s, rows, dgrad, 0, wide.reshape(-1), 1, F, 0.05, 1e-8,
+           adagrad, in_arena)
+        accv = (a[:, D + 2:].view(torch.float32).reshape(-1).clone()
+                if in_arena else s)
+        return a[:, :D + 1].float(), a[:, D + 1].float(), accv
+
+    va, pada, sa = run(ext.emb_update_unified)
+    vb, padb, sb = run(ext.emb_update_unified_binned)
+    assert torch.all(pada == 0) and torch.all(padb == 0), "pad corrupted"
+    assert torch.allclose(sa, sb, rtol=1e-4, atol=1e-6), \
+        f"acc diverged max={float((sa - sb).abs().max())}"
+    assert torch.allclose(va, vb, atol=3e-2), \
+        f"values diverged max={float((va - vb).abs().max())}"
+    # rows never touched must be bitwise identical
+    touched = torch.zeros(R, dtype=torch.bool, device="cuda")
+    touched[rows] = True
+    assert torch.equal(va[~touched], vb[~touched])
+
+
 @pytest.mark.parametrize("B,N,act", [(128, 64, 3), (513, 130, 2), (64, 33, 1),
                                      (1000, 256, 3), (70, 7, 0)])
 def test_act_grad_colsum_T(B, N, act):
