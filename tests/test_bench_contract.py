@@ -77,3 +77,22 @@ def test_bench_torchrun_4rank_table_ep():
     d = json.loads(lines[0])
     assert d["n_gpus"] == 4
     assert "ep4" in d["config"]["parallelism"]
+
+
+def test_bench_torchrun_8rank_table_ep():
+    """world=8 (gloo CPU): the driver's SCALE shape at N=8.  With n-cat=26
+    (the headline feature count) the greedy assignment puts 3-4 tables on
+    each rank; checks the full 8-way static all-to-all path end to end."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+         "--master-port", "29657", "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "16", "--vocab", "200", "--n-cat", "26"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 8
+    assert d["config"]["global_batch"] == 128
+    assert "ep8" in d["config"]["parallelism"]
