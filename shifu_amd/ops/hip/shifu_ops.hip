@@ -2739,17 +2739,42 @@ __global__ void emb_gather_split_kernel(const bf16* __restrict__ arena,
   long total = deep_total + rows_bf;           // + one wide element per row
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
-  for (long t = i; t < total; t += stride) {
+  // 2-way unroll: the ids -> arena chain is a dependent random load (~82%
+  // WAIT_ANY per PMC) and the capped grid makes each thread loop; issuing
+  // two arena loads before either store doubles the random loads in flight.
+  for (long t = i; t < total; t += 2 * stride) {
+    long u = t + stride;
+    s16x8 v1; bf16 w1; long o1 = -1, ow1 = 0;
     if (t < deep_total) {
       long rf = t / chunks_per_row;
       long c = t % chunks_per_row;
       long row = ids[rf];
       long b = rf / F, f = rf % F;
-      *(s16x8*)(out + b * out_stride + col0 + f * D + c * 8) =
-          *(const s16x8*)(arena + row * DP + c * 8);
+      v1 = *(const s16x8*)(arena + row * DP + c * 8);
+      o1 = b * out_stride + col0 + f * D + c * 8;
     } else {
       long rf = t - deep_total;
-      wide[rf] = arena[ids[rf] * DP + D];
+      w1 = arena[ids[rf] * DP + D];
+      ow1 = rf;
+    }
+    s16x8 v2; bf16 w2; long o2 = -1, ow2 = 0;
+    if (u < total) {
+      if (u < deep_total) {
+        long rf = u / chunks_per_row;
+        long c = u % chunks_per_row;
+        long row = ids[rf];
+        long b = rf / F, f = rf % F;
+        v2 = *(const s16x8*)(arena + row * DP + c * 8);
+        o2 = b * out_stride + col0 + f * D + c * 8;
+      } else {
+        long rf = u - deep_total;
+        w2 = arena[ids[rf] * DP + D];
+        ow2 = rf;
+      }
+    }
+    if (o1 >= 0) *(s16x8*)(out + o1) = v1; else wide[ow1] = w1;
+    if (u < total) {
+      if (o2 >= 0) *(s16x8*)(out + o2) = v2; else wide[ow2] = w2;
     }
   }
 }
@@ -2829,15 +2854,24 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
   // 32 lanes per entry: lane k applies deep pairs k, k+32, ...; lane 0 also
   // the (wide, pad) pair.  Shift-only index math — a 64-bit `t / 33` per
   // element (the naive pairs-flattened mapping) measured ~2x slower.
+  // The grid is capped (scat_blocks), so each group loops ~n/32768 entries;
+  // the rows[e] -> acc[row] chain is two dependent RANDOM loads (~85%
+  // WAIT_ANY per PMC), so the NEXT entry's chain is prefetched while the
+  // current one is applied — legal to hoist past the arena atomics because
+  // the scatter never touches the accumulator columns.
   long e = ((long)blockIdx.x * blockDim.x + threadIdx.x) >> 5;
   long estride = ((long)gridDim.x * blockDim.x) >> 5;
   int lane = threadIdx.x & 31;
   long hp = D >> 1;
+  long row = (e < n) ? rows[e] : 0;
+  float av = (acc && e < n) ? acc[row * astride + aoff] : 0.0f;
   for (; e < n; e += estride) {
-    long row = rows[e];
+    long en = e + estride;
+    long row_n = (en < n) ? rows[en] : 0;
+    float av_n = (acc && en < n) ? acc[row_n * astride + aoff] : 0.0f;
     // adagrad denominator inline (accsq pass completed): one broadcast
     // read replaces the separate emb_denom kernel + rowscale buffer
-    float sc = acc ? scale / (sqrtf(acc[row * astride + aoff]) + eps) : scale;
+    float sc = acc ? scale / (sqrtf(av) + eps) : scale;
     const bf16* src = dgrad + (e / F) * dstride + dcol0 + (e % F) * D;
     bf16* dst = arena + row * DP;
     for (long dp = lane; dp < hp; dp += 32) {
@@ -2852,6 +2886,8 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
       add.y = __float2bfloat16(0.0f);    // pad column stays zero
       unsafeAtomicAdd((__hip_bfloat162*)(dst + D), add);
     }
+    row = row_n;
+    av = av_n;
   }
 }
 
