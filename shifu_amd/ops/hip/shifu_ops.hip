@@ -3290,26 +3290,50 @@ void gemv_fwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ w,
 }
 
 // dw[k] = sum_b x[b,k]*dz[b] ; db = sum_b dz[b]  (column-parallel + atomics)
+// slab variant: thread t owns 8 columns (16B vector loads — the old 2B
+// per-thread column walk issued 8x the load instructions and measured ~10x
+// off roofline); each row-chunk block plain-stores its [K+1] partial
+// (dw | db) into wslab and splitk_reduce combines — no f32 atomics, bitwise
+// deterministic (the atomic version's dw contention is what made the finer
+// row chop regress in round 1).
 __global__ void gemv_wgrad_kernel(const bf16* __restrict__ x, const bf16* __restrict__ dz,
-                                  float* __restrict__ dw, float* __restrict__ db,
+                                  float* __restrict__ wslab,
                                   long B, long K, long rows_per_chunk) {
-  long k = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  long b0 = (long)blockIdx.y * rows_per_chunk;
+  long b0 = (long)blockIdx.x * rows_per_chunk;
   long b1 = min(B, b0 + rows_per_chunk);
-  if (k < K) {
-    float acc = 0.0f;
-    for (long b = b0; b < b1; ++b)
-      acc += __bfloat162float(x[b * K + k]) * __bfloat162float(dz[b]);
-    if (gridDim.y == 1) dw[k] = acc;
-    else atomicAdd(&dw[k], acc);
+  float* out = wslab + (long)blockIdx.x * (K + 1);
+  long kc = (K + 7) >> 3;
+  const bool vec = ((K & 7) == 0);
+  for (long t = threadIdx.x; t < kc; t += blockDim.x) {
+    long k8 = t * 8;
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    if (vec) {
+      for (long b = b0; b < b1; ++b) {
+        float d = __bfloat162float(dz[b]);
+        s16x8 v = *(const s16x8*)(x + b * K + k8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[j] += d * __bfloat162float(*(const bf16*)&((const short*)&v)[j]);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[k8 + j] = acc[j];
+    } else {
+      long ke = min(k8 + 8, K);
+      for (long b = b0; b < b1; ++b) {
+        float d = __bfloat162float(dz[b]);
+        for (long k = k8; k < ke; ++k)
+          acc[k - k8] += d * __bfloat162float(x[b * K + k]);
+      }
+      for (long k = k8; k < ke; ++k) out[k] = acc[k - k8];
+    }
   }
-  if (blockIdx.x == 0 && threadIdx.x < 64) {   // one wave reduces db chunk
+  if (threadIdx.x < 64) {   // one wave reduces this chunk's db partial
     float acc = 0.0f;
     for (long b = b0 + threadIdx.x; b < b1; b += 64)
       acc += __bfloat162float(dz[b]);
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
-    if (threadIdx.x == 0) atomicAdd(&db[0], acc);
+    if (threadIdx.x == 0) out[K] = acc;
   }
 }
 
@@ -3354,21 +3378,20 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
                                  bool need_dx) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_BF16(x);
   long B = x.size(0), K = x.size(1);
-  // one allocation + one memset for dw|db (small-kernel launch overhead
-  // was ~5us per fill inside the captured step)
+  // dw|db in one buffer; per-chunk partials go to a [chunks, K+1] slab and
+  // splitk_reduce combines (no memset, no atomics, deterministic)
   auto wb = at::empty({K + 1}, x.options().dtype(at::kFloat));
   auto dw = wb.narrow(0, 0, K).view({1, K});
   auto db = wb.narrow(0, K, 1);
-  hipMemsetAsync(wb.data_ptr(), 0, (size_t)(K + 1) * 4, cur_stream());
-  // NOTE a finer row chop (4096 chunks) was tried and REGRESSED 8% e2e:
-  // every chunk atomically adds all K partials onto the same ~K*4 bytes, and
-  // the line contention dwarfs the latency win.  Keep the coarse grid.
-  long gx, chunks, rpc;
-  colsum_grid(B, K, gx, chunks, rpc);
-  hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)gx, (unsigned)chunks),
-                     dim3(256), 0, cur_stream(),
+  long chunks = std::min<long>(1024, std::max<long>(B / 64, 1));
+  long rpc = (B + chunks - 1) / chunks;
+  auto wslab = at::empty({chunks, K + 1}, x.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)chunks), dim3(256), 0,
+                     cur_stream(),
                      (const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
-                     (float*)dw.data_ptr(), (float*)db.data_ptr(), B, K, rpc);
+                     (float*)wslab.data_ptr(), B, K, rpc);
+  launch_splitk_reduce((const float*)wslab.data_ptr(), (float*)wb.data_ptr(),
+                       K + 1, chunks, cur_stream(), false);
   at::Tensor dx;
   if (need_dx) {
     dx = at::empty({B, K}, x.options());
