@@ -3290,6 +3290,21 @@ void gemv_fwd_kernel(const bf16* __restrict__ x, const bf16* __restrict__ w,
 }
 
 // dw[k] = sum_b x[b,k]*dz[b] ; db = sum_b dz[b]  (column-parallel + atomics)
+// slab reduce for SMALL outputs with MANY z-parts (gemv dw|db: MN ~ a few
+// hundred, z up to 2048): 64 lanes per output element, parallel over z —
+// splitk_reduce's one-lane-per-element loop would serialize 2048 loads.
+__global__ void slab_reduce_wide_kernel(const float* __restrict__ W,
+                                        float* __restrict__ C, long MN, int z) {
+  long e = (long)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (e >= MN) return;
+  int lane = threadIdx.x & 63;
+  float s = 0.0f;
+  for (int zz = lane; zz < z; zz += 64) s += W[(long)zz * MN + e];
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) s += __shfl_down(s, off, 64);
+  if (lane == 0) C[e] = s;
+}
+
 // slab variant: thread t owns 8 columns (16B vector loads — the old 2B
 // per-thread column walk issued 8x the load instructions and measured ~10x
 // off roofline); each row-chunk block plain-stores its [K+1] partial
@@ -3303,11 +3318,10 @@ __global__ void gemv_wgrad_kernel(const bf16* __restrict__ x, const bf16* __rest
   long b1 = min(B, b0 + rows_per_chunk);
   float* out = wslab + (long)blockIdx.x * (K + 1);
   long kc = (K + 7) >> 3;
-  const bool vec = ((K & 7) == 0);
-  for (long t = threadIdx.x; t < kc; t += blockDim.x) {
-    long k8 = t * 8;
-    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-    if (vec) {
+  if ((K & 7) == 0) {
+    for (long t = threadIdx.x; t < kc; t += blockDim.x) {
+      long k8 = t * 8;
+      float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
       for (long b = b0; b < b1; ++b) {
         float d = __bfloat162float(dz[b]);
         s16x8 v = *(const s16x8*)(x + b * K + k8);
@@ -3317,14 +3331,14 @@ __global__ void gemv_wgrad_kernel(const bf16* __restrict__ x, const bf16* __rest
       }
 #pragma unroll
       for (int j = 0; j < 8; ++j) out[k8 + j] = acc[j];
-    } else {
-      long ke = min(k8 + 8, K);
-      for (long b = b0; b < b1; ++b) {
-        float d = __bfloat162float(dz[b]);
-        for (long k = k8; k < ke; ++k)
-          acc[k - k8] += d * __bfloat162float(x[b * K + k]);
-      }
-      for (long k = k8; k < ke; ++k) out[k] = acc[k - k8];
+    }
+  } else {
+    // odd K: per-column mapping (one scalar load per lane per row)
+    for (long k = threadIdx.x; k < K; k += blockDim.x) {
+      float acc = 0.0f;
+      for (long b = b0; b < b1; ++b)
+        acc += __bfloat162float(dz[b]) * __bfloat162float(x[b * K + k]);
+      out[k] = acc;
     }
   }
   if (threadIdx.x < 64) {   // one wave reduces this chunk's db partial
@@ -3383,15 +3397,20 @@ std::vector<at::Tensor> gemv_bwd(at::Tensor x, at::Tensor w, at::Tensor dz,
   auto wb = at::empty({K + 1}, x.options().dtype(at::kFloat));
   auto dw = wb.narrow(0, 0, K).view({1, K});
   auto db = wb.narrow(0, K, 1);
-  long chunks = std::min<long>(1024, std::max<long>(B / 64, 1));
+  // 64-thread blocks (one wave) x 2048 chunks: same resident-wave count as
+  // the old 256x512 grid — the vector loads alone regressed 4x when packed
+  // into 1/4 the waves (memory parallelism, not instruction count, rules)
+  long chunks = std::min<long>(2048, std::max<long>(B / 16, 1));
   long rpc = (B + chunks - 1) / chunks;
   auto wslab = at::empty({chunks, K + 1}, x.options().dtype(at::kFloat));
-  hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)chunks), dim3(256), 0,
+  hipLaunchKernelGGL(gemv_wgrad_kernel, dim3((unsigned)chunks), dim3(64), 0,
                      cur_stream(),
                      (const bf16*)x.data_ptr(), (const bf16*)dz.data_ptr(),
                      (float*)wslab.data_ptr(), B, K, rpc);
-  launch_splitk_reduce((const float*)wslab.data_ptr(), (float*)wb.data_ptr(),
-                       K + 1, chunks, cur_stream(), false);
+  hipLaunchKernelGGL(slab_reduce_wide_kernel,
+                     dim3((unsigned)((K + 1 + 3) / 4)), dim3(256), 0,
+                     cur_stream(), (const float*)wslab.data_ptr(),
+                     (float*)wb.data_ptr(), K + 1, (int)chunks);
   at::Tensor dx;
   if (need_dx) {
     dx = at::empty({B, K}, x.options());
