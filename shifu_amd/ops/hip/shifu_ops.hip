@@ -2883,12 +2883,16 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
   long estride = ((long)gridDim.x * blockDim.x) >> 5;
   int lane = threadIdx.x & 31;
   long hp = D >> 1;
+  // 2-deep: row ids prefetched two iterations ahead, accumulators one —
+  // each link of the rows -> acc chain then has a full iteration of slack,
+  // so neither random-load latency sits on the critical path.
   long row = (e < n) ? rows[e] : 0;
+  long row1 = (e + estride < n) ? rows[e + estride] : 0;
   float av = (acc && e < n) ? acc[row * astride + aoff] : 0.0f;
   for (; e < n; e += estride) {
-    long en = e + estride;
-    long row_n = (en < n) ? rows[en] : 0;
-    float av_n = (acc && en < n) ? acc[row_n * astride + aoff] : 0.0f;
+    long e2 = e + 2 * estride;
+    long row2 = (e2 < n) ? rows[e2] : 0;
+    float av_n = (acc && e + estride < n) ? acc[row1 * astride + aoff] : 0.0f;
     // adagrad denominator inline (accsq pass completed): one broadcast
     // read replaces the separate emb_denom kernel + rowscale buffer
     float sc = acc ? scale / (sqrtf(av) + eps) : scale;
@@ -2906,7 +2910,8 @@ __global__ void emb_scatter_uni_kernel(bf16* __restrict__ arena, const long* __r
       add.y = __float2bfloat16(0.0f);    // pad column stays zero
       unsafeAtomicAdd((__hip_bfloat162*)(dst + D), add);
     }
-    row = row_n;
+    row = row1;
+    row1 = row2;
     av = av_n;
   }
 }
