@@ -209,6 +209,26 @@ def test_gemm_ttv3_wgrad(R, M, N):
     assert torch.equal(c, c2)
 
 
+def test_gemv_bwd_deterministic():
+    """Head wgrad (slab reduce, no atomics): bitwise repeatable and matches
+    the fp32 reference, for both vector (K%8==0) and odd-K paths."""
+    ext = hip_ops()
+    for K in (256, 130):
+        x = _rand_bf16(4096, K, seed=K)
+        w = _rand_bf16(1, K, seed=K + 1).reshape(-1)
+        dz = _rand_bf16(4096, 1, seed=K + 2).reshape(-1)
+        dw, db, dx = ext.gemv_bwd(x, w, dz, True)
+        dw2, db2, _ = ext.gemv_bwd(x, w, dz, True)
+        assert torch.equal(dw, dw2) and torch.equal(db, db2), f"K={K} nondet"
+        ref_dw = (dz.float().unsqueeze(0) @ x.float()).reshape(-1)
+        ok, err = _rel_close(dw.reshape(-1), ref_dw)
+        assert ok, f"K={K} dw maxdiff={err}"
+        ok, err = _rel_close(db, dz.float().sum().reshape(1))
+        assert ok, f"K={K} db maxdiff={err}"
+        ok, err = _rel_close(dx, dz.float().unsqueeze(1) * w.float().unsqueeze(0))
+        assert ok, f"K={K} dx maxdiff={err}"
+
+
 def test_act_grad_colsum_into():
     """dzT-free partner of ttv3: dz = dy*act'(y), db ACCUMULATED into view."""
     torch.manual_seed(5)
