@@ -209,6 +209,33 @@ def test_gemm_ttv3_wgrad(R, M, N):
     assert torch.equal(c, c2)
 
 
+def test_wgrad_tt_route_matches_default():
+    """SHIFU_WGRAD_TT=1 routing in linear.py (ttv3 + colsum-into) must
+    produce the same grads as the default transpose+NT route."""
+    import shifu_amd.ops.linear as lin
+    from shifu_amd.ops.linear import fused_linear
+
+    def grads(tt):
+        old = lin._WGRAD_TT
+        lin._WGRAD_TT = tt
+        try:
+            torch.manual_seed(4)
+            x = torch.randn(512, 128).to(torch.bfloat16).cuda().requires_grad_(True)
+            w = torch.randn(64, 128).cuda().requires_grad_(True)
+            b = torch.zeros(64).cuda().requires_grad_(True)
+            fused_linear(x, w, b, "relu").float().pow(2).sum().backward()
+            return w.grad.clone(), b.grad.clone(), x.grad.clone()
+        finally:
+            lin._WGRAD_TT = old
+    dw1, db1, dx1 = grads(True)
+    dw0, db0, dx0 = grads(False)
+    assert torch.equal(dx1, dx0), "dx diverged"
+    ok, err = _rel_close(dw1, dw0.float(), 1e-2)
+    assert ok, f"dw tt-vs-default maxdiff={err}"
+    ok, err = _rel_close(db1, db0.float(), 1e-2)
+    assert ok, f"db tt-vs-default maxdiff={err}"
+
+
 def test_gemv_bwd_deterministic():
     """Head wgrad (slab reduce, no atomics): bitwise repeatable and matches
     the fp32 reference, for both vector (K%8==0) and odd-K paths."""
