@@ -11,7 +11,9 @@ scatter-transposes):
   dgrad : dx = nt(dz [B,N], w^T [K,N])        (w^T = one small per-step
                                                tiled-transpose kernel)
   wgrad : dw = nt(dz^T [N,B], x^T [K,B])      (activation transposes, f32
-                                               out, split-K)
+                                               out, split-K; the transpose-
+                                               free ttv3 alternative measured
+                                               slower — see _WGRAD_TT below)
 Mixed precision: fp32 master weights (autograd leaves), bf16 compute via
 per-layer views of the flat arena's bf16 mirror (ONE arena-wide cast per
 step — ops/flat.py), fp32 dW/db, bf16 dX.  Weight/bias grads can run on a
