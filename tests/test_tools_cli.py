@@ -42,3 +42,22 @@ def test_bench_help():
                        capture_output=True, text=True, timeout=120)
     assert r.returncode == 0, r.stderr
     assert "--gpus" in r.stdout
+
+
+def test_score_bench_functional_cpu():
+    """score_bench end to end on a tiny synthetic model (CPU): emits both
+    JSON lines with sane values."""
+    import json
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "tools", "score_bench.py"),
+         "--rows", "4096", "--batch", "2048", "--vocab", "50",
+         "--n-cat", "3", "--n-dense", "8"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [json.loads(l) for l in r.stdout.strip().splitlines()
+             if l.startswith("{")]
+    assert len(lines) == 2
+    batch, row = lines
+    assert batch["metric"] == "scoring_rows_per_sec" and batch["value"] > 0
+    assert batch["rows"] == 4096
+    assert row["metric"] == "scoring_row_latency_us" and row["value"] > 0
