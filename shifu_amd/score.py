@@ -11,11 +11,21 @@ score per row, and optionally report AUC against the target column.
 
 Without --column-config the row layout is assumed to be the exported feature
 order (num_dense floats, then categorical ids) with no target/weight columns.
+
+Distributed eval (the reference runs scoring as a parallel Hadoop job):
+launch the same command under torchrun — files are round-robined across
+ranks (one process per GPU with --device cuda), each rank writes
+`<output>.partR`, and rank 0 reports rows + whole-set AUC:
+
+    torchrun --nproc-per-node 8 --master-addr 127.0.0.1 -m shifu_amd.score \
+        --model final_model/ --data eval/ --column-config CC.json \
+        --output scores.csv --auc --device cuda
 """
 from __future__ import annotations
 
 import argparse
 import json
+import os
 import sys
 from typing import List, Optional
 
@@ -77,7 +87,8 @@ def main(argv=None) -> int:
     ap.add_argument("--model", required=True, help="export directory")
     ap.add_argument("--data", required=True, nargs="+", help="csv(.gz) files/dirs")
     ap.add_argument("--column-config", default=None)
-    ap.add_argument("--output", default="-", help="scores file ('-' = stdout)")
+    ap.add_argument("--output", default="-", help="scores file ('-' = stdout); "
+                    "multi-rank runs write <output>.partR per rank")
     ap.add_argument("--batch", type=int, default=65536)
     ap.add_argument("--device", default="cpu")
     ap.add_argument("--delimiter", default="|")
@@ -85,12 +96,36 @@ def main(argv=None) -> int:
                     help="also print AUC vs the target column (needs --column-config)")
     args = ap.parse_args(argv)
 
-    from shifu_amd.data.csv_loader import list_training_files
-    paths = list_training_files(list(args.data))
-    scores, targets = score_files(args.model, paths, args.column_config,
-                                  args.batch, args.device, args.delimiter)
+    # Distributed eval (successor of the reference scoring an eval set as a
+    # parallel Hadoop job): launch via torchrun — files are round-robined
+    # across ranks, each rank scores its shard (one process per GPU with
+    # --device cuda), scores land in per-rank part files in file-shard
+    # order, and rank 0 reports the AUC over the gathered whole set.
+    rank, world = 0, 1
+    dist = None
+    if os.environ.get("RANK") is not None and os.environ.get("WORLD_SIZE"):
+        import torch.distributed as dist
+        dist.init_process_group("gloo" if args.device == "cpu" else "nccl")
+        rank, world = dist.get_rank(), dist.get_world_size()
 
-    out = sys.stdout if args.output == "-" else open(args.output, "w")
+    from shifu_amd.data.csv_loader import list_training_files
+    from shifu_amd.data.sharding import shard_files
+    paths = list_training_files(list(args.data))
+    my_paths = shard_files(paths, rank, world) if world > 1 else paths
+    if world > 1 and args.device == "cuda":
+        import torch
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    scores, targets = (score_files(args.model, my_paths, args.column_config,
+                                   args.batch, args.device, args.delimiter)
+                       if my_paths else (np.empty(0), None))
+
+    if args.output == "-" and world == 1:
+        out = sys.stdout
+    else:
+        path = args.output if world == 1 else \
+            (f"{args.output}.part{rank}" if args.output != "-"
+             else f"scores.part{rank}")
+        out = open(path, "w")
     try:
         for v in scores:
             out.write(f"{v:.6f}\n")
@@ -98,7 +133,19 @@ def main(argv=None) -> int:
         if out is not sys.stdout:
             out.close()
 
-    summary = {"rows": int(len(scores))}
+    if world > 1:
+        # gather (score, target) arrays so rank 0 computes the WHOLE-set AUC
+        gathered = [None] * world if rank == 0 else None
+        dist.gather_object((scores, targets), gathered, dst=0)
+        if rank != 0:
+            dist.destroy_process_group()
+            return 0
+        scores = np.concatenate([g[0] for g in gathered])
+        tl = [g[1] for g in gathered if g[1] is not None and len(g[1])]
+        targets = np.concatenate(tl) if tl else None
+        dist.destroy_process_group()
+
+    summary = {"rows": int(len(scores)), "ranks": world}
     if args.auc and targets is not None and len(scores):
         from shifu_amd.train.trainer import auc_score
         summary["auc"] = auc_score(scores, targets)
