@@ -145,6 +145,9 @@ def main(argv=None) -> int:
         targets = np.concatenate(tl) if tl else None
         dist.destroy_process_group()
 
+    if world == 1 and dist is not None and dist.is_initialized():
+        dist.destroy_process_group()   # torchrun with a single rank
+
     summary = {"rows": int(len(scores)), "ranks": world}
     if args.auc and targets is not None and len(scores):
         from shifu_amd.train.trainer import auc_score
