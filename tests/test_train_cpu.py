@@ -291,3 +291,28 @@ def test_streaming_residency_matches_device(tmp_path):
     resident = run("auto", "a")
     streamed = run("stream", "b")
     assert abs(resident - streamed) < 1e-7, (resident, streamed)
+
+
+def test_run_to_run_determinism(tmp_path):
+    """Two identical CPU runs (same seeds/config, fresh dirs) must produce
+    BITWISE identical final parameters — the determinism guarantee the
+    precision model documents (ARCHITECTURE §8) at world=1."""
+    def run(tag):
+        train, valid = _data(800, 6, (20, 30), seed=7)
+        rc = RunConfig(tmp_model_path=str(tmp_path / f"ck{tag}"),
+                       final_model_path=str(tmp_path / f"fn{tag}"),
+                       log_dir=str(tmp_path / f"lg{tag}"), device="cpu",
+                       model_type="wide_deep", embed_dim=4, seed=3,
+                       vocab_sizes=[20, 30])
+        from shifu_amd.models.mlp import build_model
+        model = build_model(_mc(epochs=2), 6, [20, 30],
+                            model_type="wide_deep", embed_dim=4, seed=3)
+        tr = Trainer(model, _mc(epochs=2), rc, train, valid)
+        tr.fit()
+        return {k: v.detach().clone() for k, v in model.state_dict().items()}
+
+    s1 = run("a")
+    s2 = run("b")
+    assert s1.keys() == s2.keys()
+    for k in s1:
+        assert torch.equal(s1[k], s2[k]), f"nondeterministic param {k}"
