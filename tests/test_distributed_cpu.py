@@ -155,3 +155,11 @@ def test_trainer_2rank_params_converge(tmp_path):
     outs = _run_workers(_trainer_dp_worker, 29615, extra=(str(tmp_path),))
     sums = [s for _, s in outs]
     assert abs(sums[0] - sums[1]) < 1e-5, f"ranks diverged: {sums}"
+
+
+def test_trainer_2rank_run_to_run_deterministic(tmp_path):
+    """Two identical 2-rank runs must produce BITWISE identical final
+    parameters (gloo all-reduce + fp32 CPU path are deterministic)."""
+    o1 = _run_workers(_trainer_dp_worker, 29617, extra=(str(tmp_path / "a"),))
+    o2 = _run_workers(_trainer_dp_worker, 29619, extra=(str(tmp_path / "b"),))
+    assert sorted(o1) == sorted(o2), f"cross-run divergence: {o1} vs {o2}"
