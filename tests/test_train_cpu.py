@@ -316,3 +316,20 @@ def test_run_to_run_determinism(tmp_path):
     assert s1.keys() == s2.keys()
     for k in s1:
         assert torch.equal(s1[k], s2[k]), f"nondeterministic param {k}"
+
+
+def test_auc_matches_sklearn():
+    """auc_score (rank-based, tie-averaged) against sklearn's roc_auc_score
+    on random data including heavy ties."""
+    from sklearn.metrics import roc_auc_score
+    rng = np.random.default_rng(0)
+    for trial in range(20):
+        n = int(rng.integers(10, 500))
+        labels = (rng.random(n) > 0.5).astype(np.float64)
+        if labels.min() == labels.max():
+            labels[0] = 1.0 - labels[0]
+        # quantized scores force tie handling
+        scores = np.round(rng.random(n), int(rng.integers(0, 3)))
+        ours = auc_score(scores, labels)
+        ref = roc_auc_score(labels, scores)
+        assert abs(ours - ref) < 1e-9, f"trial {trial}: {ours} vs {ref}"
