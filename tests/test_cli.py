@@ -164,3 +164,13 @@ def test_score_cli_distributed_2rank(tmp_path):
                        "--column-config", cc_path, "--output", sp]) == 0
     n_single = len(open(sp).read().strip().splitlines())
     assert n_single == 400
+
+
+def test_module_alias_help():
+    """`python -m shifu_amd` is the documented CLI alias for shifu_amd.run."""
+    import subprocess
+    import sys
+    r = subprocess.run([sys.executable, "-m", "shifu_amd", "--help"],
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0
+    assert "--run-config" in r.stdout
