@@ -96,3 +96,20 @@ def test_bench_torchrun_8rank_table_ep():
     assert d["n_gpus"] == 8
     assert d["config"]["global_batch"] == 128
     assert "ep8" in d["config"]["parallelism"]
+
+
+def test_bench_torchrun_2rank_ep_row():
+    """--emb-mode ep_row at world=2: row%world-sharded arenas through the
+    real bench script (the beyond-HBM single-table layout)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29659", "bench.py", "--steps", "2", "--warmup", "1",
+         "--batch", "32", "--vocab", "1000", "--n-cat", "4",
+         "--emb-mode", "ep_row"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
