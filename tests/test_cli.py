@@ -174,3 +174,43 @@ def test_module_alias_help():
                        capture_output=True, text=True, timeout=120)
     assert r.returncode == 0
     assert "--run-config" in r.stdout
+
+
+def test_quickstart_example(tmp_path):
+    """examples/quickstart must run end to end exactly as its README says
+    (data -> train -> score with AUC)."""
+    import shutil
+    import subprocess
+    import sys
+    ex = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                      "examples", "quickstart")
+    work = tmp_path / "qs"
+    shutil.copytree(ex, work)
+    env = dict(os.environ,
+               PYTHONPATH=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    # fewer epochs for test speed
+    mc = json.loads((work / "ModelConfig.json").read_text())
+    mc["train"]["numTrainEpochs"] = 2
+    (work / "ModelConfig.json").write_text(json.dumps(mc))
+    r = subprocess.run([sys.executable, "make_data.py", "./data"], cwd=work,
+                       env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1000:]
+    r = subprocess.run([sys.executable, "-m", "shifu_amd.run",
+                        "--run-config", "run.json",
+                        "--model-config", "ModelConfig.json",
+                        "--column-config", "ColumnConfig.json"],
+                       cwd=work, env=env, capture_output=True, text=True,
+                       timeout=900)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (work / "final_model" / "GenericModelConfig.json").exists()
+    r = subprocess.run([sys.executable, "-m", "shifu_amd.score",
+                        "--model", "final_model", "--data", "./data",
+                        "--column-config", "ColumnConfig.json",
+                        "--output", "scores.csv", "--auc"],
+                       cwd=work, env=env, capture_output=True, text=True,
+                       timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert len((work / "scores.csv").read_text().strip().splitlines()) == 20000
+    summary = json.loads([l for l in r.stderr.strip().splitlines()
+                          if l.startswith("{")][-1])
+    assert summary["auc"] > 0.6, f"quickstart model failed to learn: {summary}"
