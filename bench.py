@@ -9,7 +9,10 @@ W untimed warmup steps, then exactly K timed steps bracketed by
 barrier+synchronize on both sides; elapsed is the MAX over ranks; rank 0
 prints ONE JSON line.  Each timed step is a FULL training step: embedding
 gather + fused GEMM tower forward, fused loss, backward GEMMs, bucketed
-all-reduce (dense) + sparse allgather (embeddings), fused optimizer update.
+all-reduce (dense) + EP all-to-all or sparse allgather (embeddings), fused
+optimizer update.
+At world>1 the embeddings default to table-sharded EP (static all-to-all
+routing); --emb-mode dp selects replicated arenas + sparse allgather.
 Synthetic data (no network), random-init weights, weak scaling (per-GPU
 batch fixed).
 """
