@@ -10,7 +10,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 @pytest.mark.parametrize("tool", ["gemm_bench.py", "bench_configs.py",
-                                  "auc_parity.py", "splitk_sweep.py", "score_bench.py",
+                                  "auc_parity.py", "splitk_sweep.py", "score_bench.py", "bucket_sweep.py",
                                   "profile_summary.py"])
 def test_tool_help_or_import(tool):
     path = os.path.join(ROOT, "tools", tool)

@@ -1,49 +1,61 @@
 #!/usr/bin/env python3
-"""Multi-GPU knob sweep harness (round-2 tool; needs an N-GPU box).
+"""Sweep the all-reduce bucket size on the real bench at a given world size.
 
-Runs bench.py under torchrun for each (bucket_mb, emb_mode) combination and
-prints one result line per run.  Usage on an 8-GPU node:
+The 128 MB default was chosen from xGMI topology math (7 p2p links, ring
+all-reduce per-link bound — ARCHITECTURE §2); this sweeps it empirically on
+hardware.  Run on a multi-GPU node:
 
-    python tools/bucket_sweep.py --gpus 8 [--buckets 32,64,128,256] \
-        [--emb-modes ep,dp] [--steps 30] [--warmup 8]
+    python tools/bucket_sweep.py --gpus 8 [--buckets 32,64,128,256]
+        [--steps 30 --warmup 8]
 
-The all-reduce bucket size trades hook-overlap granularity against per-link
-xGMI message efficiency (7x ~153 GB/s point-to-point links); the shipped
-default (128 MB) was chosen analytically — this measures it.
+Each point launches bench.py via torch.distributed.run (one rank per GPU)
+and reports the whole-job samples/s.  Also usable at --gpus 1 (bucketing is
+then inert — a sanity floor) and on CPU/gloo for plumbing tests.
 """
 import argparse
 import json
 import subprocess
 import sys
+import os
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_point(gpus: int, bucket_mb: int, steps: int, warmup: int,
+              extra: list) -> float:
+    cmd = [sys.executable]
+    if gpus > 1:
+        cmd += ["-m", "torch.distributed.run", "--nnodes=1",
+                "--nproc-per-node", str(gpus),
+                "--master-addr", "127.0.0.1", "--master-port", "29741"]
+    cmd += [os.path.join(ROOT, "bench.py"), "--steps", str(steps),
+            "--warmup", str(warmup), "--bucket-mb", str(bucket_mb)] + extra
+    out = subprocess.run(cmd, capture_output=True, text=True, timeout=1800)
+    if out.returncode != 0:
+        print(out.stderr[-2000:], file=sys.stderr)
+        raise SystemExit(f"bench failed at bucket_mb={bucket_mb}")
+    line = [l for l in out.stdout.strip().splitlines() if l.startswith("{")][-1]
+    return float(json.loads(line)["value"])
 
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=8)
     ap.add_argument("--buckets", default="32,64,128,256")
-    ap.add_argument("--emb-modes", default="ep")
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=8)
-    ap.add_argument("--port", type=int, default=29581)
+    ap.add_argument("--extra", default="",
+                    help="extra bench.py flags, e.g. '--batch 16384'")
     args = ap.parse_args()
-
-    for mode in args.emb_modes.split(","):
-        for mb in (int(x) for x in args.buckets.split(",")):
-            cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-                   "--nproc-per-node", str(args.gpus),
-                   "--master-addr", "127.0.0.1", "--master-port", str(args.port),
-                   "bench.py", "--gpus", str(args.gpus),
-                   "--steps", str(args.steps), "--warmup", str(args.warmup),
-                   "--bucket-mb", str(mb), "--emb-mode", mode]
-            r = subprocess.run(cmd, capture_output=True, text=True, timeout=1800)
-            line = [l for l in r.stdout.strip().splitlines() if l.startswith("{")]
-            if r.returncode != 0 or not line:
-                print(f"bucket={mb} emb={mode}: FAILED\n{r.stderr[-800:]}",
-                      flush=True)
-                continue
-            d = json.loads(line[-1])
-            print(f"bucket={mb:>4} emb={mode}: {d['value']:,.0f} samples/s "
-                  f"({d['ms_per_step']:.3f} ms/step)", flush=True)
+    extra = args.extra.split() if args.extra else []
+    results = {}
+    for mb in [int(b) for b in args.buckets.split(",")]:
+        v = run_point(args.gpus, mb, args.steps, args.warmup, extra)
+        results[mb] = v
+        print(f"bucket_mb={mb:<5d} {v/1e6:8.2f}M samples/s", flush=True)
+    best = max(results, key=results.get)
+    print(json.dumps({"metric": "bucket_sweep", "gpus": args.gpus,
+                      "results": results, "best_bucket_mb": best}))
 
 
 if __name__ == "__main__":
