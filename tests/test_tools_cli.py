@@ -61,3 +61,12 @@ def test_score_bench_functional_cpu():
     assert batch["metric"] == "scoring_rows_per_sec" and batch["value"] > 0
     assert batch["rows"] == 4096
     assert row["metric"] == "scoring_row_latency_us" and row["value"] > 0
+
+
+def test_all_tools_compile():
+    """Every tools/*.py must at least parse/compile (catches bitrot in the
+    GPU-only tools that can't run here)."""
+    import glob
+    for path in sorted(glob.glob(os.path.join(ROOT, "tools", "*.py"))):
+        src = open(path).read()
+        compile(src, os.path.basename(path), "exec")
