@@ -108,6 +108,12 @@ class Launcher:
             attempts += 1
             if attempts > self.rc.max_rank_restarts:
                 raise RuntimeError(f"training failed after {attempts} attempts: {err}")
+            # drop partial metric sets from the crashed attempt: the restarted
+            # ranks re-emit the interrupted epoch from scratch, and mixing the
+            # two attempts' entries would double-count workers in its stats
+            self.epoch_results = defaultdict(
+                list, {e: v for e, v in self.epoch_results.items()
+                       if len(v) >= self.rc.num_gpus})
             self.board.write(f"[launcher] rank failure ({err}); restarting from "
                              f"latest checkpoint (attempt {attempts})")
 
