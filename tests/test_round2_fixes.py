@@ -357,3 +357,26 @@ def test_ep_step_cadence_checkpoint(tmp_path):
                                            extra=(str(tmp_path),)):
         assert ok_exists, f"rank {rank}: no mid-epoch checkpoint written"
         assert ok_resume, f"rank {rank}: resume bookkeeping wrong"
+
+
+def test_mixed_world_shard_debris(tmp_path):
+    """An epoch carrying a COMPLETE shard set for one world size plus debris
+    from another (aborted world-change experiment) is still selectable; an
+    epoch with only partial sets for every world is skipped."""
+    from shifu_amd.parallel.ep import ShardedEmbedding
+    from shifu_amd.train import checkpoint as ckpt
+
+    class M(torch.nn.Module):
+        def __init__(self, world, rank):
+            super().__init__()
+            self.emb = ShardedEmbedding([10], 4, seed=1, world=world, rank=rank)
+
+    # epoch 0: complete at world=2 AND one stray world=4 shard
+    for r in range(2):
+        ckpt.save_checkpoint(str(tmp_path), 0, 5, M(2, r), None, rank=r, world=2)
+    ckpt.save_checkpoint(str(tmp_path), 0, 5, M(4, 0), None, rank=0, world=4)
+    # epoch 1: partial for BOTH worlds
+    ckpt.save_checkpoint(str(tmp_path), 1, 9, M(2, 0), None, rank=0, world=2)
+    ckpt.save_checkpoint(str(tmp_path), 1, 9, M(4, 1), None, rank=1, world=4)
+    chosen = ckpt.latest_checkpoint(str(tmp_path), world=2)
+    assert chosen and chosen.endswith("ckpt-0.pt"), chosen
