@@ -67,6 +67,15 @@ def load_split_cached(cache_dir: Optional[str], files: Sequence[str],
     train, valid = build()
     try:
         os.makedirs(cache_dir, exist_ok=True)
+        # prune this shard's entries under older keys (changed data/spec):
+        # they can never be read again and a 100M-row entry is tens of GB
+        prefix = f"shard-{rank}of{world}-"
+        for name in os.listdir(cache_dir):
+            if name.startswith(prefix) and name != os.path.basename(path):
+                try:
+                    os.remove(os.path.join(cache_dir, name))
+                except OSError:
+                    pass
         tmp = path + f".tmp{os.getpid()}"
         torch.save({"train": _pack(train), "valid": _pack(valid)}, tmp)
         os.replace(tmp, path)
