@@ -136,3 +136,32 @@ def test_shard_cache_roundtrip_and_invalidation(tmp_path):
     # cache disabled
     _, _, c5 = load_split_cached(None, files, spec, 0, 1, build)
     assert not c5 and calls["n"] == 4
+
+
+def test_shard_cache_prunes_stale_keys(tmp_path):
+    """Writing a new entry for a (rank, world) removes that shard's
+    older-key entries (they are unreadable forever and can be tens of GB)."""
+    import os
+    import time
+    from shifu_amd.data.synthetic import generate_synthetic_csv
+    from shifu_amd.data.csv_loader import load_csv_files, list_training_files
+    from shifu_amd.data.shard_cache import load_split_cached
+
+    data_dir = tmp_path / "data"
+    generate_synthetic_csv(str(data_dir), 200, 4, (9,), seed=4, n_files=1)
+    files = list_training_files([str(data_dir)])
+    spec = {"num": [2, 3, 4, 5], "cat": [6], "target": 0, "weight": 1,
+            "valid": 0.2, "seed": 1}
+
+    def build():
+        full = load_csv_files(files, [2, 3, 4, 5], [6], 0, 1, "|")
+        return full.split(0.2, seed=1)
+
+    cache = str(tmp_path / "cache")
+    load_split_cached(cache, files, spec, 0, 1, build)
+    assert len(os.listdir(cache)) == 1
+    time.sleep(0.01)
+    os.utime(files[0])                       # new key for the same shard
+    load_split_cached(cache, files, spec, 0, 1, build)
+    entries = [n for n in os.listdir(cache) if n.startswith("shard-0of1-")]
+    assert len(entries) == 1, f"stale entry not pruned: {entries}"
